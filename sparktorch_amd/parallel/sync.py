@@ -1,0 +1,341 @@
+"""Synchronous data-parallel training engine.
+
+Re-design of the reference's sync engine (distributed.py:66-261) for MI355X:
+
+* one worker process = one rank = one GPU; RCCL (``"nccl"`` on ROCm) over
+  xGMI instead of gloo whenever the device is a GPU;
+* gradients live in flat ~25 MB buckets whose all-reduce launches the moment
+  the bucket's grads are complete, overlapping with the rest of backward
+  (:mod:`sparktorch_amd.parallel.buckets`) — vs one blocking per-parameter
+  all-reduce in the reference (distributed.py:179-181);
+* the optimizer is a fused flat-bucket HIP kernel when on GPU
+  (:mod:`sparktorch_amd.ops.optim`), folding the 1/world grad averaging into
+  the update instead of a separate divide pass;
+* ``compileMode`` is honored (the reference's torch.compile call is broken —
+  distributed.py:117-118 tests the *builtin* ``compile``); mode
+  ``"hipgraph"`` captures the whole train step in a HIP graph and replays it.
+
+Reference behaviors preserved: long-label retry for classification criteria
+(distributed.py:152-157), autoencoder mode when no labels (distributed.py:136),
+minibatch sampling (distributed.py:145-148), early-stop consensus via two tiny
+all-reduces (distributed.py:184-196), identical final states on every rank.
+"""
+
+from __future__ import annotations
+
+import random
+from typing import Any, Iterable, List, Optional
+
+import numpy as np
+import torch
+import torch.distributed as dist
+
+from sparktorch_amd.compat.barrier import get_barrier_context
+from sparktorch_amd.parallel.buckets import FlatBuckets
+from sparktorch_amd.parallel.rendezvous import (
+    cleanup_stale_process_group,
+    init_process_group_from_barrier,
+    pick_device,
+)
+from sparktorch_amd.utils.data import handle_features
+from sparktorch_amd.utils.early_stopper import EarlyStopping
+from sparktorch_amd.utils.serialize import (
+    load_base_torch,
+    load_torch_model,
+    serialize_torch_obj,
+)
+
+
+def compute_loss(criterion, pred: torch.Tensor, y: torch.Tensor) -> torch.Tensor:
+    """Criterion with the reference's long-label retry (distributed.py:152-157):
+    classification losses want int64 class targets."""
+    if isinstance(criterion, (torch.nn.CrossEntropyLoss, torch.nn.NLLLoss)):
+        return criterion(pred, y.flatten().long())
+    try:
+        return criterion(pred, y.float())
+    except RuntimeError:
+        return criterion(pred, y.flatten().long())
+
+
+class SyncTrainer:
+    """One rank's training state: model + flat buckets + (fused) optimizer.
+
+    Usable both from barrier workers (``handle_model``) and directly from
+    torchrun-launched benchmark ranks.
+    """
+
+    def __init__(
+        self,
+        model: torch.nn.Module,
+        criterion,
+        torch_optimizer: torch.optim.Optimizer,
+        device: str = "cpu",
+        world_size: Optional[int] = None,
+        bucket_cap_mb: float = 25.0,
+        compile_mode: Optional[str] = None,
+    ):
+        self.device = device
+        self.model = model.to(device)
+        self.criterion = criterion
+        self.world_size = (
+            world_size
+            if world_size is not None
+            else (dist.get_world_size() if dist.is_initialized() else 1)
+        )
+
+        self.buckets = FlatBuckets(
+            list(self.model.parameters()), bucket_cap_mb=bucket_cap_mb, world_size=self.world_size
+        )
+        self.buckets.register_hooks()
+
+        from sparktorch_amd.ops.optim import fused_optimizer_for
+
+        self.optimizer = fused_optimizer_for(torch_optimizer, self.buckets)
+        self._torch_opt = None
+        if self.optimizer is None:
+            # No fused mapping: rebuild the torch optimizer on the re-bound
+            # (flattened) parameters; averaging happens in finalize().
+            self._torch_opt = type(torch_optimizer)(
+                self.model.parameters(), **torch_optimizer.defaults
+            )
+
+        self._graph = None
+        self._graph_inputs = None
+        self.compile_mode = compile_mode
+        if compile_mode and compile_mode not in ("hipgraph",):
+            # Correct implementation of the reference's broken torch.compile
+            # hook (distributed.py:117-118).
+            self.model = torch.compile(self.model, mode=compile_mode)
+
+    # ------------------------------------------------------------------
+    def _forward_backward(self, x: torch.Tensor, y: torch.Tensor) -> torch.Tensor:
+        pred = self.model(x)
+        loss = compute_loss(self.criterion, pred, y)
+        loss.backward()
+        return loss
+
+    def train_step(self, x: torch.Tensor, y: torch.Tensor) -> float:
+        """zero -> fwd -> bwd (all-reduce overlapped) -> finalize -> step."""
+        if self.compile_mode == "hipgraph" and x.is_cuda:
+            return self._train_step_graphed(x, y)
+        self.buckets.zero_grad()
+        loss = self._forward_backward(x, y)
+        if self.optimizer is not None:
+            self.buckets.finalize(average=False)  # scale fused into the update
+            self.optimizer.step(grad_scale=1.0 / self.world_size)
+        else:
+            self.buckets.finalize(average=True)
+            self._torch_opt.step()
+        return float(loss.detach())
+
+    # ------------------------------------------------------------------
+    def _train_step_graphed(self, x: torch.Tensor, y: torch.Tensor) -> float:
+        """Whole-step HIP-graph capture and replay (compileMode='hipgraph').
+
+        The step (zero + fwd + bwd + fused step) is captured once per input
+        shape and replayed thereafter — removes every launch gap in
+        launch-bound small-model training.  Collectives are NOT captured;
+        graph mode therefore requires world_size == 1 or external averaging.
+        """
+        key = (tuple(x.shape), tuple(y.shape) if y is not None else None)
+        if self._graph is None or self._graph_inputs[0] != key:
+            if self.world_size > 1:
+                raise RuntimeError(
+                    "compileMode='hipgraph' currently supports world_size==1 "
+                    "(collectives are not graph-captured)"
+                )
+            static_x = x.clone()
+            static_y = y.clone() if y is not None else None
+            # warmup on a side stream (required before capture)
+            s = torch.cuda.Stream()
+            s.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(s):
+                for _ in range(2):
+                    self.buckets.zero_grad()
+                    loss = self._forward_backward(static_x, static_y)
+                    if self.optimizer is not None:
+                        self.optimizer.step(grad_scale=1.0)
+                    else:
+                        self._torch_opt.step()
+            torch.cuda.current_stream().wait_stream(s)
+            g = torch.cuda.CUDAGraph()
+            static_loss = torch.zeros((), device=x.device)
+            with torch.cuda.graph(g):
+                self.buckets.zero_grad()
+                loss = self._forward_backward(static_x, static_y)
+                static_loss.copy_(loss.detach())
+                if self.optimizer is not None:
+                    self.optimizer.step(grad_scale=1.0)
+                else:
+                    self._torch_opt.step()
+            self._graph = g
+            self._graph_inputs = (key, static_x, static_y, static_loss)
+        _, static_x, static_y, static_loss = self._graph_inputs
+        static_x.copy_(x)
+        if static_y is not None:
+            static_y.copy_(y)
+        self._graph.replay()
+        return float(static_loss)
+
+    # ------------------------------------------------------------------
+    @torch.no_grad()
+    def validation_loss(self, x_val: torch.Tensor, y_val: torch.Tensor) -> float:
+        self.model.eval()
+        loss = compute_loss(self.criterion, self.model(x_val), y_val)
+        self.model.train()
+        return float(loss)
+
+    def state_dict_cpu(self) -> dict:
+        return {k: v.detach().cpu() for k, v in self.model.state_dict().items()}
+
+
+# ----------------------------------------------------------------------------
+# Barrier-worker entry (reference handle_model, distributed.py:66-205)
+# ----------------------------------------------------------------------------
+
+
+def handle_model(
+    index: int,
+    iterator: Iterable,
+    torch_obj_str: str,
+    iters: int = 10,
+    verbose: int = 0,
+    mini_batch: int = -1,
+    validation_pct: float = 0.0,
+    device: str = "cpu",
+    early_stop_patience: int = -1,
+    backend: Optional[str] = None,
+    compile_mode: Optional[str] = None,
+) -> List[dict]:
+    if index < 0:
+        raise RuntimeError("invalid partition index")
+
+    ctx = get_barrier_context()
+    cleanup_stale_process_group()
+    rank = index
+    dev = pick_device(device, rank)
+    world_size = init_process_group_from_barrier(ctx, rank, dev, backend)
+
+    try:
+        loaded = load_torch_model(torch_obj_str, device=dev)
+        data = handle_features(iterator, validation_pct)
+        if data.x_train is None:
+            raise RuntimeError(
+                "rank %d received an empty partition; repartition so every "
+                "barrier task has data" % rank
+            )
+
+        x_train = data.x_train.to(dev)
+        # autoencoder mode: no labels -> y = x (reference distributed.py:136)
+        y_train = data.y_train.to(dev) if data.y_train is not None else x_train
+        x_val = data.x_val.to(dev) if data.x_val is not None else None
+        y_val = data.y_val.to(dev) if data.y_val is not None else (x_val if x_val is not None else None)
+
+        trainer = SyncTrainer(
+            loaded.model,
+            loaded.criterion,
+            loaded.optimizer,
+            device=dev,
+            world_size=world_size,
+            compile_mode=compile_mode,
+        )
+
+        es = EarlyStopping(patience=early_stop_patience) if early_stop_patience > 0 else None
+        n = x_train.shape[0]
+        # collectives must live on the comm device (RCCL wants GPU tensors)
+        comm_dev = dev if dev.startswith("cuda") else "cpu"
+        should_stop_t = torch.zeros(1, device=comm_dev)
+
+        for i in range(iters):
+            if 0 < mini_batch < n:
+                idx = torch.from_numpy(np.random.choice(n, mini_batch, replace=False))
+                xb = x_train[idx]
+                yb = y_train[idx]
+            else:
+                xb, yb = x_train, y_train
+
+            loss = trainer.train_step(xb, yb)
+
+            if es is not None:
+                loss_for_es = (
+                    trainer.validation_loss(x_val, y_val) if x_val is not None else loss
+                )
+                lt = torch.tensor([loss_for_es], device=comm_dev)
+                dist.all_reduce(lt)
+                avg = float(lt) / world_size
+                should_stop_t[0] = 1.0 if es.step(avg) else 0.0
+                dist.all_reduce(should_stop_t)
+                if float(should_stop_t) > 0:
+                    break
+
+            if verbose:
+                print("rank %d iter %d loss %.6f" % (rank, i, loss), flush=True)
+
+        # ranks end in identical states (grads synced; same init) —
+        # reference collects all and takes [0] (distributed.py:255-261)
+        return [trainer.state_dict_cpu()]
+    finally:
+        if dist.is_initialized():
+            dist.destroy_process_group()
+
+
+# ----------------------------------------------------------------------------
+# Driver entry (reference train_distributed, distributed.py:208-261)
+# ----------------------------------------------------------------------------
+
+
+def train_distributed(
+    rdd,
+    torch_obj: str,
+    iters: int = 10,
+    partition_shuffles: int = 1,
+    verbose: int = 0,
+    mini_batch: int = -1,
+    validation_pct: float = 0.0,
+    device: str = "cpu",
+    early_stop_patience: int = -1,
+    backend: Optional[str] = None,
+    compile_mode: Optional[str] = None,
+) -> dict:
+    """Run sync data-parallel training over a (barrier) RDD; returns the
+    trained state_dict."""
+    torch_obj_str, _shapes = load_base_torch(torch_obj)
+
+    state = None
+    for shuffle_round in range(max(1, partition_shuffles)):
+        obj_str = torch_obj_str
+        if state is not None:
+            # carry trained weights into the next shuffle round
+            loaded = load_torch_model(torch_obj_str)
+            loaded.model.load_state_dict(state)
+            obj_str, _ = load_base_torch(
+                serialize_torch_obj(
+                    loaded.model,
+                    loaded.criterion,
+                    type(loaded.optimizer),
+                    **loaded.optimizer.defaults,
+                )
+            )
+
+        def worker(index, iterator, _obj=obj_str):
+            return handle_model(
+                index,
+                iterator,
+                _obj,
+                iters=iters,
+                verbose=verbose,
+                mini_batch=mini_batch,
+                validation_pct=validation_pct,
+                device=device,
+                early_stop_patience=early_stop_patience,
+                backend=backend,
+                compile_mode=compile_mode,
+            )
+
+        states = rdd.mapPartitionsWithIndex(worker).collect()
+        state = states[0]
+
+        if shuffle_round + 1 < partition_shuffles:
+            rdd = rdd.repartition(rdd.getNumPartitions())
+
+    return state

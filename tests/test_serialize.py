@@ -1,0 +1,76 @@
+import json
+
+import numpy as np
+import pytest
+import torch
+import torch.nn as nn
+
+from sparktorch_amd import serialize_torch_obj, serialize_torch_obj_lazy
+from sparktorch_amd.models.simple_net import Net, NetworkWithParameters
+from sparktorch_amd.utils.codec import (
+    b64_to_obj,
+    obj_to_b64,
+    obj_to_stopwords,
+    stopwords_to_obj,
+)
+from sparktorch_amd.utils.serialize import load_base_torch, load_torch_model
+
+
+def test_b64_roundtrip():
+    obj = {"a": 1, "t": torch.ones(3)}
+    back = b64_to_obj(obj_to_b64(obj))
+    assert back["a"] == 1
+    assert torch.equal(back["t"], torch.ones(3))
+
+
+def test_stopwords_codec_roundtrip():
+    obj = {"x": list(range(100))}
+    sw = obj_to_stopwords(obj)
+    assert sw[-1] == "4c1740b00d3c4ff6806a1402321572cb"
+    assert all(tok == "" or 0 <= int(tok) < 256 for tok in sw[0].split(","))
+    assert stopwords_to_obj(sw) == obj
+
+
+def test_serialize_torch_obj_envelope():
+    model = Net()
+    s = serialize_torch_obj(model, nn.MSELoss(), torch.optim.Adam, lr=0.001)
+    d = json.loads(s)
+    assert set(d.keys()) == {"torch_obj", "shapes"}
+    # shapes match parameters: fc1 w/b, fc2 w/b
+    assert d["shapes"] == [[20, 10], [20], [1, 20], [1]]
+
+
+def test_load_torch_model_eager():
+    model = Net()
+    s = serialize_torch_obj(model, nn.MSELoss(), torch.optim.Adam, lr=0.005)
+    loaded = load_torch_model(s, from_json=True)
+    assert isinstance(loaded.model, Net)
+    assert isinstance(loaded.criterion, nn.MSELoss)
+    # dill round-trips torch classes by value, so compare by name
+    assert type(loaded.optimizer).__name__ == "Adam"
+    assert loaded.optimizer.defaults["lr"] == 0.005
+    # weights survive the round trip
+    x = torch.randn(4, 10)
+    assert torch.allclose(model(x), loaded.model(x))
+
+
+def test_serialize_lazy_with_params():
+    s = serialize_torch_obj_lazy(
+        NetworkWithParameters,
+        nn.MSELoss,
+        torch.optim.SGD,
+        optimizer_params={"lr": 0.01},
+        model_parameters={"input_dim": 10, "hidden_dim": 30, "output_dim": 2},
+    )
+    env, shapes = load_base_torch(s)
+    assert shapes == [[30, 10], [30], [2, 30], [2]]
+    loaded = load_torch_model(s, from_json=True)
+    assert loaded.model.fc1.out_features == 30
+    assert type(loaded.criterion).__name__ == "MSELoss"
+    assert type(loaded.optimizer).__name__ == "SGD"
+
+
+def test_lazy_no_params():
+    s = serialize_torch_obj_lazy(Net, nn.MSELoss, torch.optim.Adam, optimizer_params={"lr": 0.1})
+    loaded = load_torch_model(s, from_json=True)
+    assert isinstance(loaded.model, Net)

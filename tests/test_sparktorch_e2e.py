@@ -1,0 +1,233 @@
+"""End-to-end fit -> transform tests, mirroring the reference's test matrix
+(reference sparktorch/tests/test_sparktorch.py:13-269: local[2] + 2 partitions
+=> genuine world_size=2 rendezvous + all-reduce, here over the local barrier
+engine + gloo)."""
+
+import numpy as np
+import pytest
+import torch
+import torch.nn as nn
+
+from sparktorch_amd import (
+    LocalPipeline,
+    LocalPipelineModel,
+    PysparkPipelineWrapper,
+    SparkTorch,
+    create_spark_torch_model,
+    serialize_torch_obj,
+    serialize_torch_obj_lazy,
+)
+from sparktorch_amd.compat.local import LocalDataFrame
+from sparktorch_amd.models.simple_net import (
+    AutoEncoder,
+    ClassificationNet,
+    Net,
+    NetworkWithParameters,
+)
+
+
+@pytest.fixture(scope="module")
+def data_df():
+    """400 rows, 10-dim gaussians, two classes (means 0 and 2), 2 partitions
+    (reference test fixture, test_sparktorch.py:21-26)."""
+    rng = np.random.RandomState(42)
+    a = rng.normal(0.0, 1.0, (200, 10))
+    b = rng.normal(2.0, 1.0, (200, 10))
+    feats = np.concatenate([a, b])
+    labels = [0.0] * 200 + [1.0] * 200
+    perm = rng.permutation(400)
+    return LocalDataFrame.from_arrays(feats[perm], [labels[i] for i in perm], num_partitions=2)
+
+
+@pytest.fixture(scope="module")
+def general_torch_obj():
+    model = nn.Sequential(nn.Linear(10, 20), nn.ReLU(), nn.Linear(20, 1))
+    return serialize_torch_obj(model, nn.MSELoss(), torch.optim.Adam, lr=0.01)
+
+
+def _fit_transform(df, torch_obj, **kwargs):
+    defaults = dict(
+        inputCol="features",
+        labelCol="label",
+        predictionCol="predicted",
+        torchObj=torch_obj,
+        iters=5,
+        verbose=0,
+        mode="synchronous",
+    )
+    defaults.update(kwargs)
+    est = SparkTorch(**defaults)
+    model = est.fit(df)
+    out = model.transform(df)
+    return model, out
+
+
+def test_simple_sequential(data_df, general_torch_obj):
+    model, out = _fit_transform(data_df, general_torch_obj)
+    rows = out.collect()
+    assert len(rows) == 400
+    assert all(isinstance(r["predicted"], float) for r in rows)
+
+
+def test_nn_module(data_df):
+    obj = serialize_torch_obj(Net(), nn.MSELoss(), torch.optim.SGD, lr=0.01)
+    _model, out = _fit_transform(data_df, obj)
+    assert out.count() == 400
+
+
+def test_lazy(data_df):
+    obj = serialize_torch_obj_lazy(Net, nn.MSELoss, torch.optim.Adam, optimizer_params={"lr": 0.01})
+    _model, out = _fit_transform(data_df, obj)
+    assert out.count() == 400
+
+
+def test_lazy_network_with_params(data_df):
+    obj = serialize_torch_obj_lazy(
+        NetworkWithParameters,
+        nn.MSELoss,
+        torch.optim.Adam,
+        optimizer_params={"lr": 0.01},
+        model_parameters={"input_dim": 10, "hidden_dim": 15, "output_dim": 1},
+    )
+    model, out = _fit_transform(data_df, obj)
+    net = model.getPytorchModel()
+    assert net.fc1.out_features == 15
+    assert out.count() == 400
+
+
+def test_classification_crossentropy(data_df):
+    obj = serialize_torch_obj(
+        ClassificationNet(), nn.NLLLoss(), torch.optim.Adam, lr=0.01
+    )
+    _model, out = _fit_transform(data_df, obj, iters=10)
+    preds = {r["predicted"] for r in out.collect()}
+    assert preds.issubset({0.0, 1.0})
+
+
+def test_autoencoder_vector_out(data_df):
+    obj = serialize_torch_obj(AutoEncoder(), nn.MSELoss(), torch.optim.Adam, lr=0.01)
+    est = SparkTorch(
+        inputCol="features",
+        labelCol=None,
+        predictionCol="predicted",
+        torchObj=obj,
+        iters=5,
+        useVectorOut=True,
+        mode="synchronous",
+    )
+    model = est.fit(data_df)
+    out = model.transform(data_df)
+    row = out.collect()[0]
+    assert len(np.asarray(row["predicted"])) == 10
+
+
+def test_minibatch(data_df, general_torch_obj):
+    _model, out = _fit_transform(data_df, general_torch_obj, miniBatch=32, iters=5)
+    assert out.count() == 400
+
+
+def test_validation_pct_and_early_stop(data_df, general_torch_obj):
+    _model, out = _fit_transform(
+        data_df, general_torch_obj, validationPct=0.2, earlyStopPatience=2, iters=8
+    )
+    assert out.count() == 400
+
+
+def test_inference_parity_with_create_model(data_df, general_torch_obj):
+    model, out = _fit_transform(data_df, general_torch_obj)
+    net = model.getPytorchModel()
+    wrapped = create_spark_torch_model(net, inputCol="features", predictionCol="p2")
+    out2 = wrapped.transform(data_df)
+    p1 = [r["predicted"] for r in out.collect()]
+    p2 = [r["p2"] for r in out2.collect()]
+    np.testing.assert_allclose(p1, p2, rtol=1e-5)
+
+
+def test_pipeline_save_load_roundtrip(tmp_path, data_df, general_torch_obj):
+    est = SparkTorch(
+        inputCol="features",
+        labelCol="label",
+        predictionCol="predicted",
+        torchObj=general_torch_obj,
+        iters=3,
+        mode="synchronous",
+    )
+    pipeline = LocalPipeline(stages=[est])
+    fitted = pipeline.fit(data_df)
+
+    path = str(tmp_path / "pipe")
+    fitted.write().overwrite().save(path)
+
+    loaded = PysparkPipelineWrapper.unwrap(LocalPipelineModel.load(path))
+    out = loaded.transform(data_df)
+    expected = fitted.transform(data_df)
+    p1 = [r["predicted"] for r in expected.collect()]
+    p2 = [r["predicted"] for r in out.collect()]
+    np.testing.assert_allclose(p1, p2, rtol=1e-6)
+
+
+def test_hogwild_mode(data_df):
+    from sparktorch_amd.compat.local import free_port
+
+    obj = serialize_torch_obj(Net(), nn.MSELoss(), torch.optim.Adam, lr=0.01)
+    est = SparkTorch(
+        inputCol="features",
+        labelCol="label",
+        predictionCol="predicted",
+        torchObj=obj,
+        iters=5,
+        mode="hogwild",
+        port=free_port(),
+        acquireLock=True,
+    )
+    model = est.fit(data_df)
+    out = model.transform(data_df)
+    assert out.count() == 400
+
+
+def test_hogwild_lockfree(data_df):
+    from sparktorch_amd.compat.local import free_port
+
+    obj = serialize_torch_obj(Net(), nn.MSELoss(), torch.optim.SGD, lr=0.01)
+    est = SparkTorch(
+        inputCol="features",
+        labelCol="label",
+        predictionCol="predicted",
+        torchObj=obj,
+        iters=4,
+        mode="hogwild",
+        port=free_port(),
+        acquireLock=False,
+        earlyStopPatience=3,
+    )
+    model = est.fit(data_df)
+    assert model.transform(data_df).count() == 400
+
+
+def test_bad_mode_raises(data_df, general_torch_obj):
+    est = SparkTorch(
+        inputCol="features",
+        labelCol="label",
+        torchObj=general_torch_obj,
+        iters=1,
+        mode="nonsense",
+    )
+    with pytest.raises(ValueError):
+        est.fit(data_df)
+
+
+def test_training_improves_fit(data_df):
+    """Sanity: sync training actually reduces regression loss vs init."""
+    torch.manual_seed(0)
+    net = Net()
+    obj = serialize_torch_obj(net, nn.MSELoss(), torch.optim.Adam, lr=0.05)
+    model, out = _fit_transform(data_df, obj, iters=30)
+
+    rows = data_df.collect()
+    x = torch.tensor(np.stack([np.asarray(r["features"], dtype=np.float32) for r in rows]))
+    y = torch.tensor([[float(r["label"])] for r in rows])
+    trained = model.getPytorchModel()
+    with torch.no_grad():
+        trained_loss = nn.MSELoss()(trained(x), y).item()
+        init_loss = nn.MSELoss()(net(x), y).item()
+    assert trained_loss < init_loss

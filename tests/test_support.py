@@ -1,0 +1,131 @@
+import threading
+import time
+
+import numpy as np
+import pytest
+import torch
+
+from sparktorch_amd import EarlyStopping, RWLock
+from sparktorch_amd.utils.data import handle_data, handle_features
+from sparktorch_amd.utils.serialize import DataObj
+
+
+# --- early stopping -----------------------------------------------------------
+
+
+def test_early_stop_patience_zero_disabled():
+    es = EarlyStopping(patience=0)
+    for _ in range(100):
+        assert es.step(1.0) is False
+
+
+def test_early_stop_nan_stops():
+    es = EarlyStopping(patience=5)
+    assert es.step(float("nan")) is True
+
+
+def test_early_stop_min_mode():
+    es = EarlyStopping(mode="min", patience=2)
+    assert not es.step(1.0)
+    assert not es.step(0.9)
+    assert not es.step(0.95)  # bad 1
+    assert es.step(0.95)  # bad 2 -> stop
+
+
+def test_early_stop_max_mode_percentage():
+    es = EarlyStopping(mode="max", patience=1, min_delta=10, percentage=True)
+    assert not es.step(100.0)
+    # 105 is not >10% better -> bad epoch -> patience 1 exhausted
+    assert es.step(105.0)
+
+
+# --- RW lock ------------------------------------------------------------------
+
+
+def test_rwlock_readers_shared():
+    lock = RWLock()
+    lock.acquire_read()
+    lock.acquire_read()
+    lock.release()
+    lock.release()
+
+
+def test_rwlock_writer_exclusive():
+    lock = RWLock()
+    results = []
+
+    lock.acquire_write()
+
+    def reader():
+        lock.acquire_read()
+        results.append("read")
+        lock.release()
+
+    t = threading.Thread(target=reader)
+    t.start()
+    time.sleep(0.1)
+    assert results == []  # reader blocked by writer
+    lock.release()
+    t.join(timeout=5)
+    assert results == ["read"]
+
+
+def test_rwlock_release_unheld_raises():
+    with pytest.raises(RuntimeError):
+        RWLock().release()
+
+
+# --- data layer ---------------------------------------------------------------
+
+
+def _rows(n=10, dim=4, with_label=True):
+    out = []
+    for i in range(n):
+        out.append(
+            DataObj(
+                x_train=np.arange(dim, dtype=np.float64) + i,
+                y_train=float(i % 2) if with_label else None,
+                x_val=None,
+                y_val=None,
+            )
+        )
+    return out
+
+
+def test_handle_features_stacks_float32():
+    d = handle_features(_rows(10, 4))
+    assert d.x_train.shape == (10, 4)
+    assert d.x_train.dtype == torch.float32
+    assert d.y_train.shape == (10, 1)
+    assert d.x_val is None
+
+
+def test_handle_features_empty_partition():
+    d = handle_features([])
+    assert d.x_train is None and d.y_train is None
+
+
+def test_handle_features_validation_split():
+    d = handle_features(_rows(100, 4), validation_pct=0.2)
+    assert d.x_train.shape[0] == 80
+    assert d.x_val.shape[0] == 20
+    assert d.y_val.shape[0] == 20
+
+
+def test_handle_features_no_labels():
+    d = handle_features(_rows(10, 4, with_label=False))
+    assert d.y_train is None
+
+
+def test_handle_data_mapper():
+    rows = [{"features": np.ones(3), "label": 1.0}, {"features": np.zeros(3), "label": 0.0}]
+    mapped = list(handle_data("features", "label")(iter(rows)))
+    assert len(mapped) == 2
+    assert mapped[0].y_train == 1.0
+    np.testing.assert_array_equal(mapped[1].x_train, np.zeros(3))
+
+
+def test_handle_data_autoencoder_mode():
+    rows = [{"features": np.ones(3)}]
+    mapped = list(handle_data("features", None)(iter(rows)))
+    assert mapped[0].y_train is None
