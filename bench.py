@@ -1,0 +1,142 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: MNIST-784 3-layer MLP (784-256-256-10) synchronous
+data-parallel training — the BASELINE.json headline metric
+(samples/sec/node, weak scaling over 1/2/4/8 MI355X).
+
+Contract (driver): `python bench.py --gpus N --steps K --warmup W`; for N>1
+launched via torch.distributed.run with one rank per GPU over RCCL.  W
+untimed warmup steps, then exactly K timed steps bracketed by barrier +
+synchronize on both sides; elapsed is MAX over ranks; rank 0 prints ONE JSON
+line.
+
+The timed step is the full training step: zero -> fwd (fused MFMA linear+
+relu) -> fused CE (loss+dlogits one kernel) -> bwd (MFMA dgrad/wgrad,
+bucketed RCCL all-reduce overlapped with backward) -> fused Adam.  Synthetic
+data (randn features, uniform labels), random-init weights, bf16 activations
+with fp32 master weights & fp32 MFMA accumulation.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+
+
+def main() -> int:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=30)
+    ap.add_argument("--warmup", type=int, default=10)
+    ap.add_argument("--batch", type=int, default=32768, help="per-GPU batch (weak scaling)")
+    ap.add_argument("--device", type=str, default=None)
+    ap.add_argument("--hipgraph", action="store_true", help="capture the train step in a HIP graph")
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+
+    on_gpu = torch.cuda.is_available() if args.device is None else args.device.startswith("cuda")
+    if on_gpu:
+        torch.cuda.set_device(local_rank % max(1, torch.cuda.device_count()))
+        device = "cuda:%d" % (local_rank % max(1, torch.cuda.device_count()))
+    else:
+        device = "cpu"
+
+    if world > 1:
+        dist.init_process_group("nccl" if on_gpu else "gloo")
+
+    from sparktorch_amd.parallel.sync import SyncTrainer
+
+    torch.manual_seed(1234 + rank)
+    if on_gpu:
+        from sparktorch_amd.ops.modules import MnistMLPFused
+
+        model = MnistMLPFused()
+    else:
+        from sparktorch_amd.models.mnist import MnistMLP
+
+        model = MnistMLP()
+
+    opt = torch.optim.Adam(model.parameters(), lr=1e-3)
+    trainer = SyncTrainer(
+        model,
+        nn.CrossEntropyLoss(),
+        opt,
+        device=device,
+        world_size=world,
+        compile_mode="hipgraph" if (args.hipgraph and world == 1 and on_gpu) else None,
+    )
+
+    x = torch.randn(args.batch, 784, device=device)
+    if on_gpu:
+        x = x.to(torch.bfloat16)
+    y = torch.randint(0, 10, (args.batch,), device=device)
+
+    def barrier_sync():
+        if world > 1:
+            dist.barrier()
+        if on_gpu:
+            torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        trainer.train_step(x, y)
+
+    barrier_sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        trainer.train_step(x, y)
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks
+    if world > 1:
+        et = torch.tensor([elapsed], device=device if on_gpu else "cpu")
+        dist.all_reduce(et, op=dist.ReduceOp.MAX)
+        elapsed = float(et)
+
+    n_gpus = world if world > 1 else (args.gpus if on_gpu else 1)
+    total_samples = args.batch * n_gpus * args.steps
+    samples_per_sec = total_samples / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+
+    if rank == 0:
+        out = {
+            "metric": "samples_per_sec",
+            "value": samples_per_sec,
+            "unit": "samples/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,  # reference publishes no numbers
+            "dtype": "bf16" if on_gpu else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": "mnist_mlp_784x256x256x10",
+                "global_batch": args.batch * n_gpus,
+                "seq_len": None,
+                "parallelism": "dp%d" % n_gpus,
+                "optimizer": "fused_adam",
+                "loss": "cross_entropy_fused",
+                "hipgraph": bool(args.hipgraph),
+            },
+        }
+        print(json.dumps(out), flush=True)
+
+    if world > 1:
+        dist.destroy_process_group()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -1,0 +1,208 @@
+// Python bindings for the sparktorch_amd gfx950 kernels (_sparkhip).
+// Host-only translation unit: tensor validation + launcher calls on the
+// current HIP stream.  Device code lives in the .hip files.
+
+#include <torch/extension.h>
+
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+
+typedef uint16_t bf16raw;
+
+extern "C" {
+hipError_t launch_fused_adam(float*, const float*, float*, float*, int64_t, float, float, float,
+                             float, float, float, float, float, int, hipStream_t);
+hipError_t launch_fused_sgd(float*, const float*, float*, int64_t, float, float, float, float,
+                            float, int, int, int, hipStream_t);
+hipError_t launch_relu_bwd(const bf16raw*, const bf16raw*, bf16raw*, int64_t, hipStream_t);
+hipError_t launch_bias_grad(const bf16raw*, float*, int, int, hipStream_t);
+hipError_t launch_cast_f64_f32(const double*, float*, int64_t, hipStream_t);
+hipError_t launch_cast_f32_bf16(const float*, bf16raw*, int64_t, hipStream_t);
+hipError_t launch_ce_fused(const bf16raw*, const int64_t*, float*, bf16raw*, int, int,
+                           hipStream_t);
+hipError_t launch_mse_fused(const bf16raw*, const bf16raw*, float*, bf16raw*, int64_t,
+                            hipStream_t);
+hipError_t launch_gemm_bf16(const void*, const void*, int, float*, bf16raw*, const float*, int,
+                            int, int, int64_t, int64_t, int64_t, int64_t, int, int, hipStream_t);
+}
+
+#define CHECK_HIP(err)                                                              \
+  TORCH_CHECK((err) == hipSuccess, "HIP kernel launch failed: ", hipGetErrorString(err))
+
+static hipStream_t cur_stream() {
+  return (hipStream_t)c10::hip::getCurrentHIPStream().stream();
+}
+
+static void check_gpu_contig(const at::Tensor& t, c10::ScalarType dt, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+  TORCH_CHECK(t.scalar_type() == dt, name, " has wrong dtype");
+}
+
+// ---------------------------------------------------------------------------
+
+void fused_adam(at::Tensor p, at::Tensor g, at::Tensor m, at::Tensor v, double lr, double b1,
+                double b2, double eps, double wd, double bc1, double bc2, double gscale,
+                bool adamw) {
+  check_gpu_contig(p, at::kFloat, "p");
+  check_gpu_contig(g, at::kFloat, "g");
+  CHECK_HIP(launch_fused_adam(p.data_ptr<float>(), g.data_ptr<float>(), m.data_ptr<float>(),
+                              v.data_ptr<float>(), p.numel(), (float)lr, (float)b1, (float)b2,
+                              (float)eps, (float)wd, (float)bc1, (float)bc2, (float)gscale,
+                              adamw ? 1 : 0, cur_stream()));
+}
+
+void fused_sgd(at::Tensor p, at::Tensor g, at::Tensor buf, double lr, double momentum, double wd,
+               double dampening, double gscale, bool nesterov, bool first, bool has_momentum) {
+  check_gpu_contig(p, at::kFloat, "p");
+  CHECK_HIP(launch_fused_sgd(p.data_ptr<float>(), g.data_ptr<float>(), buf.data_ptr<float>(),
+                             p.numel(), (float)lr, (float)momentum, (float)wd, (float)dampening,
+                             (float)gscale, nesterov ? 1 : 0, first ? 1 : 0, has_momentum ? 1 : 0,
+                             cur_stream()));
+}
+
+at::Tensor relu_bwd(at::Tensor dy, at::Tensor y) {
+  check_gpu_contig(dy, at::kBFloat16, "dy");
+  check_gpu_contig(y, at::kBFloat16, "y");
+  auto dz = at::empty_like(dy);
+  CHECK_HIP(launch_relu_bwd((const bf16raw*)dy.data_ptr(), (const bf16raw*)y.data_ptr(),
+                            (bf16raw*)dz.data_ptr(), dy.numel(), cur_stream()));
+  return dz;
+}
+
+at::Tensor bias_grad(at::Tensor dz) {
+  check_gpu_contig(dz, at::kBFloat16, "dz");
+  TORCH_CHECK(dz.dim() == 2);
+  auto db = at::zeros({dz.size(1)}, dz.options().dtype(at::kFloat));
+  CHECK_HIP(launch_bias_grad((const bf16raw*)dz.data_ptr(), db.data_ptr<float>(),
+                             (int)dz.size(0), (int)dz.size(1), cur_stream()));
+  return db;
+}
+
+at::Tensor cast_f64_f32(at::Tensor src) {
+  check_gpu_contig(src, at::kDouble, "src");
+  auto dst = at::empty(src.sizes(), src.options().dtype(at::kFloat));
+  CHECK_HIP(launch_cast_f64_f32(src.data_ptr<double>(), dst.data_ptr<float>(), src.numel(),
+                                cur_stream()));
+  return dst;
+}
+
+at::Tensor cast_f32_bf16(at::Tensor src) {
+  check_gpu_contig(src, at::kFloat, "src");
+  auto dst = at::empty(src.sizes(), src.options().dtype(at::kBFloat16));
+  CHECK_HIP(launch_cast_f32_bf16(src.data_ptr<float>(), (bf16raw*)dst.data_ptr(), src.numel(),
+                                 cur_stream()));
+  return dst;
+}
+
+std::tuple<at::Tensor, at::Tensor> ce_fused(at::Tensor logits, at::Tensor targets) {
+  check_gpu_contig(logits, at::kBFloat16, "logits");
+  check_gpu_contig(targets, at::kLong, "targets");
+  TORCH_CHECK(logits.dim() == 2);
+  auto loss = at::zeros({}, logits.options().dtype(at::kFloat));
+  auto dlogits = at::empty_like(logits);
+  CHECK_HIP(launch_ce_fused((const bf16raw*)logits.data_ptr(), targets.data_ptr<int64_t>(),
+                            loss.data_ptr<float>(), (bf16raw*)dlogits.data_ptr(),
+                            (int)logits.size(0), (int)logits.size(1), cur_stream()));
+  return {loss, dlogits};
+}
+
+std::tuple<at::Tensor, at::Tensor> mse_fused(at::Tensor pred, at::Tensor target) {
+  check_gpu_contig(pred, at::kBFloat16, "pred");
+  check_gpu_contig(target, at::kBFloat16, "target");
+  auto loss = at::zeros({}, pred.options().dtype(at::kFloat));
+  auto dpred = at::empty_like(pred);
+  CHECK_HIP(launch_mse_fused((const bf16raw*)pred.data_ptr(), (const bf16raw*)target.data_ptr(),
+                             loss.data_ptr<float>(), (bf16raw*)dpred.data_ptr(), pred.numel(),
+                             cur_stream()));
+  return {loss, dpred};
+}
+
+// ---------------------------------------------------------------------------
+// Linear family (see gemm.hip header comment for the stride mappings)
+// ---------------------------------------------------------------------------
+
+at::Tensor linear_fwd(at::Tensor x, at::Tensor w, c10::optional<at::Tensor> bias, bool relu) {
+  check_gpu_contig(x, at::kBFloat16, "x");
+  TORCH_CHECK(w.is_cuda() && w.is_contiguous(), "w must be contiguous GPU");
+  TORCH_CHECK(w.scalar_type() == at::kFloat || w.scalar_type() == at::kBFloat16);
+  int64_t M = x.size(0), K = x.size(1), N = w.size(0);
+  TORCH_CHECK(w.size(1) == K, "shape mismatch");
+  auto y = at::empty({M, N}, x.options());
+  const float* bptr = nullptr;
+  int epi = 1;  // EPI_BF16
+  if (bias.has_value()) {
+    check_gpu_contig(*bias, at::kFloat, "bias");
+    bptr = bias->data_ptr<float>();
+    epi = relu ? 3 : 2;
+  } else {
+    TORCH_CHECK(!relu, "relu without bias not wired");
+  }
+  CHECK_HIP(launch_gemm_bf16(x.data_ptr(), w.data_ptr(), w.scalar_type() == at::kFloat ? 1 : 0,
+                             nullptr, (bf16raw*)y.data_ptr(), bptr, (int)M, (int)N, (int)K,
+                             /*sam*/ K, /*sak*/ 1, /*sbk*/ 1, /*sbn*/ K, epi, 1, cur_stream()));
+  return y;
+}
+
+at::Tensor linear_dgrad(at::Tensor dz, at::Tensor w) {
+  check_gpu_contig(dz, at::kBFloat16, "dz");
+  TORCH_CHECK(w.is_cuda() && w.is_contiguous());
+  int64_t M = dz.size(0), N = dz.size(1), K = w.size(1);
+  TORCH_CHECK(w.size(0) == N, "shape mismatch");
+  auto dx = at::empty({M, K}, dz.options());
+  CHECK_HIP(launch_gemm_bf16(dz.data_ptr(), w.data_ptr(), w.scalar_type() == at::kFloat ? 1 : 0,
+                             nullptr, (bf16raw*)dx.data_ptr(), nullptr, (int)M, (int)K, (int)N,
+                             /*sam*/ N, /*sak*/ 1, /*sbk*/ K, /*sbn*/ 1, 1, 1, cur_stream()));
+  return dx;
+}
+
+at::Tensor linear_wgrad(at::Tensor dz, at::Tensor x, int64_t splitk) {
+  check_gpu_contig(dz, at::kBFloat16, "dz");
+  check_gpu_contig(x, at::kBFloat16, "x");
+  int64_t B = dz.size(0), N = dz.size(1), K = x.size(1);
+  TORCH_CHECK(x.size(0) == B, "shape mismatch");
+  at::Tensor dw = splitk > 1 ? at::zeros({N, K}, x.options().dtype(at::kFloat))
+                             : at::empty({N, K}, x.options().dtype(at::kFloat));
+  CHECK_HIP(launch_gemm_bf16(dz.data_ptr(), x.data_ptr(), /*b_is_f32*/ 0, dw.data_ptr<float>(),
+                             nullptr, nullptr, (int)N, (int)K, (int)B,
+                             /*sam*/ 1, /*sak*/ N, /*sbk*/ K, /*sbn*/ 1, 0, (int)splitk,
+                             cur_stream()));
+  return dw;
+}
+
+// generic C = A @ B with optional transposes, for tests/other models
+at::Tensor matmul_bf16(at::Tensor a, at::Tensor b, bool trans_a, bool trans_b) {
+  check_gpu_contig(a, at::kBFloat16, "a");
+  TORCH_CHECK(b.is_cuda() && b.is_contiguous());
+  int64_t M = trans_a ? a.size(1) : a.size(0);
+  int64_t Ka = trans_a ? a.size(0) : a.size(1);
+  int64_t Kb = trans_b ? b.size(1) : b.size(0);
+  int64_t N = trans_b ? b.size(0) : b.size(1);
+  TORCH_CHECK(Ka == Kb, "inner dims mismatch");
+  int64_t sam = trans_a ? 1 : Ka;
+  int64_t sak = trans_a ? M : 1;
+  int64_t sbk = trans_b ? 1 : N;
+  int64_t sbn = trans_b ? Kb : 1;
+  auto c = at::empty({M, N}, a.options().dtype(at::kFloat));
+  CHECK_HIP(launch_gemm_bf16(a.data_ptr(), b.data_ptr(), b.scalar_type() == at::kFloat ? 1 : 0,
+                             c.data_ptr<float>(), nullptr, nullptr, (int)M, (int)N, (int)Ka, sam,
+                             sak, sbk, sbn, 0, 1, cur_stream()));
+  return c;
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("fused_adam", &fused_adam, "fused Adam on a flat bucket");
+  m.def("fused_sgd", &fused_sgd, "fused SGD on a flat bucket");
+  m.def("relu_bwd", &relu_bwd, "dz = dy * (y>0)");
+  m.def("bias_grad", &bias_grad, "column-sum of dz");
+  m.def("cast_f64_f32", &cast_f64_f32);
+  m.def("cast_f32_bf16", &cast_f32_bf16);
+  m.def("ce_fused", &ce_fused, "cross-entropy fwd+bwd fused");
+  m.def("mse_fused", &mse_fused, "mse fwd+bwd fused");
+  m.def("linear_fwd", &linear_fwd, "Y = X W^T (+bias)(+relu), MFMA");
+  m.def("linear_dgrad", &linear_dgrad, "dX = dZ W, MFMA");
+  m.def("linear_wgrad", &linear_wgrad, "dW = dZ^T X, MFMA split-K");
+  m.def("matmul_bf16", &matmul_bf16, "generic bf16 MFMA GEMM");
+}

@@ -1,0 +1,244 @@
+// Elementwise / optimizer kernels for gfx950.
+//
+// Replaces (reference file:line):
+//  * fused_adam / fused_sgd — optimizer.step() on CPU per parameter tensor
+//    (reference distributed.py:198, server.py:139): ONE grid-stride float4
+//    kernel per flat bucket, 1/world grad averaging folded in.
+//  * relu_bwd — the dY*(Y>0) mask for the fused linear backward.
+//  * bias_grad — column sum of dZ (bf16 -> fp32).
+//  * cast_f64_f32 — Spark DenseVector rows arrive float64; pack/cast on
+//    device (reference does np.stack + .float() on CPU, util.py:87-99).
+//
+// All memory-bound: vectorized loads (float4 / 8x bf16), grid-stride loops,
+// grid capped so the scheduler keeps ~8 blocks/CU (guide §6 G11/G13).
+
+#include "common.h"
+
+static inline int ew_grid(int64_t work_items, int block) {
+  int64_t g = ceil_div_i64(work_items, block);
+  if (g > 2048) g = 2048;
+  if (g < 1) g = 1;
+  return (int)g;
+}
+
+// ---------------------------------------------------------------------------
+// fused Adam
+// ---------------------------------------------------------------------------
+
+__global__ void fused_adam_kernel(float* __restrict__ p, const float* __restrict__ g,
+                                  float* __restrict__ m, float* __restrict__ v, int64_t n,
+                                  float lr, float b1, float b2, float eps, float wd, float bc1,
+                                  float bc2, float gscale, int adamw) {
+  int64_t nvec = n >> 2;
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  const floatx4* g4 = (const floatx4*)g;
+  floatx4* p4 = (floatx4*)p;
+  floatx4* m4 = (floatx4*)m;
+  floatx4* v4 = (floatx4*)v;
+  for (int64_t k = i; k < nvec; k += stride) {
+    floatx4 gv = g4[k] * gscale;
+    floatx4 pv = p4[k];
+    floatx4 mv = m4[k];
+    floatx4 vv = v4[k];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      float gj = gv[j];
+      float pj = pv[j];
+      if (wd != 0.f) {
+        if (adamw)
+          pj *= (1.f - lr * wd);
+        else
+          gj += wd * pj;
+      }
+      float mj = b1 * mv[j] + (1.f - b1) * gj;
+      float vj = b2 * vv[j] + (1.f - b2) * gj * gj;
+      float denom = sqrtf(vj / bc2) + eps;
+      pj -= (lr / bc1) * (mj / denom);
+      pv[j] = pj;
+      mv[j] = mj;
+      vv[j] = vj;
+    }
+    p4[k] = pv;
+    m4[k] = mv;
+    v4[k] = vv;
+  }
+  // scalar tail
+  for (int64_t k = (nvec << 2) + i; k < n; k += stride) {
+    float gj = g[k] * gscale;
+    float pj = p[k];
+    if (wd != 0.f) {
+      if (adamw)
+        pj *= (1.f - lr * wd);
+      else
+        gj += wd * pj;
+    }
+    float mj = b1 * m[k] + (1.f - b1) * gj;
+    float vj = b2 * v[k] + (1.f - b2) * gj * gj;
+    pj -= (lr / bc1) * (mj / (sqrtf(vj / bc2) + eps));
+    p[k] = pj;
+    m[k] = mj;
+    v[k] = vj;
+  }
+}
+
+extern "C" hipError_t launch_fused_adam(float* p, const float* g, float* m, float* v, int64_t n,
+                                        float lr, float b1, float b2, float eps, float wd,
+                                        float bc1, float bc2, float gscale, int adamw,
+                                        hipStream_t stream) {
+  int block = 256;
+  fused_adam_kernel<<<ew_grid(n / 4 + 1, block), block, 0, stream>>>(
+      p, g, m, v, n, lr, b1, b2, eps, wd, bc1, bc2, gscale, adamw);
+  HIP_CHECK_LAUNCH();
+  return hipSuccess;
+}
+
+// ---------------------------------------------------------------------------
+// fused SGD
+// ---------------------------------------------------------------------------
+
+__global__ void fused_sgd_kernel(float* __restrict__ p, const float* __restrict__ g,
+                                 float* __restrict__ buf, int64_t n, float lr, float momentum,
+                                 float wd, float dampening, float gscale, int nesterov, int first,
+                                 int has_momentum) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t k = i; k < n; k += stride) {
+    float gj = g[k] * gscale;
+    float pj = p[k];
+    if (wd != 0.f) gj += wd * pj;
+    if (has_momentum) {
+      float bj = first ? gj : momentum * buf[k] + (1.f - dampening) * gj;
+      buf[k] = bj;
+      gj = nesterov ? gj + momentum * bj : bj;
+    }
+    p[k] = pj - lr * gj;
+  }
+}
+
+extern "C" hipError_t launch_fused_sgd(float* p, const float* g, float* buf, int64_t n, float lr,
+                                       float momentum, float wd, float dampening, float gscale,
+                                       int nesterov, int first, int has_momentum,
+                                       hipStream_t stream) {
+  int block = 256;
+  fused_sgd_kernel<<<ew_grid(n, block), block, 0, stream>>>(
+      p, g, buf, n, lr, momentum, wd, dampening, gscale, nesterov, first, has_momentum);
+  HIP_CHECK_LAUNCH();
+  return hipSuccess;
+}
+
+// ---------------------------------------------------------------------------
+// ReLU backward mask: dz = dy * (y > 0); bf16 in/out, 8-wide vectorized
+// ---------------------------------------------------------------------------
+
+__global__ void relu_bwd_kernel(const bf16raw* __restrict__ dy, const bf16raw* __restrict__ y,
+                                bf16raw* __restrict__ dz, int64_t n) {
+  int64_t nvec = n >> 3;
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  const shortx8* dy8 = (const shortx8*)dy;
+  const shortx8* y8 = (const shortx8*)y;
+  shortx8* dz8 = (shortx8*)dz;
+  for (int64_t k = i; k < nvec; k += stride) {
+    shortx8 d = dy8[k];
+    shortx8 yv = y8[k];
+    shortx8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      // y > 0 in bf16 bits: sign bit clear and magnitude nonzero
+      uint16_t yb = (uint16_t)yv[j];
+      bool pos = (yb & 0x8000u) == 0 && yb != 0;
+      o[j] = pos ? d[j] : (short)0;
+    }
+    dz8[k] = o;
+  }
+  for (int64_t k = (nvec << 3) + i; k < n; k += stride) {
+    uint16_t yb = y[k];
+    bool pos = (yb & 0x8000u) == 0 && yb != 0;
+    dz[k] = pos ? dy[k] : (bf16raw)0;
+  }
+}
+
+extern "C" hipError_t launch_relu_bwd(const bf16raw* dy, const bf16raw* y, bf16raw* dz, int64_t n,
+                                      hipStream_t stream) {
+  int block = 256;
+  relu_bwd_kernel<<<ew_grid(n / 8 + 1, block), block, 0, stream>>>(dy, y, dz, n);
+  HIP_CHECK_LAUNCH();
+  return hipSuccess;
+}
+
+// ---------------------------------------------------------------------------
+// bias grad: db[n] = sum_m dz[m][n]; dz bf16 row-major [M,N], db fp32 (zeroed)
+// grid: (ceil(N/64), m_chunks); each thread owns one column within the chunk
+// and strides rows; per-thread partial -> atomicAdd (few chunks => low
+// contention).
+// ---------------------------------------------------------------------------
+
+__global__ void bias_grad_kernel(const bf16raw* __restrict__ dz, float* __restrict__ db, int M,
+                                 int N, int rows_per_chunk) {
+  int n = blockIdx.x * blockDim.x + threadIdx.x;
+  if (n >= N) return;
+  int m0 = blockIdx.y * rows_per_chunk;
+  int m1 = min(M, m0 + rows_per_chunk);
+  float acc = 0.f;
+  for (int m = m0; m < m1; ++m) acc += bf16_to_f32(dz[(int64_t)m * N + n]);
+  atomicAdd(db + n, acc);
+}
+
+extern "C" hipError_t launch_bias_grad(const bf16raw* dz, float* db, int M, int N,
+                                       hipStream_t stream) {
+  int block = 64;
+  int rows_per_chunk = 1024;
+  dim3 grid((N + block - 1) / block, (unsigned)ceil_div_i64(M, rows_per_chunk));
+  bias_grad_kernel<<<grid, block, 0, stream>>>(dz, db, M, N, rows_per_chunk);
+  HIP_CHECK_LAUNCH();
+  return hipSuccess;
+}
+
+// ---------------------------------------------------------------------------
+// fp64 -> fp32 cast (Spark DenseVector pack)
+// ---------------------------------------------------------------------------
+
+__global__ void cast_f64_f32_kernel(const double* __restrict__ src, float* __restrict__ dst,
+                                    int64_t n) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t k = i; k < n; k += stride) dst[k] = (float)src[k];
+}
+
+extern "C" hipError_t launch_cast_f64_f32(const double* src, float* dst, int64_t n,
+                                          hipStream_t stream) {
+  int block = 256;
+  cast_f64_f32_kernel<<<ew_grid(n, block), block, 0, stream>>>(src, dst, n);
+  HIP_CHECK_LAUNCH();
+  return hipSuccess;
+}
+
+// ---------------------------------------------------------------------------
+// f32 <-> bf16 casts (activation ingress), vectorized
+// ---------------------------------------------------------------------------
+
+__global__ void cast_f32_bf16_kernel(const float* __restrict__ src, bf16raw* __restrict__ dst,
+                                     int64_t n) {
+  int64_t nvec = n >> 2;
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  const floatx4* s4 = (const floatx4*)src;
+  shortx4* d4 = (shortx4*)dst;
+  for (int64_t k = i; k < nvec; k += stride) {
+    floatx4 v = s4[k];
+    shortx4 o;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) o[j] = (short)f32_to_bf16(v[j]);
+    d4[k] = o;
+  }
+  for (int64_t k = (nvec << 2) + i; k < n; k += stride) dst[k] = f32_to_bf16(src[k]);
+}
+
+extern "C" hipError_t launch_cast_f32_bf16(const float* src, bf16raw* dst, int64_t n,
+                                           hipStream_t stream) {
+  int block = 256;
+  cast_f32_bf16_kernel<<<ew_grid(n / 4 + 1, block), block, 0, stream>>>(src, dst, n);
+  HIP_CHECK_LAUNCH();
+  return hipSuccess;
+}

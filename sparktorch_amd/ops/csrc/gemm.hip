@@ -1,0 +1,263 @@
+// Hand-written MFMA GEMM for gfx950 (CDNA4) — the hot op behind Linear
+// forward/backward (replaces the reference's eager model(x)/loss.backward()
+// CPU path, reference distributed.py:150,176).
+//
+// C[M,N] = A[M,K] x B[K,N]; A is bf16, B is bf16 or fp32 (weights are fp32
+// master copies in the flat buckets and are cast to bf16 during the LDS
+// stage); fp32 MFMA accumulation (v_mfma_f32_16x16x32_bf16).
+//
+// Geometry: 128x128 block tile, BK=32, 256 threads = 4 waves in 2x2, each
+// wave owns a 64x64 sub-tile = 4x4 fragments of 16x16x32 MFMA; accumulators
+// live in AGPRs (4 fp32 per lane per fragment).  LDS holds A as [BM][BK] and
+// B TRANSPOSED as [BN][BK] (+8 bf16 pad per row, preserving the 16-byte
+// alignment of every ds_read_b128 fragment read — guide §2/§6 G4/G17), so
+// both fragment loads are contiguous 16-byte LDS reads.
+//
+// One GEMM serves all three Linear gradients via strides:
+//   fwd   Y = X  W^T : A=X(row),   B=W^T (sbk=1,   sbn=K)
+//   dX    = dZ W     : A=dZ(row),  B=W   (sbk=Kin, sbn=1)
+//   dW    = dZ^T X   : A=dZ^T (sam=1, sak=N), B=X (sbk=Kin, sbn=1), split-K
+//           over the batch dim with fp32 atomicAdd combine (fills 256 CUs
+//           even for small weight shapes).
+// Epilogues fuse bias add + ReLU into the C-write (EPI codes below).
+
+#include "common.h"
+
+#define BM 128
+#define BN 128
+#define BK 32
+#define LDS_PAD 8          // 8 bf16 = 16 B: rows stay 16-byte aligned
+#define LDSK (BK + LDS_PAD)
+
+// EPI codes
+#define EPI_F32 0        // fp32 store (atomicAdd when SPLITK)
+#define EPI_BF16 1       // bf16 store
+#define EPI_BIAS 2       // bf16 store, + bias[n]
+#define EPI_BIAS_RELU 3  // bf16 store, + bias[n], relu
+
+typedef shortx8 frag_t;  // 8 bf16 (4 VGPRs)
+
+// Stage a [ROWS x BK] tile (ROWS=128) into LDS laid out [ROWS][LDSK].
+// Logical element (r, k) comes from src[(row0+r)*srow + (kt+k)*skol].
+// 256 threads x 16 elements.  Two layouts:
+//   skol==1 (k-contiguous): thread covers 16 consecutive k of one row.
+//   else if srow==1 (row-contiguous): thread covers 16 consecutive rows at
+//     fixed k (vectorized along rows, strided LDS column writes).
+//   else: scalar element loop.
+template <bool SRC_F32>
+__device__ __forceinline__ void stage_tile(const void* __restrict__ src, bf16raw* __restrict__ lds,
+                                           int row0, int rmax, int kt, int kmax, int64_t srow,
+                                           int64_t skol) {
+  const int t = threadIdx.x;
+  if (skol == 1) {
+    // 2 threads per row; 16 consecutive k each
+    int r = t >> 1;
+    int k0 = (t & 1) * 16;
+    int gr = row0 + r;
+    bf16raw* dst = lds + r * LDSK + k0;
+    if (gr < rmax) {
+      const char* base = (const char*)src + (int64_t)gr * srow * (SRC_F32 ? 4 : 2);
+      int krem = kmax - kt - k0;  // how many of our 16 k are in range
+      if (krem >= 16) {
+        if (SRC_F32) {
+          const float* s = (const float*)base + kt + k0;
+#pragma unroll
+          for (int v = 0; v < 4; ++v) {
+            floatx4 x = *(const floatx4*)(s + v * 4);
+#pragma unroll
+            for (int j = 0; j < 4; ++j) dst[v * 4 + j] = f32_to_bf16(x[j]);
+          }
+        } else {
+          const bf16raw* s = (const bf16raw*)base + kt + k0;
+          *(shortx8*)dst = *(const shortx8*)s;
+          *(shortx8*)(dst + 8) = *(const shortx8*)(s + 8);
+        }
+      } else {
+#pragma unroll
+        for (int j = 0; j < 16; ++j) {
+          int k = kt + k0 + j;
+          if (k < kmax)
+            dst[j] = SRC_F32 ? f32_to_bf16(((const float*)base)[k]) : ((const bf16raw*)base)[k];
+          else
+            dst[j] = 0;
+        }
+      }
+    } else {
+#pragma unroll
+      for (int j = 0; j < 16; ++j) dst[j] = 0;
+    }
+  } else if (srow == 1) {
+    // 16 consecutive rows at fixed k; 8 threads per k column
+    int k = t >> 3;           // 0..31
+    int r0 = (t & 7) * 16;    // 0,16,...,112
+    int gk = kt + k;
+    if (gk < kmax) {
+      const char* base = (const char*)src + (int64_t)gk * skol * (SRC_F32 ? 4 : 2);
+      int rrem = rmax - row0 - r0;
+      if (rrem >= 16) {
+        if (SRC_F32) {
+          const float* s = (const float*)base + row0 + r0;
+#pragma unroll
+          for (int v = 0; v < 4; ++v) {
+            floatx4 x = *(const floatx4*)(s + v * 4);
+#pragma unroll
+            for (int j = 0; j < 4; ++j) lds[(r0 + v * 4 + j) * LDSK + k] = f32_to_bf16(x[j]);
+          }
+        } else {
+          const bf16raw* s = (const bf16raw*)base + row0 + r0;
+#pragma unroll
+          for (int v = 0; v < 2; ++v) {
+            shortx8 x = *(const shortx8*)(s + v * 8);
+#pragma unroll
+            for (int j = 0; j < 8; ++j) lds[(r0 + v * 8 + j) * LDSK + k] = (bf16raw)x[j];
+          }
+        }
+      } else {
+#pragma unroll
+        for (int j = 0; j < 16; ++j) {
+          int r = r0 + j;
+          bf16raw v = 0;
+          if (r < rrem)
+            v = SRC_F32 ? f32_to_bf16(((const float*)base)[row0 + r])
+                        : ((const bf16raw*)base)[row0 + r];
+          lds[r * LDSK + k] = v;
+        }
+      }
+    } else {
+#pragma unroll
+      for (int j = 0; j < 16; ++j) lds[(r0 + j) * LDSK + k] = 0;
+    }
+  } else {
+    // generic scalar path
+    int idx0 = t * 16;
+#pragma unroll
+    for (int j = 0; j < 16; ++j) {
+      int idx = idx0 + j;
+      int r = idx / BK, k = idx % BK;
+      int gr = row0 + r, gk = kt + k;
+      bf16raw v = 0;
+      if (gr < rmax && gk < kmax) {
+        const char* p = (const char*)src + ((int64_t)gr * srow + (int64_t)gk * skol) * (SRC_F32 ? 4 : 2);
+        v = SRC_F32 ? f32_to_bf16(*(const float*)p) : *(const bf16raw*)p;
+      }
+      lds[r * LDSK + k] = v;
+    }
+  }
+}
+
+template <bool B_IS_F32, int EPI, bool SPLITK>
+__global__ __launch_bounds__(256) void gemm_kernel(const void* __restrict__ Ap,
+                                                   const void* __restrict__ Bp,
+                                                   float* __restrict__ Cf,
+                                                   bf16raw* __restrict__ Cb,
+                                                   const float* __restrict__ bias, int M, int N,
+                                                   int K, int64_t sam, int64_t sak, int64_t sbk,
+                                                   int64_t sbn, int k_per_split) {
+  __shared__ bf16raw As[BM * LDSK];
+  __shared__ bf16raw Bs[BN * LDSK];
+
+  const int m0 = blockIdx.x * BM;
+  const int n0 = blockIdx.y * BN;
+
+  int k_begin = 0, k_end = K;
+  if (SPLITK) {
+    k_begin = blockIdx.z * k_per_split;
+    k_end = min(K, k_begin + k_per_split);
+  }
+
+  const int wid = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  const int wr = wid >> 1, wc = wid & 1;
+  const int l15 = lane & 15, kg = lane >> 4;
+
+  floatx4 acc[4][4] = {};
+
+  for (int kt = k_begin; kt < k_end; kt += BK) {
+    stage_tile<false>(Ap, As, m0, M, kt, k_end, sam, sak);
+    // B staged TRANSPOSED: LDS row = n, LDS col = k -> srow := sbn, skol := sbk
+    stage_tile<B_IS_F32>(Bp, Bs, n0, N, kt, k_end, sbn, sbk);
+    __syncthreads();
+
+    frag_t a[4], b[4];
+#pragma unroll
+    for (int mi = 0; mi < 4; ++mi)
+      a[mi] = *(const frag_t*)&As[(wr * 64 + mi * 16 + l15) * LDSK + kg * 8];
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni)
+      b[ni] = *(const frag_t*)&Bs[(wc * 64 + ni * 16 + l15) * LDSK + kg * 8];
+
+#pragma unroll
+    for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+        acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[mi], b[ni], acc[mi][ni], 0, 0, 0);
+
+    __syncthreads();
+  }
+
+  // epilogue: C/D fragment layout col = lane&15, row = 4*(lane>>4) + reg
+  const int m_base = m0 + wr * 64;
+  const int n_base = n0 + wc * 64;
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi) {
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+      int n = n_base + ni * 16 + l15;
+      if (n >= N) continue;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int m = m_base + mi * 16 + kg * 4 + r;
+        if (m >= M) continue;
+        float val = acc[mi][ni][r];
+        int64_t off = (int64_t)m * N + n;
+        if (EPI == EPI_F32) {
+          if (SPLITK)
+            atomicAdd(Cf + off, val);
+          else
+            Cf[off] = val;
+        } else {
+          if (EPI >= EPI_BIAS) val += bias[n];
+          if (EPI == EPI_BIAS_RELU) val = fmaxf(val, 0.f);
+          Cb[off] = f32_to_bf16(val);
+        }
+      }
+    }
+  }
+}
+
+extern "C" hipError_t launch_gemm_bf16(const void* A, const void* B, int b_is_f32, float* Cf,
+                                       bf16raw* Cb, const float* bias, int M, int N, int K,
+                                       int64_t sam, int64_t sak, int64_t sbk, int64_t sbn, int epi,
+                                       int splitk, hipStream_t stream) {
+  dim3 block(256);
+  int kps = 0;
+  if (splitk < 1) splitk = 1;
+  if (splitk > 1) {
+    kps = (int)ceil_div_i64(ceil_div_i64(K, splitk), BK) * BK;
+    splitk = (int)ceil_div_i64(K, kps);
+  }
+  dim3 grid((unsigned)ceil_div_i64(M, BM), (unsigned)ceil_div_i64(N, BN), (unsigned)splitk);
+
+#define DISPATCH(BF32, EPIC, SPK)                                                          \
+  gemm_kernel<BF32, EPIC, SPK><<<grid, block, 0, stream>>>(A, B, Cf, Cb, bias, M, N, K, sam, \
+                                                           sak, sbk, sbn, kps)
+
+  if (b_is_f32) {
+    if (splitk > 1 && epi == EPI_F32) DISPATCH(true, EPI_F32, true);
+    else if (epi == EPI_F32) DISPATCH(true, EPI_F32, false);
+    else if (epi == EPI_BF16) DISPATCH(true, EPI_BF16, false);
+    else if (epi == EPI_BIAS) DISPATCH(true, EPI_BIAS, false);
+    else if (epi == EPI_BIAS_RELU) DISPATCH(true, EPI_BIAS_RELU, false);
+    else return hipErrorInvalidValue;
+  } else {
+    if (splitk > 1 && epi == EPI_F32) DISPATCH(false, EPI_F32, true);
+    else if (epi == EPI_F32) DISPATCH(false, EPI_F32, false);
+    else if (epi == EPI_BF16) DISPATCH(false, EPI_BF16, false);
+    else if (epi == EPI_BIAS) DISPATCH(false, EPI_BIAS, false);
+    else if (epi == EPI_BIAS_RELU) DISPATCH(false, EPI_BIAS_RELU, false);
+    else return hipErrorInvalidValue;
+  }
+#undef DISPATCH
+  HIP_CHECK_LAUNCH();
+  return hipSuccess;
+}

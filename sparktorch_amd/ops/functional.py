@@ -1,0 +1,119 @@
+"""Autograd wiring for the hand-written gfx950 kernels.
+
+Forward math (bf16 activations, fp32 master weights, fp32 MFMA accumulate):
+  y = relu(x @ W^T + b)        — one fused kernel (gemm.hip EPI_BIAS_RELU)
+backward:
+  dz = dy * (y > 0)            — relu_bwd (elementwise.hip)
+  dx = dz @ W                  — MFMA
+  dW = dz^T @ x                — MFMA, split-K over batch (atomic fp32)
+  db = colsum(dz)              — bias_grad
+
+The CPU fallback keeps identical semantics so the same model code runs in the
+no-GPU sandbox; on a GPU the extension is REQUIRED (ops.ext() raises).
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+import torch.nn.functional as F
+
+from sparktorch_amd import ops
+
+
+def _choose_splitk(batch: int, n_out: int, k_in: int) -> int:
+    """Fill 256 CUs: base grid is ceil(N/128)*ceil(K/128) blocks; split the
+    batch reduction until ~512 blocks, keeping >=512 rows per slice."""
+    base = -(-n_out // 128) * (-(-k_in) // 128)
+    want = max(1, 512 // max(1, base))
+    max_by_rows = max(1, batch // 512)
+    return int(min(want, max_by_rows, 64))
+
+
+class _LinearFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x: torch.Tensor, w: torch.Tensor, b: Optional[torch.Tensor], relu: bool):
+        y = ops.ext().linear_fwd(x, w, b, relu)
+        ctx.save_for_backward(x, w, y)
+        ctx.relu = relu
+        ctx.has_bias = b is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, dy: torch.Tensor):
+        x, w, y = ctx.saved_tensors
+        ext = ops.ext()
+        dy = dy.contiguous()
+        if dy.dtype != torch.bfloat16:
+            dy = dy.to(torch.bfloat16)
+        dz = ext.relu_bwd(dy, y) if ctx.relu else dy
+        dx = ext.linear_dgrad(dz, w) if ctx.needs_input_grad[0] else None
+        dw = None
+        if ctx.needs_input_grad[1]:
+            sk = _choose_splitk(dz.shape[0], w.shape[0], w.shape[1])
+            dw = ext.linear_wgrad(dz, x, sk)
+        db = ext.bias_grad(dz) if (ctx.has_bias and ctx.needs_input_grad[2]) else None
+        return dx, dw, db, None
+
+
+def hip_linear(
+    x: torch.Tensor,
+    weight: torch.Tensor,
+    bias: Optional[torch.Tensor] = None,
+    relu: bool = False,
+) -> torch.Tensor:
+    """Fused linear (+bias)(+relu).  GPU: hand-written MFMA kernels; CPU:
+    identical-semantics torch ops (bf16-sim not applied on CPU)."""
+    if x.is_cuda:
+        if x.dtype != torch.bfloat16:
+            x = x.to(torch.bfloat16)
+        if not x.is_contiguous():
+            x = x.contiguous()
+        y = _LinearFn.apply(x, weight, bias, relu)
+        return y
+    y = F.linear(x, weight, bias)
+    return F.relu(y) if relu else y
+
+
+class _CEFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, logits: torch.Tensor, target: torch.Tensor):
+        loss, dlogits = ops.ext().ce_fused(logits, target)
+        ctx.save_for_backward(dlogits)
+        return loss
+
+    @staticmethod
+    def backward(ctx, gout):
+        (dlogits,) = ctx.saved_tensors
+        return dlogits * gout.to(dlogits.dtype), None
+
+
+def hip_cross_entropy(logits: torch.Tensor, target: torch.Tensor) -> torch.Tensor:
+    """Fused CE fwd+bwd (one kernel computes loss and dlogits)."""
+    if logits.is_cuda:
+        if logits.dtype != torch.bfloat16:
+            logits = logits.to(torch.bfloat16)
+        return _CEFn.apply(logits.contiguous(), target.flatten().long().contiguous())
+    return F.cross_entropy(logits.float(), target.flatten().long())
+
+
+class _MSEFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, pred: torch.Tensor, target: torch.Tensor):
+        loss, dpred = ops.ext().mse_fused(pred, target)
+        ctx.save_for_backward(dpred)
+        return loss
+
+    @staticmethod
+    def backward(ctx, gout):
+        (dpred,) = ctx.saved_tensors
+        return dpred * gout.to(dpred.dtype), None
+
+
+def hip_mse(pred: torch.Tensor, target: torch.Tensor) -> torch.Tensor:
+    if pred.is_cuda:
+        if pred.dtype != torch.bfloat16:
+            pred = pred.to(torch.bfloat16)
+        return _MSEFn.apply(pred.contiguous(), target.to(torch.bfloat16).contiguous())
+    return F.mse_loss(pred.float(), target.float())

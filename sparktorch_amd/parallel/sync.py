@@ -49,10 +49,15 @@ from sparktorch_amd.utils.serialize import (
 def compute_loss(criterion, pred: torch.Tensor, y: torch.Tensor) -> torch.Tensor:
     """Criterion with the reference's long-label retry (distributed.py:152-157):
     classification losses want int64 class targets."""
-    if isinstance(criterion, (torch.nn.CrossEntropyLoss, torch.nn.NLLLoss)):
+    cname = type(criterion).__name__
+    if cname in ("HipCrossEntropy", "HipMSE"):
+        return criterion(pred, y)  # fused losses handle their own casts
+    if cname in ("CrossEntropyLoss", "NLLLoss") or isinstance(
+        criterion, (torch.nn.CrossEntropyLoss, torch.nn.NLLLoss)
+    ):
         return criterion(pred, y.flatten().long())
     try:
-        return criterion(pred, y.float())
+        return criterion(pred, y.float() if y.dtype != pred.dtype else y)
     except RuntimeError:
         return criterion(pred, y.flatten().long())
 
@@ -77,6 +82,20 @@ class SyncTrainer:
         self.device = device
         self.model = model.to(device)
         self.criterion = criterion
+        self.native = device.startswith("cuda")
+        if self.native:
+            # GPU: the hand-written MFMA/fused-kernel path is mandatory —
+            # fail loudly rather than silently running eager torch.
+            from sparktorch_amd import ops as _ops
+
+            _ops.ext()
+            from sparktorch_amd.ops.modules import (
+                convert_criterion_for_mi355x,
+                convert_model_for_mi355x,
+            )
+
+            self.model = convert_model_for_mi355x(self.model)
+            self.criterion = convert_criterion_for_mi355x(self.criterion)
         self.world_size = (
             world_size
             if world_size is not None
@@ -228,6 +247,12 @@ def handle_model(
         x_train = data.x_train.to(dev)
         # autoencoder mode: no labels -> y = x (reference distributed.py:136)
         y_train = data.y_train.to(dev) if data.y_train is not None else x_train
+        if dev.startswith("cuda"):
+            # bf16 activations once, outside the hot loop
+            ae = y_train is x_train
+            x_train = x_train.to(torch.bfloat16)
+            if ae:
+                y_train = x_train
         x_val = data.x_val.to(dev) if data.x_val is not None else None
         y_val = data.y_val.to(dev) if data.y_val is not None else (x_val if x_val is not None else None)
 
