@@ -117,7 +117,7 @@ __device__ __forceinline__ void stage_tile(const void* __restrict__ src, bf16raw
         for (int j = 0; j < 16; ++j) {
           int r = r0 + j;
           bf16raw v = 0;
-          if (r < rrem)
+          if (j < rrem)  // row0 + r < rmax
             v = SRC_F32 ? f32_to_bf16(((const float*)base)[row0 + r])
                         : ((const bf16raw*)base)[row0 + r];
           lds[r * LDSK + k] = v;
