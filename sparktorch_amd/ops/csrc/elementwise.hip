@@ -174,22 +174,53 @@ extern "C" hipError_t launch_relu_bwd(const bf16raw* dy, const bf16raw* y, bf16r
 // contention).
 // ---------------------------------------------------------------------------
 
+// Block = (64, 4): lane x owns 8 consecutive columns (shortx8 vector load),
+// consecutive x lanes read consecutive 16 B => fully coalesced row reads; the
+// 4 y-lanes stripe rows.  Per-thread fp32 accumulators, LDS-reduce the 4 row
+// stripes, one atomicAdd per column per block (guide §6 G12/G13).
 __global__ void bias_grad_kernel(const bf16raw* __restrict__ dz, float* __restrict__ db, int M,
                                  int N, int rows_per_chunk) {
-  int n = blockIdx.x * blockDim.x + threadIdx.x;
-  if (n >= N) return;
-  int m0 = blockIdx.y * rows_per_chunk;
-  int m1 = min(M, m0 + rows_per_chunk);
-  float acc = 0.f;
-  for (int m = m0; m < m1; ++m) acc += bf16_to_f32(dz[(int64_t)m * N + n]);
-  atomicAdd(db + n, acc);
+  const int n0 = (blockIdx.x * 64 + threadIdx.x) * 8;
+  const int m0 = blockIdx.y * rows_per_chunk;
+  const int m1 = min(M, m0 + rows_per_chunk);
+  float acc[8] = {};
+  if (n0 < N) {
+    const bool full = (n0 + 8 <= N);
+    for (int m = m0 + threadIdx.y; m < m1; m += 4) {
+      const bf16raw* row = dz + (int64_t)m * N + n0;
+      if (full) {
+        shortx8 v = *(const shortx8*)row;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) acc[j] += bf16_to_f32((bf16raw)v[j]);
+      } else {
+        for (int j = 0; j < N - n0; ++j) acc[j] += bf16_to_f32(row[j]);
+      }
+    }
+  }
+  // reduce the 4 y-stripes through LDS
+  __shared__ float red[4][64][8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) red[threadIdx.y][threadIdx.x][j] = acc[j];
+  __syncthreads();
+  if (threadIdx.y == 0 && n0 < N) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      if (n0 + j < N) {
+        float s = red[0][threadIdx.x][j] + red[1][threadIdx.x][j] + red[2][threadIdx.x][j] +
+                  red[3][threadIdx.x][j];
+        atomicAdd(db + n0 + j, s);
+      }
+    }
+  }
 }
 
 extern "C" hipError_t launch_bias_grad(const bf16raw* dz, float* db, int M, int N,
                                        hipStream_t stream) {
-  int block = 64;
-  int rows_per_chunk = 1024;
-  dim3 grid((N + block - 1) / block, (unsigned)ceil_div_i64(M, rows_per_chunk));
+  dim3 block(64, 4);
+  // <=128 row chunks (bounds atomic traffic), >=256 rows each
+  int rows_per_chunk = (int)ceil_div_i64(M, 128);
+  if (rows_per_chunk < 256) rows_per_chunk = 256;
+  dim3 grid((unsigned)ceil_div_i64(N, 64 * 8), (unsigned)ceil_div_i64(M, rows_per_chunk));
   bias_grad_kernel<<<grid, block, 0, stream>>>(dz, db, M, N, rows_per_chunk);
   HIP_CHECK_LAUNCH();
   return hipSuccess;
