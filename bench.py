@@ -34,7 +34,7 @@ def main() -> int:
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=30)
     ap.add_argument("--warmup", type=int, default=10)
-    ap.add_argument("--batch", type=int, default=32768, help="per-GPU batch (weak scaling)")
+    ap.add_argument("--batch", type=int, default=65536, help="per-GPU batch (weak scaling)")
     ap.add_argument("--device", type=str, default=None)
     ap.add_argument("--hipgraph", action="store_true", help="capture the train step in a HIP graph")
     args = ap.parse_args()

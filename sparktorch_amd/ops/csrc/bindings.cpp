@@ -25,7 +25,8 @@ hipError_t launch_ce_fused(const bf16raw*, const int64_t*, float*, bf16raw*, int
 hipError_t launch_mse_fused(const bf16raw*, const bf16raw*, float*, bf16raw*, int64_t,
                             hipStream_t);
 hipError_t launch_gemm_bf16(const void*, const void*, int, float*, bf16raw*, const float*, int,
-                            int, int, int64_t, int64_t, int64_t, int64_t, int, int, hipStream_t);
+                            int, int, int64_t, int64_t, int64_t, int64_t, int, int, float*, int,
+                            hipStream_t);
 }
 
 #define CHECK_HIP(err)                                                              \
@@ -142,7 +143,7 @@ at::Tensor linear_fwd(at::Tensor x, at::Tensor w, c10::optional<at::Tensor> bias
   }
   CHECK_HIP(launch_gemm_bf16(x.data_ptr(), w.data_ptr(), w.scalar_type() == at::kFloat ? 1 : 0,
                              nullptr, (bf16raw*)y.data_ptr(), bptr, (int)M, (int)N, (int)K,
-                             /*sam*/ K, /*sak*/ 1, /*sbk*/ 1, /*sbn*/ K, epi, 1, cur_stream()));
+                             /*sam*/ K, /*sak*/ 1, /*sbk*/ 1, /*sbn*/ K, epi, 1, nullptr, -1, cur_stream()));
   return y;
 }
 
@@ -154,7 +155,7 @@ at::Tensor linear_dgrad(at::Tensor dz, at::Tensor w) {
   auto dx = at::empty({M, K}, dz.options());
   CHECK_HIP(launch_gemm_bf16(dz.data_ptr(), w.data_ptr(), w.scalar_type() == at::kFloat ? 1 : 0,
                              nullptr, (bf16raw*)dx.data_ptr(), nullptr, (int)M, (int)K, (int)N,
-                             /*sam*/ N, /*sak*/ 1, /*sbk*/ K, /*sbn*/ 1, 1, 1, cur_stream()));
+                             /*sam*/ N, /*sak*/ 1, /*sbk*/ K, /*sbn*/ 1, 1, 1, nullptr, -1, cur_stream()));
   return dx;
 }
 
@@ -168,8 +169,47 @@ at::Tensor linear_wgrad(at::Tensor dz, at::Tensor x, int64_t splitk) {
   CHECK_HIP(launch_gemm_bf16(dz.data_ptr(), x.data_ptr(), /*b_is_f32*/ 0, dw.data_ptr<float>(),
                              nullptr, nullptr, (int)N, (int)K, (int)B,
                              /*sam*/ 1, /*sak*/ N, /*sbk*/ K, /*sbn*/ 1, 0, (int)splitk,
-                             cur_stream()));
+                             nullptr, -1, cur_stream()));
   return dw;
+}
+
+// Direct-accumulate variants: write into a pre-zeroed (or accumulating) grad
+// view — no fresh allocation, no autograd add pass.  Always the atomic
+// (accumulate) epilogue so gradient accumulation semantics match autograd.
+void linear_wgrad_into(at::Tensor dz, at::Tensor x, at::Tensor dw, int64_t splitk) {
+  check_gpu_contig(dz, at::kBFloat16, "dz");
+  check_gpu_contig(x, at::kBFloat16, "x");
+  check_gpu_contig(dw, at::kFloat, "dw");
+  int64_t B = dz.size(0), N = dz.size(1), K = x.size(1);
+  TORCH_CHECK(x.size(0) == B && dw.size(0) == N && dw.size(1) == K, "shape mismatch");
+  if (splitk < 1) splitk = 1;
+  CHECK_HIP(launch_gemm_bf16(dz.data_ptr(), x.data_ptr(), 0, dw.data_ptr<float>(), nullptr,
+                             nullptr, (int)N, (int)K, (int)B, 1, N, K, 1, 0, -(int)splitk,
+                             nullptr, -1, cur_stream()));
+}
+
+// dW and db in ONE MFMA launch: dW_ext = dz^T @ [x | 1]; the virtual ones
+// column's output lands in db via the epilogue (gemm.hip ones_row).
+void linear_wgrad_bias_into(at::Tensor dz, at::Tensor x, at::Tensor dw, at::Tensor db,
+                            int64_t splitk) {
+  check_gpu_contig(dz, at::kBFloat16, "dz");
+  check_gpu_contig(x, at::kBFloat16, "x");
+  check_gpu_contig(dw, at::kFloat, "dw");
+  check_gpu_contig(db, at::kFloat, "db");
+  int64_t B = dz.size(0), N = dz.size(1), K = x.size(1);
+  TORCH_CHECK(x.size(0) == B && dw.size(0) == N && dw.size(1) == K && db.numel() == N);
+  if (splitk < 1) splitk = 1;
+  CHECK_HIP(launch_gemm_bf16(dz.data_ptr(), x.data_ptr(), 0, dw.data_ptr<float>(), nullptr,
+                             nullptr, (int)N, (int)(K + 1), (int)B, 1, N, K, 1, 0, -(int)splitk,
+                             db.data_ptr<float>(), (int)K, cur_stream()));
+}
+
+void bias_grad_into(at::Tensor dz, at::Tensor db) {
+  check_gpu_contig(dz, at::kBFloat16, "dz");
+  check_gpu_contig(db, at::kFloat, "db");
+  TORCH_CHECK(dz.dim() == 2 && db.numel() == dz.size(1));
+  CHECK_HIP(launch_bias_grad((const bf16raw*)dz.data_ptr(), db.data_ptr<float>(),
+                             (int)dz.size(0), (int)dz.size(1), cur_stream()));
 }
 
 // generic C = A @ B with optional transposes, for tests/other models
@@ -188,7 +228,7 @@ at::Tensor matmul_bf16(at::Tensor a, at::Tensor b, bool trans_a, bool trans_b) {
   auto c = at::empty({M, N}, a.options().dtype(at::kFloat));
   CHECK_HIP(launch_gemm_bf16(a.data_ptr(), b.data_ptr(), b.scalar_type() == at::kFloat ? 1 : 0,
                              c.data_ptr<float>(), nullptr, nullptr, (int)M, (int)N, (int)Ka, sam,
-                             sak, sbk, sbn, 0, 1, cur_stream()));
+                             sak, sbk, sbn, 0, 1, nullptr, -1, cur_stream()));
   return c;
 }
 
@@ -204,5 +244,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("linear_fwd", &linear_fwd, "Y = X W^T (+bias)(+relu), MFMA");
   m.def("linear_dgrad", &linear_dgrad, "dX = dZ W, MFMA");
   m.def("linear_wgrad", &linear_wgrad, "dW = dZ^T X, MFMA split-K");
+  m.def("linear_wgrad_into", &linear_wgrad_into, "dW accumulated into grad view");
+  m.def("linear_wgrad_bias_into", &linear_wgrad_bias_into, "dW + db in one MFMA launch");
+  m.def("bias_grad_into", &bias_grad_into, "db accumulated into grad view");
   m.def("matmul_bf16", &matmul_bf16, "generic bf16 MFMA GEMM");
 }

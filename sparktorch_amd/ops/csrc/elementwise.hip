@@ -217,9 +217,10 @@ __global__ void bias_grad_kernel(const bf16raw* __restrict__ dz, float* __restri
 extern "C" hipError_t launch_bias_grad(const bf16raw* dz, float* db, int M, int N,
                                        hipStream_t stream) {
   dim3 block(64, 4);
-  // <=128 row chunks (bounds atomic traffic), >=256 rows each
-  int rows_per_chunk = (int)ceil_div_i64(M, 128);
-  if (rows_per_chunk < 256) rows_per_chunk = 256;
+  // ~512 row chunks: 2048 waves keeps 256 CUs latency-hidden; per-column
+  // atomic chains stay ~512 deep (few us, parallel across columns)
+  int rows_per_chunk = (int)ceil_div_i64(M, 512);
+  if (rows_per_chunk < 64) rows_per_chunk = 64;
   dim3 grid((unsigned)ceil_div_i64(N, 64 * 8), (unsigned)ceil_div_i64(M, rows_per_chunk));
   bias_grad_kernel<<<grid, block, 0, stream>>>(dz, db, M, N, rows_per_chunk);
   HIP_CHECK_LAUNCH();

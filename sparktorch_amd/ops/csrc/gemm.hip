@@ -44,18 +44,27 @@ typedef shortx8 frag_t;  // 8 bf16 (4 VGPRs)
 //   else if srow==1 (row-contiguous): thread covers 16 consecutive rows at
 //     fixed k (vectorized along rows, strided LDS column writes).
 //   else: scalar element loop.
+#define BF16_ONE ((bf16raw)0x3F80)
+
+// ones_row >= 0 marks a VIRTUAL row whose every element is 1.0 (the bias
+// column of dW_ext = dz^T @ [x | 1]); memory is only touched for rows below
+// it.  -1 = no virtual row.
 template <bool SRC_F32>
 __device__ __forceinline__ void stage_tile(const void* __restrict__ src, bf16raw* __restrict__ lds,
                                            int row0, int rmax, int kt, int kmax, int64_t srow,
-                                           int64_t skol) {
+                                           int64_t skol, int ones_row = -1) {
   const int t = threadIdx.x;
+  const int mem_rows = ones_row >= 0 ? ones_row : rmax;
   if (skol == 1) {
     // 2 threads per row; 16 consecutive k each
     int r = t >> 1;
     int k0 = (t & 1) * 16;
     int gr = row0 + r;
     bf16raw* dst = lds + r * LDSK + k0;
-    if (gr < rmax) {
+    if (ones_row >= 0 && gr == ones_row) {
+#pragma unroll
+      for (int j = 0; j < 16; ++j) dst[j] = (kt + k0 + j < kmax) ? BF16_ONE : (bf16raw)0;
+    } else if (gr < mem_rows) {
       const char* base = (const char*)src + (int64_t)gr * srow * (SRC_F32 ? 4 : 2);
       int krem = kmax - kt - k0;  // how many of our 16 k are in range
       if (krem >= 16) {
@@ -94,7 +103,8 @@ __device__ __forceinline__ void stage_tile(const void* __restrict__ src, bf16raw
     if (gk < kmax) {
       const char* base = (const char*)src + (int64_t)gk * skol * (SRC_F32 ? 4 : 2);
       int rrem = rmax - row0 - r0;
-      if (rrem >= 16) {
+      int mem_rem = mem_rows - row0 - r0;
+      if (mem_rem >= 16) {
         if (SRC_F32) {
           const float* s = (const float*)base + row0 + r0;
 #pragma unroll
@@ -117,7 +127,9 @@ __device__ __forceinline__ void stage_tile(const void* __restrict__ src, bf16raw
         for (int j = 0; j < 16; ++j) {
           int r = r0 + j;
           bf16raw v = 0;
-          if (j < rrem)  // row0 + r < rmax
+          if (ones_row >= 0 && row0 + r == ones_row)
+            v = BF16_ONE;
+          else if (j < mem_rem)  // row0 + r < mem_rows
             v = SRC_F32 ? f32_to_bf16(((const float*)base)[row0 + r])
                         : ((const bf16raw*)base)[row0 + r];
           lds[r * LDSK + k] = v;
@@ -136,7 +148,9 @@ __device__ __forceinline__ void stage_tile(const void* __restrict__ src, bf16raw
       int r = idx / BK, k = idx % BK;
       int gr = row0 + r, gk = kt + k;
       bf16raw v = 0;
-      if (gr < rmax && gk < kmax) {
+      if (ones_row >= 0 && gr == ones_row && gk < kmax) {
+        v = BF16_ONE;
+      } else if (gr < mem_rows && gk < kmax) {
         const char* p = (const char*)src + ((int64_t)gr * srow + (int64_t)gk * skol) * (SRC_F32 ? 4 : 2);
         v = SRC_F32 ? f32_to_bf16(*(const float*)p) : *(const bf16raw*)p;
       }
@@ -152,12 +166,22 @@ __global__ __launch_bounds__(256) void gemm_kernel(const void* __restrict__ Ap,
                                                    bf16raw* __restrict__ Cb,
                                                    const float* __restrict__ bias, int M, int N,
                                                    int K, int64_t sam, int64_t sak, int64_t sbk,
-                                                   int64_t sbn, int k_per_split) {
+                                                   int64_t sbn, int k_per_split,
+                                                   float* __restrict__ Db, int ones_row) {
   __shared__ bf16raw As[BM * LDSK];
   __shared__ bf16raw Bs[BN * LDSK];
 
-  const int m0 = blockIdx.x * BM;
-  const int n0 = blockIdx.y * BN;
+  // bijective XCD-aware swizzle of the flattened block id (guide §5.5 T1):
+  // consecutive output tiles land on one XCD so shared operand rows stay in
+  // that XCD's L2.
+  const int gx = gridDim.x;
+  int nwg = gx * gridDim.y;
+  int orig = blockIdx.y * gx + blockIdx.x;
+  int q = nwg >> 3, rr = nwg & 7;
+  int wg = ((orig & 7) < rr ? (orig & 7) * (q + 1) : rr * (q + 1) + ((orig & 7) - rr) * q) +
+           (orig >> 3);
+  const int m0 = (wg % gx) * BM;
+  const int n0 = (wg / gx) * BN;
 
   int k_begin = 0, k_end = K;
   if (SPLITK) {
@@ -175,7 +199,7 @@ __global__ __launch_bounds__(256) void gemm_kernel(const void* __restrict__ Ap,
   for (int kt = k_begin; kt < k_end; kt += BK) {
     stage_tile<false>(Ap, As, m0, M, kt, k_end, sam, sak);
     // B staged TRANSPOSED: LDS row = n, LDS col = k -> srow := sbn, skol := sbk
-    stage_tile<B_IS_F32>(Bp, Bs, n0, N, kt, k_end, sbn, sbk);
+    stage_tile<B_IS_F32>(Bp, Bs, n0, N, kt, k_end, sbn, sbk, ones_row);
     __syncthreads();
 
     frag_t a[4], b[4];
@@ -209,13 +233,16 @@ __global__ __launch_bounds__(256) void gemm_kernel(const void* __restrict__ Ap,
         int m = m_base + mi * 16 + kg * 4 + r;
         if (m >= M) continue;
         float val = acc[mi][ni][r];
-        int64_t off = (int64_t)m * N + n;
         if (EPI == EPI_F32) {
-          if (SPLITK)
-            atomicAdd(Cf + off, val);
-          else
-            Cf[off] = val;
+          if (Db != nullptr && n == ones_row) {
+            atomicAdd(Db + m, val);  // bias column of dW_ext
+          } else if (SPLITK) {
+            atomicAdd(Cf + (int64_t)m * (N - (ones_row >= 0 ? 1 : 0)) + n, val);
+          } else {
+            Cf[(int64_t)m * (N - (ones_row >= 0 ? 1 : 0)) + n] = val;
+          }
         } else {
+          int64_t off = (int64_t)m * N + n;
           if (EPI >= EPI_BIAS) val += bias[n];
           if (EPI == EPI_BIAS_RELU) val = fmaxf(val, 0.f);
           Cb[off] = f32_to_bf16(val);
@@ -228,29 +255,38 @@ __global__ __launch_bounds__(256) void gemm_kernel(const void* __restrict__ Ap,
 extern "C" hipError_t launch_gemm_bf16(const void* A, const void* B, int b_is_f32, float* Cf,
                                        bf16raw* Cb, const float* bias, int M, int N, int K,
                                        int64_t sam, int64_t sak, int64_t sbk, int64_t sbn, int epi,
-                                       int splitk, hipStream_t stream) {
+                                       int splitk, float* Db, int ones_row, hipStream_t stream) {
   dim3 block(256);
+  // splitk < 0 => |splitk| slices AND force the atomic (accumulate) epilogue
+  // even if the recomputed slice count collapses to 1.
+  int force_atomic = 0;
+  if (splitk < 0) {
+    force_atomic = 1;
+    splitk = -splitk;
+  }
   int kps = 0;
   if (splitk < 1) splitk = 1;
   if (splitk > 1) {
     kps = (int)ceil_div_i64(ceil_div_i64(K, splitk), BK) * BK;
     splitk = (int)ceil_div_i64(K, kps);
   }
+  if (splitk == 1) kps = K;
+  const bool atomic = (splitk > 1 || force_atomic);
   dim3 grid((unsigned)ceil_div_i64(M, BM), (unsigned)ceil_div_i64(N, BN), (unsigned)splitk);
 
 #define DISPATCH(BF32, EPIC, SPK)                                                          \
   gemm_kernel<BF32, EPIC, SPK><<<grid, block, 0, stream>>>(A, B, Cf, Cb, bias, M, N, K, sam, \
-                                                           sak, sbk, sbn, kps)
+                                                           sak, sbk, sbn, kps, Db, ones_row)
 
   if (b_is_f32) {
-    if (splitk > 1 && epi == EPI_F32) DISPATCH(true, EPI_F32, true);
+    if (atomic && epi == EPI_F32) DISPATCH(true, EPI_F32, true);
     else if (epi == EPI_F32) DISPATCH(true, EPI_F32, false);
     else if (epi == EPI_BF16) DISPATCH(true, EPI_BF16, false);
     else if (epi == EPI_BIAS) DISPATCH(true, EPI_BIAS, false);
     else if (epi == EPI_BIAS_RELU) DISPATCH(true, EPI_BIAS_RELU, false);
     else return hipErrorInvalidValue;
   } else {
-    if (splitk > 1 && epi == EPI_F32) DISPATCH(false, EPI_F32, true);
+    if (atomic && epi == EPI_F32) DISPATCH(false, EPI_F32, true);
     else if (epi == EPI_F32) DISPATCH(false, EPI_F32, false);
     else if (epi == EPI_BF16) DISPATCH(false, EPI_BF16, false);
     else if (epi == EPI_BIAS) DISPATCH(false, EPI_BIAS, false);

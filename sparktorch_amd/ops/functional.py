@@ -38,6 +38,7 @@ class _LinearFn(torch.autograd.Function):
         ctx.save_for_backward(x, w, y)
         ctx.relu = relu
         ctx.has_bias = b is not None
+        ctx.bias_ref = b
         return y
 
     @staticmethod
@@ -49,11 +50,37 @@ class _LinearFn(torch.autograd.Function):
             dy = dy.to(torch.bfloat16)
         dz = ext.relu_bwd(dy, y) if ctx.relu else dy
         dx = ext.linear_dgrad(dz, w) if ctx.needs_input_grad[0] else None
-        dw = None
-        if ctx.needs_input_grad[1]:
-            sk = _choose_splitk(dz.shape[0], w.shape[0], w.shape[1])
-            dw = ext.linear_wgrad(dz, x, sk)
-        db = ext.bias_grad(dz) if (ctx.has_bias and ctx.needs_input_grad[2]) else None
+
+        # Direct-grad path: when FlatBuckets installed this param, its .grad
+        # is a pre-zeroed flat-bucket view — accumulate straight into it (no
+        # fresh alloc, no autograd add pass) and notify the bucket so the
+        # overlapped all-reduce launches; autograd sees None.
+        dw = db = None
+        b = ctx.bias_ref
+        w_notify = getattr(w, "_bucket_notify", None)
+        b_notify = getattr(b, "_bucket_notify", None) if b is not None else None
+        sk = _choose_splitk(dz.shape[0], w.shape[0], w.shape[1])
+        w_direct = w_notify is not None and w.grad is not None
+        b_direct = b_notify is not None and b is not None and b.grad is not None
+
+        if ctx.needs_input_grad[1] and w_direct and ctx.has_bias and b_direct:
+            # dW and db in ONE MFMA launch (virtual ones column)
+            ext.linear_wgrad_bias_into(dz, x, w.grad, b.grad, sk)
+            w_notify()
+            b_notify()
+        else:
+            if ctx.needs_input_grad[1]:
+                if w_direct:
+                    ext.linear_wgrad_into(dz, x, w.grad, sk)
+                    w_notify()
+                else:
+                    dw = ext.linear_wgrad(dz, x, sk)
+            if ctx.has_bias and ctx.needs_input_grad[2]:
+                if b_direct:
+                    ext.bias_grad_into(dz, b.grad)
+                    b_notify()
+                else:
+                    db = ext.bias_grad(dz)
         return dx, dw, db, None
 
 

@@ -123,7 +123,12 @@ class FlatBuckets:
             return hook
 
         for p, b in param_to_bucket.items():
-            self._hooks.append(p.register_post_accumulate_grad_hook(make_hook(b)))
+            hook = make_hook(b)
+            self._hooks.append(p.register_post_accumulate_grad_hook(hook))
+            # Direct-grad path: native ops accumulate straight into p.grad
+            # (the flat view), return None to autograd (so the torch hook
+            # never fires), and call this notifier instead.
+            p._bucket_notify = (lambda h=hook, q=p: h(q))
 
     def _launch(self, bucket: Bucket) -> None:
         if self._comm_enabled and self.world_size > 1 and dist.is_initialized():
@@ -159,3 +164,7 @@ class FlatBuckets:
         for h in self._hooks:
             h.remove()
         self._hooks = []
+        for b in self.buckets:
+            for p in b.params:
+                if hasattr(p, "_bucket_notify"):
+                    del p._bucket_notify

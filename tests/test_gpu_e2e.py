@@ -1,0 +1,83 @@
+"""GPU end-to-end tests: estimator fit/transform on MI355X, multi-rank
+collectives with GPU tensors, hogwild GPU worker."""
+
+import numpy as np
+import pytest
+import torch
+import torch.nn as nn
+
+pytestmark = pytest.mark.gpu
+
+if not torch.cuda.is_available():
+    pytest.skip("needs MI355X", allow_module_level=True)
+
+from sparktorch_amd import SparkTorch, serialize_torch_obj
+from sparktorch_amd.compat.local import LocalDataFrame, free_port
+from sparktorch_amd.models.mnist import MnistMLP
+
+
+def _df(n=400, dim=20, parts=2):
+    rng = np.random.RandomState(0)
+    feats = rng.normal(0, 1, (n, dim)).astype(np.float64)
+    labels = (feats.sum(axis=1) > 0).astype(np.float64)
+    return LocalDataFrame.from_arrays(feats, list(labels), num_partitions=parts)
+
+
+def test_estimator_fit_transform_on_gpu_single_partition():
+    df = _df(parts=1)
+    obj = serialize_torch_obj(
+        MnistMLP(in_dim=20, hidden=32, classes=2), nn.CrossEntropyLoss(), torch.optim.Adam, lr=0.01
+    )
+    est = SparkTorch(
+        inputCol="features",
+        labelCol="label",
+        predictionCol="predicted",
+        torchObj=obj,
+        iters=10,
+        mode="synchronous",
+        device="cuda:0",
+    )
+    model = est.fit(df)
+    out = model.transform(df)
+    preds = {r["predicted"] for r in out.collect()}
+    assert preds.issubset({0.0, 1.0})
+
+
+def test_sync_two_ranks_one_gpu_gloo():
+    """world_size=2 on one GPU over gloo: exercises the full bucket/direct-
+    grad/all-reduce/consensus path with GPU tensors (RCCL needs one rank per
+    GPU; gloo stands in so the logic is testable on a 1-GPU box)."""
+    from sparktorch_amd.parallel.sync import train_distributed
+    from sparktorch_amd.utils.data import handle_data
+
+    df = _df(parts=2)
+    obj = serialize_torch_obj(
+        MnistMLP(in_dim=20, hidden=16, classes=2), nn.CrossEntropyLoss(), torch.optim.Adam, lr=0.01
+    )
+    rdd = df.rdd.mapPartitions(handle_data("features", "label"))
+    state = train_distributed(
+        rdd, obj, iters=3, device="cuda:0", backend="gloo", early_stop_patience=2,
+        validation_pct=0.25,
+    )
+    for v in state.values():
+        assert torch.isfinite(v).all()
+
+
+def test_hogwild_gpu_worker():
+    df = _df(parts=2)
+    obj = serialize_torch_obj(
+        MnistMLP(in_dim=20, hidden=16, classes=2), nn.MSELoss(), torch.optim.Adam, lr=0.01
+    )
+    est = SparkTorch(
+        inputCol="features",
+        labelCol="label",
+        predictionCol="predicted",
+        torchObj=obj,
+        iters=3,
+        mode="hogwild",
+        port=free_port(),
+        device="cuda:0",
+        acquireLock=True,
+    )
+    model = est.fit(df)
+    assert model.transform(df).count() == 400

@@ -79,6 +79,24 @@ def test_linear_wgrad_splitk(splitk):
     assert rel < 0.02, rel
 
 
+@pytest.mark.parametrize("K", [96, 128, 784])  # K%128==0 exercises the extra tile
+def test_linear_wgrad_bias_fused(K):
+    torch.manual_seed(30)
+    B, N = 4096, 64
+    dz = torch.randn(B, N, device=DEV) * 0.03
+    x = torch.randn(B, K, device=DEV)
+    dw = torch.zeros(N, K, device=DEV)
+    db = torch.zeros(N, device=DEV)
+    ops.ext().linear_wgrad_bias_into(bf(dz).contiguous(), bf(x).contiguous(), dw, db, 8)
+    ref_dw = ref_of(dz).t() @ ref_of(x)
+    ref_db = ref_of(dz).sum(0)
+    assert (dw - ref_dw).abs().max().item() / (ref_dw.abs().max().item() + 1e-6) < 0.02
+    assert (db - ref_db).abs().max().item() / (ref_db.abs().max().item() + 1e-6) < 0.02
+    # accumulate semantics: second call doubles
+    ops.ext().linear_wgrad_bias_into(bf(dz).contiguous(), bf(x).contiguous(), dw, db, 8)
+    assert (dw - 2 * ref_dw).abs().max().item() / (2 * ref_dw.abs().max().item() + 1e-6) < 0.02
+
+
 def test_bias_grad():
     dz = torch.randn(5000, 37, device=DEV)
     db = ops.ext().bias_grad(bf(dz).contiguous())
