@@ -212,6 +212,87 @@ void bias_grad_into(at::Tensor dz, at::Tensor db) {
                              (int)dz.size(0), (int)dz.size(1), cur_stream()));
 }
 
+// ---------------------------------------------------------------------------
+// Conv2d lowering + pooling + dropout (conv.hip)
+// ---------------------------------------------------------------------------
+
+extern "C" {
+hipError_t launch_im2col(const bf16raw*, bf16raw*, int, int, int, int, int, int, int, int, int,
+                         int, int, int, hipStream_t);
+hipError_t launch_col2im(const bf16raw*, bf16raw*, int, int, int, int, int, int, int, int, int,
+                         int, int, int, hipStream_t);
+hipError_t launch_maxpool_fwd(const bf16raw*, bf16raw*, uint8_t*, int, int, int, int, int, int,
+                              hipStream_t);
+hipError_t launch_maxpool_bwd(const bf16raw*, const uint8_t*, bf16raw*, int, int, int, int, int,
+                              int, hipStream_t);
+hipError_t launch_dropout(const bf16raw*, bf16raw*, int64_t, int64_t, float, uint32_t,
+                          hipStream_t);
+}
+
+at::Tensor im2col(at::Tensor x, int64_t KH, int64_t KW, int64_t sh, int64_t sw, int64_t ph,
+                  int64_t pw) {
+  check_gpu_contig(x, at::kBFloat16, "x");
+  TORCH_CHECK(x.dim() == 4, "x must be NCHW");
+  int64_t B = x.size(0), CI = x.size(1), H = x.size(2), W = x.size(3);
+  int64_t HO = (H + 2 * ph - KH) / sh + 1;
+  int64_t WO = (W + 2 * pw - KW) / sw + 1;
+  auto col = at::empty({B * HO * WO, CI * KH * KW}, x.options());
+  CHECK_HIP(launch_im2col((const bf16raw*)x.data_ptr(), (bf16raw*)col.data_ptr(), (int)B, (int)CI,
+                          (int)H, (int)W, (int)KH, (int)KW, (int)HO, (int)WO, (int)sh, (int)sw,
+                          (int)ph, (int)pw, cur_stream()));
+  return col;
+}
+
+at::Tensor col2im(at::Tensor dcol, int64_t B, int64_t CI, int64_t H, int64_t W, int64_t KH,
+                  int64_t KW, int64_t sh, int64_t sw, int64_t ph, int64_t pw) {
+  check_gpu_contig(dcol, at::kBFloat16, "dcol");
+  int64_t HO = (H + 2 * ph - KH) / sh + 1;
+  int64_t WO = (W + 2 * pw - KW) / sw + 1;
+  TORCH_CHECK(dcol.size(0) == B * HO * WO && dcol.size(1) == CI * KH * KW);
+  auto dx = at::empty({B, CI, H, W}, dcol.options());
+  CHECK_HIP(launch_col2im((const bf16raw*)dcol.data_ptr(), (bf16raw*)dx.data_ptr(), (int)B,
+                          (int)CI, (int)H, (int)W, (int)KH, (int)KW, (int)HO, (int)WO, (int)sh,
+                          (int)sw, (int)ph, (int)pw, cur_stream()));
+  return dx;
+}
+
+std::tuple<at::Tensor, at::Tensor> maxpool_fwd(at::Tensor x, int64_t ks) {
+  check_gpu_contig(x, at::kBFloat16, "x");
+  TORCH_CHECK(x.dim() == 4);
+  int64_t B = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  int64_t HO = H / ks, WO = W / ks;
+  auto y = at::empty({B, C, HO, WO}, x.options());
+  auto arg = at::empty({B, C, HO, WO}, x.options().dtype(at::kByte));
+  CHECK_HIP(launch_maxpool_fwd((const bf16raw*)x.data_ptr(), (bf16raw*)y.data_ptr(),
+                               arg.data_ptr<uint8_t>(), (int)(B * C), (int)H, (int)W, (int)ks,
+                               (int)HO, (int)WO, cur_stream()));
+  return {y, arg};
+}
+
+at::Tensor maxpool_bwd(at::Tensor dy, at::Tensor arg, int64_t H, int64_t W, int64_t ks) {
+  check_gpu_contig(dy, at::kBFloat16, "dy");
+  int64_t B = dy.size(0), C = dy.size(1);
+  auto dx = at::empty({B, C, H, W}, dy.options());
+  // windows are disjoint (stride == kernel): bwd writes every slot, no zero
+  CHECK_HIP(launch_maxpool_bwd((const bf16raw*)dy.data_ptr(), arg.data_ptr<uint8_t>(),
+                               (bf16raw*)dx.data_ptr(), (int)(B * C), (int)H, (int)W, (int)ks,
+                               (int)dy.size(2), (int)dy.size(3), cur_stream()));
+  // rows/cols beyond HO*ks / WO*ks (non-divisible inputs) keep garbage: zero
+  if (H % ks || W % ks) {
+    dx.slice(2, (H / ks) * ks, H).zero_();
+    dx.slice(3, (W / ks) * ks, W).zero_();
+  }
+  return dx;
+}
+
+at::Tensor dropout_apply(at::Tensor x, double p, int64_t seed, int64_t units_div) {
+  check_gpu_contig(x, at::kBFloat16, "x");
+  auto y = at::empty_like(x);
+  CHECK_HIP(launch_dropout((const bf16raw*)x.data_ptr(), (bf16raw*)y.data_ptr(), x.numel(),
+                           units_div, (float)p, (uint32_t)(seed & 0xffffffff), cur_stream()));
+  return y;
+}
+
 // generic C = A @ B with optional transposes, for tests/other models
 at::Tensor matmul_bf16(at::Tensor a, at::Tensor b, bool trans_a, bool trans_b) {
   check_gpu_contig(a, at::kBFloat16, "a");
@@ -248,4 +329,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("linear_wgrad_bias_into", &linear_wgrad_bias_into, "dW + db in one MFMA launch");
   m.def("bias_grad_into", &bias_grad_into, "db accumulated into grad view");
   m.def("matmul_bf16", &matmul_bf16, "generic bf16 MFMA GEMM");
+  m.def("im2col", &im2col, "NCHW -> implicit-GEMM col matrix");
+  m.def("col2im", &col2im, "col gradient -> NCHW input gradient");
+  m.def("maxpool_fwd", &maxpool_fwd, "max_pool2d (kernel==stride) + argmax");
+  m.def("maxpool_bwd", &maxpool_bwd, "max_pool2d backward scatter");
+  m.def("dropout_apply", &dropout_apply, "counter-based dropout / dropout2d");
 }

@@ -103,6 +103,119 @@ def hip_linear(
     return F.relu(y) if relu else y
 
 
+class _Conv2dFn(torch.autograd.Function):
+    """Conv2d as implicit GEMM: im2col -> MFMA linear (+bias)(+relu) ->
+    NCHW permute; backward reuses the fused wgrad+bias kernel on the saved
+    col matrix and col2im for the input gradient."""
+
+    @staticmethod
+    def forward(ctx, x, w, b, stride, padding, relu):
+        ext = ops.ext()
+        B, CI, H, W = x.shape
+        CO, _, KH, KW = w.shape
+        sh, sw = stride
+        ph, pw = padding
+        HO = (H + 2 * ph - KH) // sh + 1
+        WO = (W + 2 * pw - KW) // sw + 1
+        col = ext.im2col(x, KH, KW, sh, sw, ph, pw)
+        w2d = w.reshape(CO, CI * KH * KW).contiguous()
+        y2d = ext.linear_fwd(col, w2d, b, relu)  # [B*HO*WO, CO]
+        ctx.save_for_backward(col, w2d, y2d)
+        ctx.meta = (B, CI, H, W, CO, KH, KW, sh, sw, ph, pw, HO, WO, relu, b is not None)
+        ctx.w_ref, ctx.b_ref = w, b
+        return y2d.view(B, HO, WO, CO).permute(0, 3, 1, 2).contiguous()
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = ops.ext()
+        col, w2d, y2d = ctx.saved_tensors
+        B, CI, H, W, CO, KH, KW, sh, sw, ph, pw, HO, WO, relu, has_bias = ctx.meta
+        dy2d = dy.permute(0, 2, 3, 1).reshape(B * HO * WO, CO).contiguous()
+        if dy2d.dtype != torch.bfloat16:
+            dy2d = dy2d.to(torch.bfloat16)
+        dz = ext.relu_bwd(dy2d, y2d) if relu else dy2d
+
+        w, b = ctx.w_ref, ctx.b_ref
+        dw = db = None
+        sk = _choose_splitk(dz.shape[0], CO, col.shape[1])
+        w_notify = getattr(w, "_bucket_notify", None)
+        b_notify = getattr(b, "_bucket_notify", None) if b is not None else None
+        if ctx.needs_input_grad[1]:
+            if w_notify is not None and w.grad is not None and has_bias and b_notify is not None and b.grad is not None:
+                ext.linear_wgrad_bias_into(dz, col, w.grad.reshape(CO, -1), b.grad, sk)
+                w_notify()
+                b_notify()
+            else:
+                dw = ext.linear_wgrad(dz, col, sk).view(CO, CI, KH, KW)
+                if has_bias and ctx.needs_input_grad[2]:
+                    db = ext.bias_grad(dz)
+        elif has_bias and ctx.needs_input_grad[2]:
+            db = ext.bias_grad(dz)
+
+        dx = None
+        if ctx.needs_input_grad[0]:
+            dcol = ext.linear_dgrad(dz, w2d)
+            dx = ext.col2im(dcol, B, CI, H, W, KH, KW, sh, sw, ph, pw)
+        return dx, dw, db, None, None, None
+
+
+def hip_conv2d(x, weight, bias=None, stride=(1, 1), padding=(0, 0), relu=False):
+    if x.is_cuda:
+        if x.dtype != torch.bfloat16:
+            x = x.to(torch.bfloat16)
+        return _Conv2dFn.apply(x.contiguous(), weight, bias, stride, padding, relu)
+    y = F.conv2d(x, weight, bias, stride=stride, padding=padding)
+    return F.relu(y) if relu else y
+
+
+class _MaxPool2dFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, ks):
+        y, arg = ops.ext().maxpool_fwd(x, ks)
+        ctx.save_for_backward(arg)
+        ctx.meta = (x.shape[2], x.shape[3], ks)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (arg,) = ctx.saved_tensors
+        H, W, ks = ctx.meta
+        return ops.ext().maxpool_bwd(dy.contiguous().to(torch.bfloat16), arg, H, W, ks), None
+
+
+def hip_max_pool2d(x, kernel_size: int):
+    if x.is_cuda:
+        if x.dtype != torch.bfloat16:
+            x = x.to(torch.bfloat16)
+        return _MaxPool2dFn.apply(x.contiguous(), kernel_size)
+    return F.max_pool2d(x, kernel_size)
+
+
+class _DropoutFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, p, seed, units_div):
+        ctx.meta = (p, seed, units_div)
+        return ops.ext().dropout_apply(x, p, seed, units_div)
+
+    @staticmethod
+    def backward(ctx, dy):
+        p, seed, units_div = ctx.meta
+        # same (seed, unit) hash -> same mask; scale applies to dy too
+        return ops.ext().dropout_apply(dy.contiguous().to(torch.bfloat16), p, seed, units_div), None, None, None
+
+
+def hip_dropout(x, p: float, training: bool = True, channel_wise: bool = False):
+    if not training or p <= 0.0:
+        return x
+    if x.is_cuda:
+        if x.dtype != torch.bfloat16:
+            x = x.to(torch.bfloat16)
+        units_div = (x.shape[2] * x.shape[3]) if (channel_wise and x.dim() == 4) else 1
+        seed = int(torch.randint(0, 2**31 - 1, (1,)).item())
+        return _DropoutFn.apply(x.contiguous(), p, seed, units_div)
+    return F.dropout2d(x, p, training) if channel_wise else F.dropout(x, p, training)
+
+
 class _CEFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, logits: torch.Tensor, target: torch.Tensor):

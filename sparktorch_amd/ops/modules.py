@@ -15,7 +15,14 @@ from typing import Optional
 import torch
 import torch.nn as nn
 
-from sparktorch_amd.ops.functional import hip_cross_entropy, hip_linear, hip_mse
+from sparktorch_amd.ops.functional import (
+    hip_conv2d,
+    hip_cross_entropy,
+    hip_dropout,
+    hip_linear,
+    hip_max_pool2d,
+    hip_mse,
+)
 
 
 class HipLinear(nn.Module):
@@ -49,6 +56,83 @@ class HipLinear(nn.Module):
         return "in=%d, out=%d, act=%s" % (self.in_features, self.out_features, self.activation)
 
 
+class HipConv2d(nn.Module):
+    """Drop-in nn.Conv2d on the implicit-GEMM MFMA path (stride/padding;
+    dilation=1, groups=1), optional fused ReLU."""
+
+    def __init__(self, in_ch: int, out_ch: int, kernel_size: int, stride: int = 1,
+                 padding: int = 0, bias: bool = True, activation: Optional[str] = None):
+        super().__init__()
+        ks = (kernel_size, kernel_size) if isinstance(kernel_size, int) else kernel_size
+        self.stride = (stride, stride) if isinstance(stride, int) else stride
+        self.padding = (padding, padding) if isinstance(padding, int) else padding
+        self.weight = nn.Parameter(torch.empty(out_ch, in_ch, *ks))
+        self.bias = nn.Parameter(torch.zeros(out_ch)) if bias else None
+        self.activation = activation
+        nn.init.kaiming_uniform_(self.weight, a=5 ** 0.5)
+
+    @classmethod
+    def from_conv2d(cls, conv: nn.Conv2d, activation: Optional[str] = None) -> "HipConv2d":
+        if (conv.dilation != (1, 1)) or (conv.groups != 1) or conv.padding_mode != "zeros":
+            raise ValueError("HipConv2d supports dilation=1, groups=1, zero padding")
+        m = cls.__new__(cls)
+        nn.Module.__init__(m)
+        m.stride = conv.stride
+        m.padding = conv.padding
+        m.weight = conv.weight  # SAME Parameter objects
+        m.bias = conv.bias
+        m.activation = activation
+        return m
+
+    def forward(self, x):
+        return hip_conv2d(
+            x, self.weight, self.bias, self.stride, self.padding, relu=self.activation == "relu"
+        )
+
+
+class HipMaxPool2d(nn.Module):
+    def __init__(self, kernel_size: int):
+        super().__init__()
+        self.kernel_size = kernel_size
+
+    def forward(self, x):
+        return hip_max_pool2d(x, self.kernel_size)
+
+
+class HipDropout(nn.Module):
+    def __init__(self, p: float = 0.5, channel_wise: bool = False):
+        super().__init__()
+        self.p = p
+        self.channel_wise = channel_wise
+
+    def forward(self, x):
+        return hip_dropout(x, self.p, training=self.training, channel_wise=self.channel_wise)
+
+
+class MnistCNNFused(nn.Module):
+    """The reference example CNN (examples/cnn_network.py:6-24) on the fully
+    native path: implicit-GEMM convs with fused ReLU, native maxpool,
+    counter-based Dropout2d.  state_dict-compatible with models.mnist.MnistCNN."""
+
+    def __init__(self):
+        super().__init__()
+        self.conv1 = HipConv2d(1, 16, kernel_size=5, activation="relu")
+        self.conv2 = HipConv2d(16, 32, kernel_size=3, activation="relu")
+        self.dropout = HipDropout(p=0.25, channel_wise=True)
+        self.fc = HipLinear(3872, 10)
+
+    def forward(self, x):
+        if x.is_cuda and x.dtype != torch.bfloat16:
+            x = x.to(torch.bfloat16)
+        x = x.view(-1, 1, 28, 28)
+        x = self.conv1(x)
+        x = self.conv2(x)
+        x = hip_max_pool2d(x, 2)
+        x = self.dropout(x)
+        x = torch.flatten(x, 1)
+        return self.fc(x)
+
+
 class HipCrossEntropy(nn.Module):
     def forward(self, logits, target):
         return hip_cross_entropy(logits, target)
@@ -76,10 +160,27 @@ class MnistMLPFused(nn.Module):
 
 
 def convert_model_for_mi355x(model: nn.Module) -> nn.Module:
-    """Swap nn.Linear -> HipLinear in place (same Parameters)."""
+    """Swap torch modules for their native-kernel equivalents in place
+    (adopting the SAME Parameter objects, so optimizers/buckets are
+    unaffected).  Modules with unsupported configs are left as-is (they run
+    through torch-ROCm eager, e.g. grouped/dilated convs)."""
     for name, child in model.named_children():
-        if isinstance(child, nn.Linear):
+        cname = type(child).__name__
+        if cname in ("Linear",) or isinstance(child, nn.Linear):
             setattr(model, name, HipLinear.from_linear(child))
+        elif cname in ("Conv2d",) or isinstance(child, nn.Conv2d):
+            try:
+                setattr(model, name, HipConv2d.from_conv2d(child))
+            except (ValueError, AttributeError):
+                pass
+        elif cname == "MaxPool2d" and getattr(child, "kernel_size", None) == getattr(child, "stride", 0):
+            ks = child.kernel_size
+            if isinstance(ks, int):
+                setattr(model, name, HipMaxPool2d(ks))
+        elif cname == "Dropout2d":
+            setattr(model, name, HipDropout(p=child.p, channel_wise=True))
+        elif cname == "Dropout":
+            setattr(model, name, HipDropout(p=child.p, channel_wise=False))
         else:
             convert_model_for_mi355x(child)
     return model
