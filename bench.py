@@ -37,8 +37,10 @@ def main() -> int:
     ap.add_argument("--batch", type=int, default=65536, help="per-GPU batch (weak scaling)")
     ap.add_argument("--device", type=str, default=None)
     ap.add_argument("--hipgraph", action="store_true", help="capture the train step in a HIP graph")
-    ap.add_argument("--model", type=str, default="mnist_mlp", choices=["mnist_mlp", "mnist_cnn"],
-                    help="mnist_mlp = BASELINE flagship; mnist_cnn = examples/simple_cnn config")
+    ap.add_argument("--model", type=str, default="mnist_mlp",
+                    choices=["mnist_mlp", "mnist_cnn", "resnet18"],
+                    help="mnist_mlp = BASELINE flagship; mnist_cnn = examples/simple_cnn config; "
+                         "resnet18 = BASELINE config 4 (synthetic 3x224x224)")
     ap.add_argument("--mode", type=str, default="train", choices=["train", "infer"],
                     help="infer = saved-pipeline batch inference (HIP-graph forward)")
     args = ap.parse_args()
@@ -69,6 +71,17 @@ def main() -> int:
             from sparktorch_amd.models.mnist import MnistCNN
 
             model = MnistCNN()
+    elif args.model == "resnet18":
+        if args.batch == 65536:  # default was sized for the MLP
+            args.batch = 256
+        if on_gpu:
+            from sparktorch_amd.ops.modules import ResNet18Fused
+
+            model = ResNet18Fused()
+        else:
+            from sparktorch_amd.models.resnet import ResNet18
+
+            model = ResNet18()
     elif on_gpu:
         from sparktorch_amd.ops.modules import MnistMLPFused
 
@@ -91,10 +104,12 @@ def main() -> int:
         compile_mode="hipgraph" if (args.hipgraph and world == 1 and on_gpu) else None,
     )
 
-    x = torch.randn(args.batch, 784, device=device)
+    in_dim = 3 * 224 * 224 if args.model == "resnet18" else 784
+    n_classes = 1000 if args.model == "resnet18" else 10
+    x = torch.randn(args.batch, in_dim, device=device)
     if on_gpu:
         x = x.to(torch.bfloat16)
-    y = torch.randint(0, 10, (args.batch,), device=device)
+    y = torch.randint(0, n_classes, (args.batch,), device=device)
 
     def barrier_sync():
         if world > 1:
@@ -138,8 +153,9 @@ def main() -> int:
             "dtype": "bf16" if on_gpu else "fp32",
             "data": "synthetic",
             "config": {
-                "model": "mnist_mlp_784x256x256x10" if args.model == "mnist_mlp"
-                         else "mnist_cnn_conv16x5_conv32x3_fc3872x10",
+                "model": {"mnist_mlp": "mnist_mlp_784x256x256x10",
+                          "mnist_cnn": "mnist_cnn_conv16x5_conv32x3_fc3872x10",
+                          "resnet18": "resnet18_3x224x224_1000cls"}[args.model],
                 "global_batch": args.batch * n_gpus,
                 "seq_len": None,
                 "parallelism": "dp%d" % n_gpus,
@@ -162,7 +178,8 @@ def run_infer(args, model, device, on_gpu) -> int:
 
     model = model.to(device).eval()
     bs = args.batch
-    x = torch.randn(bs, 784, device=device)
+    in_dim = 3 * 224 * 224 if args.model == "resnet18" else 784
+    x = torch.randn(bs, in_dim, device=device)
     if on_gpu:
         from sparktorch_amd.ops.graph import GraphedForward
 
@@ -172,7 +189,7 @@ def run_infer(args, model, device, on_gpu) -> int:
         def step():
             runner._static_in.copy_(x)
             runner._graph.replay() if runner._graph else runner(x.cpu())
-        runner(torch.randn(bs, 784))  # capture
+        runner(torch.randn(bs, in_dim))  # capture
     else:
         def step():
             with torch.no_grad():

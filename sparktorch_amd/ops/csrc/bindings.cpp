@@ -27,6 +27,23 @@ hipError_t launch_mse_fused(const bf16raw*, const bf16raw*, float*, bf16raw*, in
 hipError_t launch_gemm_bf16(const void*, const void*, int, float*, bf16raw*, const float*, int,
                             int, int, int64_t, int64_t, int64_t, int64_t, int, int, float*, int,
                             hipStream_t);
+hipError_t launch_bn_stats(const bf16raw*, float*, float*, float*, float*, float*, float*, int,
+                           int, int64_t, float, float, int, hipStream_t);
+hipError_t launch_bn_apply(const bf16raw*, const bf16raw*, bf16raw*, const float*, const float*,
+                           const float*, const float*, int, int64_t, int64_t, int, hipStream_t);
+hipError_t launch_bn_bwd_reduce(const bf16raw*, const bf16raw*, const bf16raw*, const float*,
+                                const float*, float*, float*, int, int, int64_t, int,
+                                hipStream_t);
+hipError_t launch_bn_bwd_dx(const bf16raw*, const bf16raw*, const bf16raw*, const float*,
+                            const float*, const float*, const float*, const float*, bf16raw*,
+                            int, int64_t, int64_t, float, int, hipStream_t);
+hipError_t launch_add_relu(const bf16raw*, const bf16raw*, bf16raw*, int64_t, hipStream_t);
+hipError_t launch_maxpool_gen_fwd(const bf16raw*, bf16raw*, uint8_t*, int64_t, int, int, int,
+                                  int, int, int, int, hipStream_t);
+hipError_t launch_maxpool_gen_bwd(const bf16raw*, const uint8_t*, bf16raw*, int64_t, int, int,
+                                  int, int, int, int, int, hipStream_t);
+hipError_t launch_gap_fwd(const bf16raw*, bf16raw*, int64_t, int64_t, hipStream_t);
+hipError_t launch_gap_bwd(const bf16raw*, bf16raw*, int64_t, int64_t, hipStream_t);
 }
 
 #define CHECK_HIP(err)                                                              \
@@ -313,6 +330,154 @@ at::Tensor matmul_bf16(at::Tensor a, at::Tensor b, bool trans_a, bool trans_b) {
   return c;
 }
 
+// --------------------------- BatchNorm / ResNet ops ------------------------
+
+static int bn_split(int C) {
+  // aim for ~2048 blocks so small-C layers still fill 256 CUs / 8 XCDs
+  int s = 2048 / (C > 0 ? C : 1);
+  return s < 1 ? 1 : (s > 64 ? 64 : s);
+}
+
+// training-mode stats: returns (mean, invstd); updates running stats in-place
+// when given.
+std::tuple<at::Tensor, at::Tensor> bn_stats(at::Tensor x, c10::optional<at::Tensor> running_mean,
+                                            c10::optional<at::Tensor> running_var,
+                                            double momentum, double eps) {
+  check_gpu_contig(x, at::kBFloat16, "x");
+  TORCH_CHECK(x.dim() == 4, "x must be NCHW");
+  int B = (int)x.size(0), C = (int)x.size(1);
+  int64_t HW = x.size(2) * x.size(3);
+  auto fopt = x.options().dtype(at::kFloat);
+  auto sum = at::zeros({C}, fopt), sumsq = at::zeros({C}, fopt);
+  auto mean = at::empty({C}, fopt), invstd = at::empty({C}, fopt);
+  float* rm = nullptr;
+  float* rv = nullptr;
+  if (running_mean.has_value()) {
+    check_gpu_contig(*running_mean, at::kFloat, "running_mean");
+    check_gpu_contig(*running_var, at::kFloat, "running_var");
+    rm = running_mean->data_ptr<float>();
+    rv = running_var->data_ptr<float>();
+  }
+  CHECK_HIP(launch_bn_stats((const bf16raw*)x.data_ptr(), sum.data_ptr<float>(),
+                            sumsq.data_ptr<float>(), mean.data_ptr<float>(),
+                            invstd.data_ptr<float>(), rm, rv, B, C, HW, (float)momentum,
+                            (float)eps, bn_split(C), cur_stream()));
+  return {mean, invstd};
+}
+
+at::Tensor bn_apply(at::Tensor x, c10::optional<at::Tensor> res, at::Tensor mean,
+                    at::Tensor invstd, at::Tensor gamma, at::Tensor beta, bool relu) {
+  check_gpu_contig(x, at::kBFloat16, "x");
+  TORCH_CHECK(x.dim() == 4, "x must be NCHW");
+  const bf16raw* rp = nullptr;
+  if (res.has_value()) {
+    check_gpu_contig(*res, at::kBFloat16, "res");
+    TORCH_CHECK(res->sizes() == x.sizes(), "residual shape mismatch");
+    rp = (const bf16raw*)res->data_ptr();
+  }
+  auto y = at::empty_like(x);
+  CHECK_HIP(launch_bn_apply((const bf16raw*)x.data_ptr(), rp, (bf16raw*)y.data_ptr(),
+                            mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                            gamma.data_ptr<float>(), beta.data_ptr<float>(), (int)x.size(1),
+                            x.size(2) * x.size(3), x.numel(), relu ? 1 : 0, cur_stream()));
+  return y;
+}
+
+// dbeta/dgamma are accumulated IN-PLACE (callers pass pre-zeroed fp32
+// buffers — possibly flat-bucket grad views, same convention as
+// linear_wgrad_into).
+void bn_bwd_reduce(at::Tensor dy, c10::optional<at::Tensor> yrelu, at::Tensor x, at::Tensor mean,
+                   at::Tensor invstd, at::Tensor dbeta, at::Tensor dgamma) {
+  check_gpu_contig(dy, at::kBFloat16, "dy");
+  check_gpu_contig(x, at::kBFloat16, "x");
+  check_gpu_contig(dbeta, at::kFloat, "dbeta");
+  check_gpu_contig(dgamma, at::kFloat, "dgamma");
+  const bf16raw* yp = nullptr;
+  if (yrelu.has_value()) yp = (const bf16raw*)yrelu->data_ptr();
+  int B = (int)x.size(0), C = (int)x.size(1);
+  int64_t HW = x.size(2) * x.size(3);
+  CHECK_HIP(launch_bn_bwd_reduce((const bf16raw*)dy.data_ptr(), yp, (const bf16raw*)x.data_ptr(),
+                                 mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                                 dbeta.data_ptr<float>(), dgamma.data_ptr<float>(), B, C, HW,
+                                 bn_split(C), cur_stream()));
+}
+
+at::Tensor bn_bwd_dx(at::Tensor dy, c10::optional<at::Tensor> yrelu, at::Tensor x,
+                     at::Tensor mean, at::Tensor invstd, at::Tensor gamma, at::Tensor dbeta,
+                     at::Tensor dgamma, bool train_stats) {
+  check_gpu_contig(dy, at::kBFloat16, "dy");
+  check_gpu_contig(x, at::kBFloat16, "x");
+  const bf16raw* yp = nullptr;
+  if (yrelu.has_value()) yp = (const bf16raw*)yrelu->data_ptr();
+  int C = (int)x.size(1);
+  int64_t HW = x.size(2) * x.size(3);
+  auto dx = at::empty_like(x);
+  float inv_count = 1.0f / (float)(x.size(0) * HW);
+  CHECK_HIP(launch_bn_bwd_dx((const bf16raw*)dy.data_ptr(), yp, (const bf16raw*)x.data_ptr(),
+                             mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                             gamma.data_ptr<float>(), dbeta.data_ptr<float>(),
+                             dgamma.data_ptr<float>(), (bf16raw*)dx.data_ptr(), C, HW, x.numel(),
+                             inv_count, train_stats ? 1 : 0, cur_stream()));
+  return dx;
+}
+
+at::Tensor add_relu(at::Tensor a, at::Tensor b) {
+  check_gpu_contig(a, at::kBFloat16, "a");
+  check_gpu_contig(b, at::kBFloat16, "b");
+  TORCH_CHECK(a.sizes() == b.sizes());
+  auto out = at::empty_like(a);
+  CHECK_HIP(launch_add_relu((const bf16raw*)a.data_ptr(), (const bf16raw*)b.data_ptr(),
+                            (bf16raw*)out.data_ptr(), a.numel(), cur_stream()));
+  return out;
+}
+
+std::tuple<at::Tensor, at::Tensor> maxpool_gen_fwd(at::Tensor x, int64_t ks, int64_t stride,
+                                                   int64_t pad) {
+  check_gpu_contig(x, at::kBFloat16, "x");
+  TORCH_CHECK(x.dim() == 4, "x must be NCHW");
+  TORCH_CHECK(ks <= 15, "kernel too large for uint8 argmax");
+  int H = (int)x.size(2), W = (int)x.size(3);
+  int HO = (int)((H + 2 * pad - ks) / stride + 1);
+  int WO = (int)((W + 2 * pad - ks) / stride + 1);
+  int64_t BC = x.size(0) * x.size(1);
+  auto y = at::empty({x.size(0), x.size(1), HO, WO}, x.options());
+  auto arg = at::empty({x.size(0), x.size(1), HO, WO}, x.options().dtype(at::kByte));
+  CHECK_HIP(launch_maxpool_gen_fwd((const bf16raw*)x.data_ptr(), (bf16raw*)y.data_ptr(),
+                                   arg.data_ptr<uint8_t>(), BC, H, W, HO, WO, (int)ks,
+                                   (int)stride, (int)pad, cur_stream()));
+  return {y, arg};
+}
+
+at::Tensor maxpool_gen_bwd(at::Tensor dy, at::Tensor arg, int64_t H, int64_t W, int64_t ks,
+                           int64_t stride, int64_t pad) {
+  check_gpu_contig(dy, at::kBFloat16, "dy");
+  check_gpu_contig(arg, at::kByte, "arg");
+  int HO = (int)dy.size(2), WO = (int)dy.size(3);
+  int64_t BC = dy.size(0) * dy.size(1);
+  auto dx = at::empty({dy.size(0), dy.size(1), H, W}, dy.options());
+  CHECK_HIP(launch_maxpool_gen_bwd((const bf16raw*)dy.data_ptr(), arg.data_ptr<uint8_t>(),
+                                   (bf16raw*)dx.data_ptr(), BC, (int)H, (int)W, HO, WO, (int)ks,
+                                   (int)stride, (int)pad, cur_stream()));
+  return dx;
+}
+
+at::Tensor gap_fwd(at::Tensor x) {
+  check_gpu_contig(x, at::kBFloat16, "x");
+  TORCH_CHECK(x.dim() == 4, "x must be NCHW");
+  auto y = at::empty({x.size(0), x.size(1)}, x.options());
+  CHECK_HIP(launch_gap_fwd((const bf16raw*)x.data_ptr(), (bf16raw*)y.data_ptr(),
+                           x.size(0) * x.size(1), x.size(2) * x.size(3), cur_stream()));
+  return y;
+}
+
+at::Tensor gap_bwd(at::Tensor dy, int64_t H, int64_t W) {
+  check_gpu_contig(dy, at::kBFloat16, "dy");
+  auto dx = at::empty({dy.size(0), dy.size(1), H, W}, dy.options());
+  CHECK_HIP(launch_gap_bwd((const bf16raw*)dy.data_ptr(), (bf16raw*)dx.data_ptr(), H * W,
+                           dx.numel(), cur_stream()));
+  return dx;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_adam", &fused_adam, "fused Adam on a flat bucket");
   m.def("fused_sgd", &fused_sgd, "fused SGD on a flat bucket");
@@ -334,4 +499,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("maxpool_fwd", &maxpool_fwd, "max_pool2d (kernel==stride) + argmax");
   m.def("maxpool_bwd", &maxpool_bwd, "max_pool2d backward scatter");
   m.def("dropout_apply", &dropout_apply, "counter-based dropout / dropout2d");
+  m.def("bn_stats", &bn_stats, "BatchNorm2d training stats (mean, invstd) + running update");
+  m.def("bn_apply", &bn_apply, "BN normalize+affine (+residual)(+relu)");
+  m.def("bn_bwd_reduce", &bn_bwd_reduce, "BN backward: dbeta/dgamma accumulated in-place");
+  m.def("bn_bwd_dx", &bn_bwd_dx, "BN backward input gradient");
+  m.def("add_relu", &add_relu, "out = relu(a+b) (residual join)");
+  m.def("maxpool_gen_fwd", &maxpool_gen_fwd, "general max_pool2d (stride/pad) + argmax");
+  m.def("maxpool_gen_bwd", &maxpool_gen_bwd, "general max_pool2d backward (gather)");
+  m.def("gap_fwd", &gap_fwd, "global average pool [B,C,H,W]->[B,C]");
+  m.def("gap_bwd", &gap_bwd, "global average pool backward");
 }

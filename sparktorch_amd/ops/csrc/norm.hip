@@ -1,0 +1,376 @@
+// BatchNorm2d / pooling / residual kernels for gfx950 — the ResNet family
+// ops (BASELINE config 4: ResNet-18 on synthetic 3x224x224 rows).  The
+// reference has no GPU code at all; its conv models run
+// eager CPU torch.  Here BN training statistics, normalization (+fused
+// ReLU, +fused residual add), BN backward, overlapping max_pool2d, global
+// average pool and fused add+relu are hand-written CDNA4 kernels: bf16
+// activations, fp32 statistics/params, LDS wave reductions, grids split
+// over (channel x batch-slice) so small-C layers still cover all 256 CUs.
+//
+// All NCHW contiguous.
+
+#include "common.h"
+
+// ---------------------------------------------------------------------------
+// BatchNorm2d forward statistics.
+// Pass 1: grid (C, S) blocks; block (c, s) reduces a slice of the B*HW
+// per-channel elements into LDS, then one atomicAdd per block into
+// sum[c] / sumsq[c] (fp32, caller-zeroed).
+// ---------------------------------------------------------------------------
+
+__global__ void bn_stats_partial_kernel(const bf16raw* __restrict__ x, float* __restrict__ sum,
+                                        float* __restrict__ sumsq, int B, int C, int64_t HW) {
+  int c = blockIdx.x;
+  int64_t total = (int64_t)B * HW;
+  int64_t per = ceil_div_i64(total, gridDim.y);
+  int64_t lo = (int64_t)blockIdx.y * per;
+  int64_t hi = lo + per < total ? lo + per : total;
+  float s = 0.f, ss = 0.f;
+  for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+    int64_t b = i / HW;
+    int64_t hw = i - b * HW;
+    float v = bf16_to_f32(x[(b * C + c) * HW + hw]);
+    s += v;
+    ss += v * v;
+  }
+  __shared__ float ls[256], lss[256];
+  ls[threadIdx.x] = s;
+  lss[threadIdx.x] = ss;
+  __syncthreads();
+  for (int off = blockDim.x >> 1; off > 0; off >>= 1) {
+    if (threadIdx.x < off) {
+      ls[threadIdx.x] += ls[threadIdx.x + off];
+      lss[threadIdx.x] += lss[threadIdx.x + off];
+    }
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) {
+    atomicAdd(&sum[c], ls[0]);
+    atomicAdd(&sumsq[c], lss[0]);
+  }
+}
+
+// Pass 2: mean/invstd for normalization (biased var) + running-stat update
+// (unbiased var, torch semantics).  One tiny launch over C.
+__global__ void bn_stats_finalize_kernel(const float* __restrict__ sum,
+                                         const float* __restrict__ sumsq,
+                                         float* __restrict__ mean, float* __restrict__ invstd,
+                                         float* __restrict__ running_mean,
+                                         float* __restrict__ running_var, int64_t count,
+                                         float momentum, float eps, int C) {
+  for (int c = blockIdx.x * blockDim.x + threadIdx.x; c < C; c += gridDim.x * blockDim.x) {
+    float mu = sum[c] / (float)count;
+    float var = sumsq[c] / (float)count - mu * mu;
+    var = var > 0.f ? var : 0.f;
+    mean[c] = mu;
+    invstd[c] = rsqrtf(var + eps);
+    if (running_mean != nullptr) {
+      float unbiased = count > 1 ? var * (float)count / (float)(count - 1) : var;
+      running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * mu;
+      running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
+    }
+  }
+}
+
+extern "C" hipError_t launch_bn_stats(const bf16raw* x, float* sum, float* sumsq, float* mean,
+                                      float* invstd, float* running_mean, float* running_var,
+                                      int B, int C, int64_t HW, float momentum, float eps,
+                                      int nsplit, hipStream_t stream) {
+  dim3 grid(C, nsplit);
+  bn_stats_partial_kernel<<<grid, 256, 0, stream>>>(x, sum, sumsq, B, C, HW);
+  HIP_CHECK_LAUNCH();
+  int blocks = (int)ceil_div_i64(C, 256);
+  bn_stats_finalize_kernel<<<blocks, 256, 0, stream>>>(sum, sumsq, mean, invstd, running_mean,
+                                                       running_var, (int64_t)B * HW, momentum,
+                                                       eps, C);
+  HIP_CHECK_LAUNCH();
+  return hipSuccess;
+}
+
+// ---------------------------------------------------------------------------
+// BN apply: y = gamma*(x-mean)*invstd + beta [+ residual] [relu].
+// Elementwise over B*C*HW; residual may be null.
+// ---------------------------------------------------------------------------
+
+__global__ void bn_apply_kernel(const bf16raw* __restrict__ x, const bf16raw* __restrict__ res,
+                                bf16raw* __restrict__ y, const float* __restrict__ mean,
+                                const float* __restrict__ invstd, const float* __restrict__ gamma,
+                                const float* __restrict__ beta, int C, int64_t HW, int64_t total,
+                                int do_relu) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int c = (int)((i / HW) % C);
+    float v = (bf16_to_f32(x[i]) - mean[c]) * invstd[c] * gamma[c] + beta[c];
+    if (res != nullptr) v += bf16_to_f32(res[i]);
+    if (do_relu && v < 0.f) v = 0.f;
+    y[i] = f32_to_bf16(v);
+  }
+}
+
+extern "C" hipError_t launch_bn_apply(const bf16raw* x, const bf16raw* res, bf16raw* y,
+                                      const float* mean, const float* invstd, const float* gamma,
+                                      const float* beta, int C, int64_t HW, int64_t total,
+                                      int do_relu, hipStream_t stream) {
+  int64_t g = ceil_div_i64(total, 256);
+  int grid = (int)(g > 8192 ? 8192 : (g < 1 ? 1 : g));
+  bn_apply_kernel<<<grid, 256, 0, stream>>>(x, res, y, mean, invstd, gamma, beta, C, HW, total,
+                                            do_relu);
+  HIP_CHECK_LAUNCH();
+  return hipSuccess;
+}
+
+// ---------------------------------------------------------------------------
+// BN backward reduce: per channel sum of dy' and dy'*xhat, where
+// dy' = dy * (y>0) when the forward fused a ReLU (y = saved post-relu out).
+// Accumulates straight into dbeta[c] (=Σdy') and dgamma[c] (=Σdy'·xhat) —
+// callers pass pre-zeroed fp32 buffers (possibly flat-bucket grad views).
+// ---------------------------------------------------------------------------
+
+__global__ void bn_bwd_reduce_kernel(const bf16raw* __restrict__ dy,
+                                     const bf16raw* __restrict__ yrelu,
+                                     const bf16raw* __restrict__ x, const float* __restrict__ mean,
+                                     const float* __restrict__ invstd, float* __restrict__ dbeta,
+                                     float* __restrict__ dgamma, int B, int C, int64_t HW) {
+  int c = blockIdx.x;
+  int64_t total = (int64_t)B * HW;
+  int64_t per = ceil_div_i64(total, gridDim.y);
+  int64_t lo = (int64_t)blockIdx.y * per;
+  int64_t hi = lo + per < total ? lo + per : total;
+  float mu = mean[c], is = invstd[c];
+  float sdy = 0.f, sdyx = 0.f;
+  for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+    int64_t b = i / HW;
+    int64_t hw = i - b * HW;
+    int64_t off = (b * C + c) * HW + hw;
+    float g = bf16_to_f32(dy[off]);
+    if (yrelu != nullptr && bf16_to_f32(yrelu[off]) <= 0.f) g = 0.f;
+    float xh = (bf16_to_f32(x[off]) - mu) * is;
+    sdy += g;
+    sdyx += g * xh;
+  }
+  __shared__ float l0[256], l1[256];
+  l0[threadIdx.x] = sdy;
+  l1[threadIdx.x] = sdyx;
+  __syncthreads();
+  for (int off = blockDim.x >> 1; off > 0; off >>= 1) {
+    if (threadIdx.x < off) {
+      l0[threadIdx.x] += l0[threadIdx.x + off];
+      l1[threadIdx.x] += l1[threadIdx.x + off];
+    }
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) {
+    atomicAdd(&dbeta[c], l0[0]);
+    atomicAdd(&dgamma[c], l1[0]);
+  }
+}
+
+extern "C" hipError_t launch_bn_bwd_reduce(const bf16raw* dy, const bf16raw* yrelu,
+                                           const bf16raw* x, const float* mean,
+                                           const float* invstd, float* dbeta, float* dgamma,
+                                           int B, int C, int64_t HW, int nsplit,
+                                           hipStream_t stream) {
+  dim3 grid(C, nsplit);
+  bn_bwd_reduce_kernel<<<grid, 256, 0, stream>>>(dy, yrelu, x, mean, invstd, dbeta, dgamma, B, C,
+                                                 HW);
+  HIP_CHECK_LAUNCH();
+  return hipSuccess;
+}
+
+// dx = gamma*invstd * (dy' - Σdy'/N - xhat * Σ(dy'·xhat)/N)   (training)
+//    = gamma*invstd * dy'                                      (eval stats)
+__global__ void bn_bwd_dx_kernel(const bf16raw* __restrict__ dy, const bf16raw* __restrict__ yrelu,
+                                 const bf16raw* __restrict__ x, const float* __restrict__ mean,
+                                 const float* __restrict__ invstd,
+                                 const float* __restrict__ gamma, const float* __restrict__ dbeta,
+                                 const float* __restrict__ dgamma, bf16raw* __restrict__ dx,
+                                 int C, int64_t HW, int64_t total, float inv_count,
+                                 int train_stats) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int c = (int)((i / HW) % C);
+    float g = bf16_to_f32(dy[i]);
+    if (yrelu != nullptr && bf16_to_f32(yrelu[i]) <= 0.f) g = 0.f;
+    float is = invstd[c];
+    float out;
+    if (train_stats) {
+      float xh = (bf16_to_f32(x[i]) - mean[c]) * is;
+      out = gamma[c] * is * (g - dbeta[c] * inv_count - xh * dgamma[c] * inv_count);
+    } else {
+      out = gamma[c] * is * g;
+    }
+    dx[i] = f32_to_bf16(out);
+  }
+}
+
+extern "C" hipError_t launch_bn_bwd_dx(const bf16raw* dy, const bf16raw* yrelu, const bf16raw* x,
+                                       const float* mean, const float* invstd, const float* gamma,
+                                       const float* dbeta, const float* dgamma, bf16raw* dx,
+                                       int C, int64_t HW, int64_t total, float inv_count,
+                                       int train_stats, hipStream_t stream) {
+  int64_t g = ceil_div_i64(total, 256);
+  int grid = (int)(g > 8192 ? 8192 : (g < 1 ? 1 : g));
+  bn_bwd_dx_kernel<<<grid, 256, 0, stream>>>(dy, yrelu, x, mean, invstd, gamma, dbeta, dgamma, dx,
+                                             C, HW, total, inv_count, train_stats);
+  HIP_CHECK_LAUNCH();
+  return hipSuccess;
+}
+
+// ---------------------------------------------------------------------------
+// Fused residual add + relu: out = relu(a + b).  Backward reuses relu_bwd
+// (elementwise.hip) against the saved output; both branches receive dz.
+// ---------------------------------------------------------------------------
+
+__global__ void add_relu_kernel(const bf16raw* __restrict__ a, const bf16raw* __restrict__ b,
+                                bf16raw* __restrict__ out, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    float v = bf16_to_f32(a[i]) + bf16_to_f32(b[i]);
+    out[i] = f32_to_bf16(v > 0.f ? v : 0.f);
+  }
+}
+
+extern "C" hipError_t launch_add_relu(const bf16raw* a, const bf16raw* b, bf16raw* out, int64_t n,
+                                      hipStream_t stream) {
+  int64_t g = ceil_div_i64(n, 256);
+  int grid = (int)(g > 8192 ? 8192 : (g < 1 ? 1 : g));
+  add_relu_kernel<<<grid, 256, 0, stream>>>(a, b, out, n);
+  HIP_CHECK_LAUNCH();
+  return hipSuccess;
+}
+
+// ---------------------------------------------------------------------------
+// General max_pool2d (overlapping windows, padding) — ResNet stem 3x3/s2/p1.
+// argmax stored as uint8 offset (kh*KW+kw) inside the window; backward is a
+// gather over the <= ceil(KS/S)^2 windows covering each input element (no
+// atomics).
+// ---------------------------------------------------------------------------
+
+__global__ void maxpool_gen_fwd_kernel(const bf16raw* __restrict__ x, bf16raw* __restrict__ y,
+                                       uint8_t* __restrict__ arg, int64_t BC, int H, int W,
+                                       int HO, int WO, int KS, int S, int P) {
+  int64_t total = BC * HO * WO;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int wo = (int)(i % WO);
+    int ho = (int)((i / WO) % HO);
+    int64_t bc = i / ((int64_t)WO * HO);
+    const bf16raw* xp = x + bc * H * W;
+    float best = -3.4e38f;
+    int bestk = 0;
+    for (int kh = 0; kh < KS; ++kh) {
+      int h = ho * S - P + kh;
+      if (h < 0 || h >= H) continue;
+      for (int kw = 0; kw < KS; ++kw) {
+        int w = wo * S - P + kw;
+        if (w < 0 || w >= W) continue;
+        float v = bf16_to_f32(xp[h * W + w]);
+        if (v > best) {
+          best = v;
+          bestk = kh * KS + kw;
+        }
+      }
+    }
+    y[i] = f32_to_bf16(best);
+    arg[i] = (uint8_t)bestk;
+  }
+}
+
+__global__ void maxpool_gen_bwd_kernel(const bf16raw* __restrict__ dy,
+                                       const uint8_t* __restrict__ arg, bf16raw* __restrict__ dx,
+                                       int64_t BC, int H, int W, int HO, int WO, int KS, int S,
+                                       int P) {
+  int64_t total = BC * H * W;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int w = (int)(i % W);
+    int h = (int)((i / W) % H);
+    int64_t bc = i / ((int64_t)W * H);
+    float acc = 0.f;
+    // windows (ho,wo) with ho*S - P <= h < ho*S - P + KS
+    int ho_lo = (h + P - KS + S) / S;  // ceil((h+P-KS+1)/S) for non-negative
+    if (ho_lo < 0) ho_lo = 0;
+    int ho_hi = (h + P) / S;
+    if (ho_hi >= HO) ho_hi = HO - 1;
+    int wo_lo = (w + P - KS + S) / S;
+    if (wo_lo < 0) wo_lo = 0;
+    int wo_hi = (w + P) / S;
+    if (wo_hi >= WO) wo_hi = WO - 1;
+    for (int ho = ho_lo; ho <= ho_hi; ++ho) {
+      int kh = h - (ho * S - P);
+      if (kh < 0 || kh >= KS) continue;
+      for (int wo = wo_lo; wo <= wo_hi; ++wo) {
+        int kw = w - (wo * S - P);
+        if (kw < 0 || kw >= KS) continue;
+        int64_t o = (bc * HO + ho) * WO + wo;
+        if (arg[o] == (uint8_t)(kh * KS + kw)) acc += bf16_to_f32(dy[o]);
+      }
+    }
+    dx[i] = f32_to_bf16(acc);
+  }
+}
+
+extern "C" hipError_t launch_maxpool_gen_fwd(const bf16raw* x, bf16raw* y, uint8_t* arg,
+                                             int64_t BC, int H, int W, int HO, int WO, int KS,
+                                             int S, int P, hipStream_t stream) {
+  int64_t g = ceil_div_i64(BC * HO * WO, 256);
+  int grid = (int)(g > 8192 ? 8192 : (g < 1 ? 1 : g));
+  maxpool_gen_fwd_kernel<<<grid, 256, 0, stream>>>(x, y, arg, BC, H, W, HO, WO, KS, S, P);
+  HIP_CHECK_LAUNCH();
+  return hipSuccess;
+}
+
+extern "C" hipError_t launch_maxpool_gen_bwd(const bf16raw* dy, const uint8_t* arg, bf16raw* dx,
+                                             int64_t BC, int H, int W, int HO, int WO, int KS,
+                                             int S, int P, hipStream_t stream) {
+  int64_t g = ceil_div_i64(BC * H * W, 256);
+  int grid = (int)(g > 8192 ? 8192 : (g < 1 ? 1 : g));
+  maxpool_gen_bwd_kernel<<<grid, 256, 0, stream>>>(dy, arg, dx, BC, H, W, HO, WO, KS, S, P);
+  HIP_CHECK_LAUNCH();
+  return hipSuccess;
+}
+
+// ---------------------------------------------------------------------------
+// Global average pool: [B,C,H,W] -> [B,C].  One block per (b,c), LDS reduce;
+// backward broadcasts dy/HW.
+// ---------------------------------------------------------------------------
+
+__global__ void gap_fwd_kernel(const bf16raw* __restrict__ x, bf16raw* __restrict__ y,
+                               int64_t HW) {
+  int64_t bc = blockIdx.x;
+  const bf16raw* xp = x + bc * HW;
+  float s = 0.f;
+  for (int64_t i = threadIdx.x; i < HW; i += blockDim.x) s += bf16_to_f32(xp[i]);
+  __shared__ float ls[256];
+  ls[threadIdx.x] = s;
+  __syncthreads();
+  for (int off = blockDim.x >> 1; off > 0; off >>= 1) {
+    if (threadIdx.x < off) ls[threadIdx.x] += ls[threadIdx.x + off];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) y[bc] = f32_to_bf16(ls[0] / (float)HW);
+}
+
+__global__ void gap_bwd_kernel(const bf16raw* __restrict__ dy, bf16raw* __restrict__ dx,
+                               int64_t HW, int64_t total) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    dx[i] = f32_to_bf16(bf16_to_f32(dy[i / HW]) / (float)HW);
+  }
+}
+
+extern "C" hipError_t launch_gap_fwd(const bf16raw* x, bf16raw* y, int64_t BC, int64_t HW,
+                                     hipStream_t stream) {
+  gap_fwd_kernel<<<(int)BC, 256, 0, stream>>>(x, y, HW);
+  HIP_CHECK_LAUNCH();
+  return hipSuccess;
+}
+
+extern "C" hipError_t launch_gap_bwd(const bf16raw* dy, bf16raw* dx, int64_t HW, int64_t total,
+                                     hipStream_t stream) {
+  int64_t g = ceil_div_i64(total, 256);
+  int grid = (int)(g > 8192 ? 8192 : (g < 1 ? 1 : g));
+  gap_bwd_kernel<<<grid, 256, 0, stream>>>(dy, dx, HW, total);
+  HIP_CHECK_LAUNCH();
+  return hipSuccess;
+}

@@ -168,6 +168,138 @@ def hip_conv2d(x, weight, bias=None, stride=(1, 1), padding=(0, 0), relu=False):
     return F.relu(y) if relu else y
 
 
+class _BatchNorm2dFn(torch.autograd.Function):
+    """BatchNorm2d on the native kernels: training stats (channel×slice grid
+    + atomics, fp32), normalize+affine with optional fused ReLU; backward is
+    one reduce kernel (dbeta/dgamma, accumulated into pre-zeroed fp32 buffers
+    — flat-bucket grad views on the direct-grad path, like linear_wgrad_into)
+    + one elementwise dx kernel.  Replaces the reference's eager CPU
+    nn.BatchNorm2d (reference has no BN kernels)."""
+
+    @staticmethod
+    def forward(ctx, x, gamma, beta, running_mean, running_var, training, momentum, eps, relu):
+        ext = ops.ext()
+        if training:
+            mean, invstd = ext.bn_stats(x, running_mean, running_var, momentum, eps)
+        else:
+            mean = running_mean.contiguous()
+            invstd = (running_var + eps).rsqrt().contiguous()
+        y = ext.bn_apply(x, None, mean, invstd, gamma, beta, relu)
+        ctx.save_for_backward(x, y if relu else None, mean, invstd, gamma)
+        ctx.relu = relu
+        ctx.train_stats = training
+        ctx.gamma_ref, ctx.beta_ref = gamma, beta
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = ops.ext()
+        x, yrelu, mean, invstd, gamma = ctx.saved_tensors
+        dy = dy.contiguous()
+        if dy.dtype != torch.bfloat16:
+            dy = dy.to(torch.bfloat16)
+
+        g, b = ctx.gamma_ref, ctx.beta_ref
+        g_notify = getattr(g, "_bucket_notify", None)
+        b_notify = getattr(b, "_bucket_notify", None)
+        direct = (
+            g_notify is not None and g.grad is not None
+            and b_notify is not None and b.grad is not None
+        )
+        if direct:
+            dgamma_buf, dbeta_buf = g.grad, b.grad
+        else:
+            dgamma_buf = torch.zeros_like(mean)
+            dbeta_buf = torch.zeros_like(mean)
+        ext.bn_bwd_reduce(dy, yrelu, x, mean, invstd, dbeta_buf, dgamma_buf)
+        dx = None
+        if ctx.needs_input_grad[0]:
+            dx = ext.bn_bwd_dx(dy, yrelu, x, mean, invstd, gamma, dbeta_buf, dgamma_buf,
+                               ctx.train_stats)
+        if direct:
+            g_notify()
+            b_notify()
+            return dx, None, None, None, None, None, None, None, None
+        return dx, dgamma_buf, dbeta_buf, None, None, None, None, None, None
+
+
+def hip_batch_norm2d(x, gamma, beta, running_mean, running_var, training: bool,
+                     momentum: float = 0.1, eps: float = 1e-5, relu: bool = False):
+    if x.is_cuda:
+        if x.dtype != torch.bfloat16:
+            x = x.to(torch.bfloat16)
+        return _BatchNorm2dFn.apply(x.contiguous(), gamma, beta, running_mean, running_var,
+                                    training, momentum, eps, relu)
+    y = F.batch_norm(x, running_mean, running_var, gamma, beta, training, momentum, eps)
+    return F.relu(y) if relu else y
+
+
+class _AddReluFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, a, b):
+        out = ops.ext().add_relu(a, b)
+        ctx.save_for_backward(out)
+        return out
+
+    @staticmethod
+    def backward(ctx, dy):
+        (out,) = ctx.saved_tensors
+        dz = ops.ext().relu_bwd(dy.contiguous().to(torch.bfloat16), out)
+        return dz, dz
+
+
+def hip_add_relu(a, b):
+    """Residual join: relu(a + b), one fused elementwise kernel."""
+    if a.is_cuda:
+        if a.dtype != torch.bfloat16:
+            a = a.to(torch.bfloat16)
+        if b.dtype != torch.bfloat16:
+            b = b.to(torch.bfloat16)
+        return _AddReluFn.apply(a.contiguous(), b.contiguous())
+    return F.relu(a + b)
+
+
+class _MaxPool2dGenFn(torch.autograd.Function):
+    """Overlapping-window max_pool2d (e.g. ResNet stem 3x3/s2/p1); backward
+    gathers over covering windows — no atomics."""
+
+    @staticmethod
+    def forward(ctx, x, ks, stride, pad):
+        y, arg = ops.ext().maxpool_gen_fwd(x, ks, stride, pad)
+        ctx.save_for_backward(arg)
+        ctx.meta = (x.shape[2], x.shape[3], ks, stride, pad)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (arg,) = ctx.saved_tensors
+        H, W, ks, stride, pad = ctx.meta
+        dx = ops.ext().maxpool_gen_bwd(dy.contiguous().to(torch.bfloat16), arg, H, W, ks,
+                                       stride, pad)
+        return dx, None, None, None
+
+
+class _GlobalAvgPoolFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        ctx.meta = (x.shape[2], x.shape[3])
+        return ops.ext().gap_fwd(x)
+
+    @staticmethod
+    def backward(ctx, dy):
+        H, W = ctx.meta
+        return ops.ext().gap_bwd(dy.contiguous().to(torch.bfloat16), H, W)
+
+
+def hip_global_avg_pool(x):
+    """[B,C,H,W] -> [B,C] channel means (adaptive_avg_pool2d(x,1).flatten(1))."""
+    if x.is_cuda:
+        if x.dtype != torch.bfloat16:
+            x = x.to(torch.bfloat16)
+        return _GlobalAvgPoolFn.apply(x.contiguous())
+    return F.adaptive_avg_pool2d(x, 1).flatten(1)
+
+
 class _MaxPool2dFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, ks):
@@ -183,12 +315,17 @@ class _MaxPool2dFn(torch.autograd.Function):
         return ops.ext().maxpool_bwd(dy.contiguous().to(torch.bfloat16), arg, H, W, ks), None
 
 
-def hip_max_pool2d(x, kernel_size: int):
+def hip_max_pool2d(x, kernel_size: int, stride: Optional[int] = None, padding: int = 0):
+    if stride is None:
+        stride = kernel_size
     if x.is_cuda:
         if x.dtype != torch.bfloat16:
             x = x.to(torch.bfloat16)
-        return _MaxPool2dFn.apply(x.contiguous(), kernel_size)
-    return F.max_pool2d(x, kernel_size)
+        x = x.contiguous()
+        if stride == kernel_size and padding == 0:
+            return _MaxPool2dFn.apply(x, kernel_size)  # disjoint fast path
+        return _MaxPool2dGenFn.apply(x, kernel_size, stride, padding)
+    return F.max_pool2d(x, kernel_size, stride=stride, padding=padding)
 
 
 class _DropoutFn(torch.autograd.Function):

@@ -16,9 +16,12 @@ import torch
 import torch.nn as nn
 
 from sparktorch_amd.ops.functional import (
+    hip_add_relu,
+    hip_batch_norm2d,
     hip_conv2d,
     hip_cross_entropy,
     hip_dropout,
+    hip_global_avg_pool,
     hip_linear,
     hip_max_pool2d,
     hip_mse,
@@ -91,12 +94,67 @@ class HipConv2d(nn.Module):
 
 
 class HipMaxPool2d(nn.Module):
-    def __init__(self, kernel_size: int):
+    def __init__(self, kernel_size: int, stride: Optional[int] = None, padding: int = 0):
         super().__init__()
         self.kernel_size = kernel_size
+        self.stride = kernel_size if stride is None else stride
+        self.padding = padding
 
     def forward(self, x):
-        return hip_max_pool2d(x, self.kernel_size)
+        return hip_max_pool2d(x, self.kernel_size, self.stride, self.padding)
+
+
+class HipBatchNorm2d(nn.Module):
+    """Drop-in nn.BatchNorm2d on the native kernels (affine, running stats),
+    with optional fused ReLU.  state_dict-compatible with nn.BatchNorm2d."""
+
+    def __init__(self, num_features: int, eps: float = 1e-5, momentum: float = 0.1,
+                 activation: Optional[str] = None):
+        super().__init__()
+        self.num_features = num_features
+        self.eps = eps
+        self.momentum = momentum
+        self.activation = activation
+        self.weight = nn.Parameter(torch.ones(num_features))
+        self.bias = nn.Parameter(torch.zeros(num_features))
+        self.register_buffer("running_mean", torch.zeros(num_features))
+        self.register_buffer("running_var", torch.ones(num_features))
+        self.register_buffer("num_batches_tracked", torch.tensor(0, dtype=torch.long))
+
+    @classmethod
+    def from_batchnorm(cls, bn: nn.BatchNorm2d, activation: Optional[str] = None) -> "HipBatchNorm2d":
+        if not bn.affine or not bn.track_running_stats:
+            raise ValueError("HipBatchNorm2d needs affine=True, track_running_stats=True")
+        m = cls.__new__(cls)
+        nn.Module.__init__(m)
+        m.num_features = bn.num_features
+        m.eps = bn.eps
+        m.momentum = bn.momentum if bn.momentum is not None else 0.1
+        m.activation = activation
+        m.weight = bn.weight  # SAME Parameter objects
+        m.bias = bn.bias
+        m.register_buffer("running_mean", bn.running_mean)
+        m.register_buffer("running_var", bn.running_var)
+        m.register_buffer("num_batches_tracked", bn.num_batches_tracked)
+        return m
+
+    def forward(self, x):
+        if self.training:
+            self.num_batches_tracked += 1
+        return hip_batch_norm2d(
+            x, self.weight, self.bias, self.running_mean, self.running_var,
+            self.training, self.momentum, self.eps, relu=self.activation == "relu",
+        )
+
+    def extra_repr(self) -> str:
+        return "%d, eps=%g, act=%s" % (self.num_features, self.eps, self.activation)
+
+
+class HipGlobalAvgPool(nn.Module):
+    """adaptive_avg_pool2d(x, 1).flatten(1) as one reduction kernel."""
+
+    def forward(self, x):
+        return hip_global_avg_pool(x)
 
 
 class HipDropout(nn.Module):
@@ -131,6 +189,65 @@ class MnistCNNFused(nn.Module):
         x = self.dropout(x)
         x = torch.flatten(x, 1)
         return self.fc(x)
+
+
+class FusedBasicBlock(nn.Module):
+    """ResNet basic block on the native path: implicit-GEMM convs, BN with
+    fused ReLU, and a fused residual add+relu join.  state_dict-compatible
+    with models.resnet.BasicBlock."""
+
+    def __init__(self, in_ch: int, out_ch: int, stride: int = 1):
+        super().__init__()
+        self.conv1 = HipConv2d(in_ch, out_ch, 3, stride=stride, padding=1, bias=False)
+        self.bn1 = HipBatchNorm2d(out_ch, activation="relu")
+        self.conv2 = HipConv2d(out_ch, out_ch, 3, stride=1, padding=1, bias=False)
+        self.bn2 = HipBatchNorm2d(out_ch)
+        self.down = None
+        if stride != 1 or in_ch != out_ch:
+            self.down = nn.Sequential(
+                HipConv2d(in_ch, out_ch, 1, stride=stride, bias=False),
+                HipBatchNorm2d(out_ch),
+            )
+
+    def forward(self, x):
+        identity = x if self.down is None else self.down(x)
+        out = self.bn1(self.conv1(x))          # BN+ReLU fused
+        out = self.conv2(out)
+        out = self.bn2(out)
+        return hip_add_relu(out, identity)     # residual join fused
+
+
+class ResNet18Fused(nn.Module):
+    """ResNet-18 (BASELINE config 4: synthetic 3x224x224 Vector rows) fully
+    on the hand-written CDNA4 kernels: 7x7/s2 stem conv, overlapping
+    3x3/s2/p1 maxpool, 8 basic blocks, global-avg-pool reduction, MFMA fc.
+    state_dict-compatible with models.resnet.ResNet18."""
+
+    def __init__(self, num_classes: int = 1000, in_ch: int = 3):
+        super().__init__()
+        self.conv1 = HipConv2d(in_ch, 64, 7, stride=2, padding=3, bias=False)
+        self.bn1 = HipBatchNorm2d(64, activation="relu")
+        self.maxpool = HipMaxPool2d(3, stride=2, padding=1)
+        layers = []
+        cfg = [(64, 1), (128, 2), (256, 2), (512, 2)]
+        ch = 64
+        for out_ch, stride in cfg:
+            layers.append(FusedBasicBlock(ch, out_ch, stride))
+            layers.append(FusedBasicBlock(out_ch, out_ch, 1))
+            ch = out_ch
+        self.layers = nn.Sequential(*layers)
+        self.gap = HipGlobalAvgPool()
+        self.fc = HipLinear(512, num_classes)
+
+    def forward(self, x):
+        if x.is_cuda and x.dtype != torch.bfloat16:
+            x = x.to(torch.bfloat16)
+        if x.dim() == 2:  # flattened Vector rows, like the CNN unflatten idiom
+            x = x.view(-1, 3, 224, 224)
+        x = self.bn1(self.conv1(x))
+        x = self.maxpool(x)
+        x = self.layers(x)
+        return self.fc(self.gap(x))
 
 
 class HipCrossEntropy(nn.Module):
@@ -173,10 +290,19 @@ def convert_model_for_mi355x(model: nn.Module) -> nn.Module:
                 setattr(model, name, HipConv2d.from_conv2d(child))
             except (ValueError, AttributeError):
                 pass
-        elif cname == "MaxPool2d" and getattr(child, "kernel_size", None) == getattr(child, "stride", 0):
-            ks = child.kernel_size
-            if isinstance(ks, int):
-                setattr(model, name, HipMaxPool2d(ks))
+        elif cname == "MaxPool2d" or isinstance(child, nn.MaxPool2d):
+            ks, st, pd = child.kernel_size, child.stride, child.padding
+            if isinstance(ks, int) and isinstance(st, (int, type(None))) and isinstance(pd, int) \
+                    and child.dilation == 1 and not child.ceil_mode:
+                setattr(model, name, HipMaxPool2d(ks, stride=st, padding=pd))
+        elif cname == "BatchNorm2d" or isinstance(child, nn.BatchNorm2d):
+            try:
+                setattr(model, name, HipBatchNorm2d.from_batchnorm(child))
+            except (ValueError, AttributeError):
+                pass
+        elif cname == "AdaptiveAvgPool2d" and child.output_size in (1, (1, 1)):
+            # NB: produces [B,C] (flattened) — matches the F.adaptive_avg_pool2d(x,1).flatten(1) idiom
+            setattr(model, name, HipGlobalAvgPool())
         elif cname == "Dropout2d":
             setattr(model, name, HipDropout(p=child.p, channel_wise=True))
         elif cname == "Dropout":
