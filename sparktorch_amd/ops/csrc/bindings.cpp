@@ -44,6 +44,27 @@ hipError_t launch_maxpool_gen_bwd(const bf16raw*, const uint8_t*, bf16raw*, int6
                                   int, int, int, int, int, hipStream_t);
 hipError_t launch_gap_fwd(const bf16raw*, bf16raw*, int64_t, int64_t, hipStream_t);
 hipError_t launch_gap_bwd(const bf16raw*, bf16raw*, int64_t, int64_t, hipStream_t);
+hipError_t launch_im2col_nhwc(const bf16raw*, bf16raw*, int, int, int, int, int, int, int, int,
+                              int, int, int, int, hipStream_t);
+hipError_t launch_col2im_nhwc(const bf16raw*, bf16raw*, int, int, int, int, int, int, int, int,
+                              int, int, int, int, hipStream_t);
+hipError_t launch_maxpool_nhwc_fwd(const bf16raw*, bf16raw*, uint8_t*, int, int, int, int, int,
+                                   int, int, int, int, hipStream_t);
+hipError_t launch_maxpool_nhwc_bwd(const bf16raw*, const uint8_t*, bf16raw*, int, int, int, int,
+                                   int, int, int, int, int, hipStream_t);
+hipError_t launch_gap_nhwc_fwd(const bf16raw*, bf16raw*, int, int, int64_t, hipStream_t);
+hipError_t launch_gap_nhwc_bwd(const bf16raw*, bf16raw*, int, int, int64_t, hipStream_t);
+hipError_t launch_bn_stats_nhwc(const bf16raw*, float*, float*, float*, float*, float*, float*,
+                                int64_t, int, float, float, int, hipStream_t);
+hipError_t launch_bn_apply_nhwc(const bf16raw*, const bf16raw*, bf16raw*, const float*,
+                                const float*, const float*, const float*, int, int64_t, int,
+                                hipStream_t);
+hipError_t launch_bn_bwd_reduce_nhwc(const bf16raw*, const bf16raw*, const bf16raw*,
+                                     const float*, const float*, float*, float*, int64_t, int,
+                                     int, hipStream_t);
+hipError_t launch_bn_bwd_dx_nhwc(const bf16raw*, const bf16raw*, const bf16raw*, const float*,
+                                 const float*, const float*, const float*, const float*,
+                                 bf16raw*, int, int64_t, float, int, hipStream_t);
 }
 
 #define CHECK_HIP(err)                                                              \
@@ -478,6 +499,169 @@ at::Tensor gap_bwd(at::Tensor dy, int64_t H, int64_t W) {
   return dx;
 }
 
+// ------------------------------- NHWC ops ----------------------------------
+// x is [B, H, W, C] contiguous; col matrices are [B*HO*WO, KH*KW*CI] with
+// (kh, kw, ci) fastest — matching w.permute(0,2,3,1).reshape(CO, K).
+
+at::Tensor im2col_nhwc(at::Tensor x, int64_t KH, int64_t KW, int64_t sh, int64_t sw, int64_t ph,
+                       int64_t pw) {
+  check_gpu_contig(x, at::kBFloat16, "x");
+  TORCH_CHECK(x.dim() == 4, "x must be BHWC");
+  int B = (int)x.size(0), H = (int)x.size(1), W = (int)x.size(2), CI = (int)x.size(3);
+  int HO = (int)((H + 2 * ph - KH) / sh + 1);
+  int WO = (int)((W + 2 * pw - KW) / sw + 1);
+  auto col = at::empty({(int64_t)B * HO * WO, KH * KW * CI}, x.options());
+  CHECK_HIP(launch_im2col_nhwc((const bf16raw*)x.data_ptr(), (bf16raw*)col.data_ptr(), B, CI, H,
+                               W, (int)KH, (int)KW, HO, WO, (int)sh, (int)sw, (int)ph, (int)pw,
+                               cur_stream()));
+  return col;
+}
+
+at::Tensor col2im_nhwc(at::Tensor dcol, int64_t B, int64_t CI, int64_t H, int64_t W, int64_t KH,
+                       int64_t KW, int64_t sh, int64_t sw, int64_t ph, int64_t pw) {
+  check_gpu_contig(dcol, at::kBFloat16, "dcol");
+  int HO = (int)((H + 2 * ph - KH) / sh + 1);
+  int WO = (int)((W + 2 * pw - KW) / sw + 1);
+  auto dx = at::empty({B, H, W, CI}, dcol.options());
+  CHECK_HIP(launch_col2im_nhwc((const bf16raw*)dcol.data_ptr(), (bf16raw*)dx.data_ptr(), (int)B,
+                               (int)CI, (int)H, (int)W, (int)KH, (int)KW, HO, WO, (int)sh,
+                               (int)sw, (int)ph, (int)pw, cur_stream()));
+  return dx;
+}
+
+std::tuple<at::Tensor, at::Tensor> maxpool_nhwc_fwd(at::Tensor x, int64_t ks, int64_t stride,
+                                                    int64_t pad) {
+  check_gpu_contig(x, at::kBFloat16, "x");
+  TORCH_CHECK(x.dim() == 4, "x must be BHWC");
+  TORCH_CHECK(ks <= 15, "kernel too large for uint8 argmax");
+  int B = (int)x.size(0), H = (int)x.size(1), W = (int)x.size(2), C = (int)x.size(3);
+  int HO = (int)((H + 2 * pad - ks) / stride + 1);
+  int WO = (int)((W + 2 * pad - ks) / stride + 1);
+  auto y = at::empty({B, HO, WO, C}, x.options());
+  auto arg = at::empty({B, HO, WO, C}, x.options().dtype(at::kByte));
+  CHECK_HIP(launch_maxpool_nhwc_fwd((const bf16raw*)x.data_ptr(), (bf16raw*)y.data_ptr(),
+                                    arg.data_ptr<uint8_t>(), B, C, H, W, HO, WO, (int)ks,
+                                    (int)stride, (int)pad, cur_stream()));
+  return {y, arg};
+}
+
+at::Tensor maxpool_nhwc_bwd(at::Tensor dy, at::Tensor arg, int64_t H, int64_t W, int64_t ks,
+                            int64_t stride, int64_t pad) {
+  check_gpu_contig(dy, at::kBFloat16, "dy");
+  check_gpu_contig(arg, at::kByte, "arg");
+  int B = (int)dy.size(0), HO = (int)dy.size(1), WO = (int)dy.size(2), C = (int)dy.size(3);
+  auto dx = at::empty({B, H, W, (int64_t)C}, dy.options());
+  CHECK_HIP(launch_maxpool_nhwc_bwd((const bf16raw*)dy.data_ptr(), arg.data_ptr<uint8_t>(),
+                                    (bf16raw*)dx.data_ptr(), B, C, (int)H, (int)W, HO, WO,
+                                    (int)ks, (int)stride, (int)pad, cur_stream()));
+  return dx;
+}
+
+at::Tensor gap_nhwc_fwd(at::Tensor x) {
+  check_gpu_contig(x, at::kBFloat16, "x");
+  TORCH_CHECK(x.dim() == 4, "x must be BHWC");
+  int B = (int)x.size(0), C = (int)x.size(3);
+  auto y = at::empty({B, C}, x.options());
+  CHECK_HIP(launch_gap_nhwc_fwd((const bf16raw*)x.data_ptr(), (bf16raw*)y.data_ptr(), B, C,
+                                x.size(1) * x.size(2), cur_stream()));
+  return y;
+}
+
+at::Tensor gap_nhwc_bwd(at::Tensor dy, int64_t H, int64_t W) {
+  check_gpu_contig(dy, at::kBFloat16, "dy");
+  int B = (int)dy.size(0), C = (int)dy.size(1);
+  auto dx = at::empty({B, H, W, (int64_t)C}, dy.options());
+  CHECK_HIP(launch_gap_nhwc_bwd((const bf16raw*)dy.data_ptr(), (bf16raw*)dx.data_ptr(), B, C,
+                                H * W, cur_stream()));
+  return dx;
+}
+
+std::tuple<at::Tensor, at::Tensor> bn_stats_nhwc(at::Tensor x,
+                                                 c10::optional<at::Tensor> running_mean,
+                                                 c10::optional<at::Tensor> running_var,
+                                                 double momentum, double eps) {
+  check_gpu_contig(x, at::kBFloat16, "x");
+  TORCH_CHECK(x.dim() == 4, "x must be BHWC");
+  int C = (int)x.size(3);
+  int64_t M = x.numel() / C;
+  auto fopt = x.options().dtype(at::kFloat);
+  auto sum = at::zeros({C}, fopt), sumsq = at::zeros({C}, fopt);
+  auto mean = at::empty({C}, fopt), invstd = at::empty({C}, fopt);
+  float* rm = nullptr;
+  float* rv = nullptr;
+  if (running_mean.has_value()) {
+    check_gpu_contig(*running_mean, at::kFloat, "running_mean");
+    check_gpu_contig(*running_var, at::kFloat, "running_var");
+    rm = running_mean->data_ptr<float>();
+    rv = running_var->data_ptr<float>();
+  }
+  int tiles = (C + 63) / 64;
+  int nsplit = 2048 / (tiles > 0 ? tiles : 1);
+  nsplit = nsplit < 1 ? 1 : (nsplit > 256 ? 256 : nsplit);
+  CHECK_HIP(launch_bn_stats_nhwc((const bf16raw*)x.data_ptr(), sum.data_ptr<float>(),
+                                 sumsq.data_ptr<float>(), mean.data_ptr<float>(),
+                                 invstd.data_ptr<float>(), rm, rv, M, C, (float)momentum,
+                                 (float)eps, nsplit, cur_stream()));
+  return {mean, invstd};
+}
+
+at::Tensor bn_apply_nhwc(at::Tensor x, c10::optional<at::Tensor> res, at::Tensor mean,
+                         at::Tensor invstd, at::Tensor gamma, at::Tensor beta, bool relu) {
+  check_gpu_contig(x, at::kBFloat16, "x");
+  TORCH_CHECK(x.dim() == 4, "x must be BHWC");
+  const bf16raw* rp = nullptr;
+  if (res.has_value()) {
+    check_gpu_contig(*res, at::kBFloat16, "res");
+    TORCH_CHECK(res->sizes() == x.sizes(), "residual shape mismatch");
+    rp = (const bf16raw*)res->data_ptr();
+  }
+  auto y = at::empty_like(x);
+  CHECK_HIP(launch_bn_apply_nhwc((const bf16raw*)x.data_ptr(), rp, (bf16raw*)y.data_ptr(),
+                                 mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                                 gamma.data_ptr<float>(), beta.data_ptr<float>(),
+                                 (int)x.size(3), x.numel(), relu ? 1 : 0, cur_stream()));
+  return y;
+}
+
+void bn_bwd_reduce_nhwc(at::Tensor dy, c10::optional<at::Tensor> yrelu, at::Tensor x,
+                        at::Tensor mean, at::Tensor invstd, at::Tensor dbeta,
+                        at::Tensor dgamma) {
+  check_gpu_contig(dy, at::kBFloat16, "dy");
+  check_gpu_contig(x, at::kBFloat16, "x");
+  check_gpu_contig(dbeta, at::kFloat, "dbeta");
+  check_gpu_contig(dgamma, at::kFloat, "dgamma");
+  const bf16raw* yp = nullptr;
+  if (yrelu.has_value()) yp = (const bf16raw*)yrelu->data_ptr();
+  int C = (int)x.size(3);
+  int64_t M = x.numel() / C;
+  int tiles = (C + 63) / 64;
+  int nsplit = 2048 / (tiles > 0 ? tiles : 1);
+  nsplit = nsplit < 1 ? 1 : (nsplit > 256 ? 256 : nsplit);
+  CHECK_HIP(launch_bn_bwd_reduce_nhwc((const bf16raw*)dy.data_ptr(), yp,
+                                      (const bf16raw*)x.data_ptr(), mean.data_ptr<float>(),
+                                      invstd.data_ptr<float>(), dbeta.data_ptr<float>(),
+                                      dgamma.data_ptr<float>(), M, C, nsplit, cur_stream()));
+}
+
+at::Tensor bn_bwd_dx_nhwc(at::Tensor dy, c10::optional<at::Tensor> yrelu, at::Tensor x,
+                          at::Tensor mean, at::Tensor invstd, at::Tensor gamma, at::Tensor dbeta,
+                          at::Tensor dgamma, bool train_stats) {
+  check_gpu_contig(dy, at::kBFloat16, "dy");
+  check_gpu_contig(x, at::kBFloat16, "x");
+  const bf16raw* yp = nullptr;
+  if (yrelu.has_value()) yp = (const bf16raw*)yrelu->data_ptr();
+  int C = (int)x.size(3);
+  auto dx = at::empty_like(x);
+  float inv_count = 1.0f / (float)(x.numel() / C);
+  CHECK_HIP(launch_bn_bwd_dx_nhwc((const bf16raw*)dy.data_ptr(), yp,
+                                  (const bf16raw*)x.data_ptr(), mean.data_ptr<float>(),
+                                  invstd.data_ptr<float>(), gamma.data_ptr<float>(),
+                                  dbeta.data_ptr<float>(), dgamma.data_ptr<float>(),
+                                  (bf16raw*)dx.data_ptr(), C, x.numel(), inv_count,
+                                  train_stats ? 1 : 0, cur_stream()));
+  return dx;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_adam", &fused_adam, "fused Adam on a flat bucket");
   m.def("fused_sgd", &fused_sgd, "fused SGD on a flat bucket");
@@ -508,4 +692,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("maxpool_gen_bwd", &maxpool_gen_bwd, "general max_pool2d backward (gather)");
   m.def("gap_fwd", &gap_fwd, "global average pool [B,C,H,W]->[B,C]");
   m.def("gap_bwd", &gap_bwd, "global average pool backward");
+  m.def("im2col_nhwc", &im2col_nhwc, "BHWC -> implicit-GEMM col (vectorized over C)");
+  m.def("col2im_nhwc", &col2im_nhwc, "col gradient -> BHWC input gradient");
+  m.def("maxpool_nhwc_fwd", &maxpool_nhwc_fwd, "NHWC max_pool2d (stride/pad) + argmax");
+  m.def("maxpool_nhwc_bwd", &maxpool_nhwc_bwd, "NHWC max_pool2d backward (gather)");
+  m.def("gap_nhwc_fwd", &gap_nhwc_fwd, "NHWC global average pool -> [B,C]");
+  m.def("gap_nhwc_bwd", &gap_nhwc_bwd, "NHWC global average pool backward");
+  m.def("bn_stats_nhwc", &bn_stats_nhwc, "NHWC BN training stats + running update");
+  m.def("bn_apply_nhwc", &bn_apply_nhwc, "NHWC BN normalize+affine (+residual)(+relu)");
+  m.def("bn_bwd_reduce_nhwc", &bn_bwd_reduce_nhwc, "NHWC BN backward reduce (in-place)");
+  m.def("bn_bwd_dx_nhwc", &bn_bwd_dx_nhwc, "NHWC BN backward input gradient");
 }

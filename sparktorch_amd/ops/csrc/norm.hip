@@ -88,6 +88,175 @@ extern "C" hipError_t launch_bn_stats(const bf16raw* x, float* sum, float* sumsq
 }
 
 // ---------------------------------------------------------------------------
+// NHWC BatchNorm statistics: x viewed as [M, C] (M = B*H*W).  Block =
+// 64-channel tile x 4 row-groups (coalesced 128-byte row reads); grid
+// (ceil(C/64), S) with one atomicAdd per channel per block.
+// ---------------------------------------------------------------------------
+
+__global__ void bn_stats_partial_nhwc_kernel(const bf16raw* __restrict__ x,
+                                             float* __restrict__ sum, float* __restrict__ sumsq,
+                                             int64_t M, int C) {
+  int lane_c = threadIdx.x & 63;
+  int row_g = threadIdx.x >> 6;  // 0..3
+  int c = blockIdx.x * 64 + lane_c;
+  int64_t per = ceil_div_i64(M, gridDim.y);
+  int64_t lo = (int64_t)blockIdx.y * per;
+  int64_t hi = lo + per < M ? lo + per : M;
+  float s = 0.f, ss = 0.f;
+  if (c < C) {
+    for (int64_t r = lo + row_g; r < hi; r += 4) {
+      float v = bf16_to_f32(x[r * C + c]);
+      s += v;
+      ss += v * v;
+    }
+  }
+  __shared__ float ls[256], lss[256];
+  ls[threadIdx.x] = s;
+  lss[threadIdx.x] = ss;
+  __syncthreads();
+  if (row_g == 0 && c < C) {
+    s = ls[lane_c] + ls[lane_c + 64] + ls[lane_c + 128] + ls[lane_c + 192];
+    ss = lss[lane_c] + lss[lane_c + 64] + lss[lane_c + 128] + lss[lane_c + 192];
+    atomicAdd(&sum[c], s);
+    atomicAdd(&sumsq[c], ss);
+  }
+}
+
+extern "C" hipError_t launch_bn_stats_nhwc(const bf16raw* x, float* sum, float* sumsq,
+                                           float* mean, float* invstd, float* running_mean,
+                                           float* running_var, int64_t M, int C, float momentum,
+                                           float eps, int nsplit, hipStream_t stream) {
+  dim3 grid((C + 63) / 64, nsplit);
+  bn_stats_partial_nhwc_kernel<<<grid, 256, 0, stream>>>(x, sum, sumsq, M, C);
+  HIP_CHECK_LAUNCH();
+  int blocks = (int)ceil_div_i64(C, 256);
+  bn_stats_finalize_kernel<<<blocks, 256, 0, stream>>>(sum, sumsq, mean, invstd, running_mean,
+                                                       running_var, M, momentum, eps, C);
+  HIP_CHECK_LAUNCH();
+  return hipSuccess;
+}
+
+// NHWC apply: elementwise, c = i % C (fully coalesced).
+__global__ void bn_apply_nhwc_kernel(const bf16raw* __restrict__ x,
+                                     const bf16raw* __restrict__ res, bf16raw* __restrict__ y,
+                                     const float* __restrict__ mean,
+                                     const float* __restrict__ invstd,
+                                     const float* __restrict__ gamma,
+                                     const float* __restrict__ beta, int C, int64_t total,
+                                     int do_relu) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int c = (int)(i % C);
+    float v = (bf16_to_f32(x[i]) - mean[c]) * invstd[c] * gamma[c] + beta[c];
+    if (res != nullptr) v += bf16_to_f32(res[i]);
+    if (do_relu && v < 0.f) v = 0.f;
+    y[i] = f32_to_bf16(v);
+  }
+}
+
+extern "C" hipError_t launch_bn_apply_nhwc(const bf16raw* x, const bf16raw* res, bf16raw* y,
+                                           const float* mean, const float* invstd,
+                                           const float* gamma, const float* beta, int C,
+                                           int64_t total, int do_relu, hipStream_t stream) {
+  int64_t g = ceil_div_i64(total, 256);
+  int grid = (int)(g > 8192 ? 8192 : (g < 1 ? 1 : g));
+  bn_apply_nhwc_kernel<<<grid, 256, 0, stream>>>(x, res, y, mean, invstd, gamma, beta, C, total,
+                                                 do_relu);
+  HIP_CHECK_LAUNCH();
+  return hipSuccess;
+}
+
+// NHWC backward reduce: per-channel Σdy' and Σdy'*xhat (same tile scheme as
+// stats; dy' = dy masked by saved post-relu output when fused).
+__global__ void bn_bwd_reduce_nhwc_kernel(const bf16raw* __restrict__ dy,
+                                          const bf16raw* __restrict__ yrelu,
+                                          const bf16raw* __restrict__ x,
+                                          const float* __restrict__ mean,
+                                          const float* __restrict__ invstd,
+                                          float* __restrict__ dbeta, float* __restrict__ dgamma,
+                                          int64_t M, int C) {
+  int lane_c = threadIdx.x & 63;
+  int row_g = threadIdx.x >> 6;
+  int c = blockIdx.x * 64 + lane_c;
+  int64_t per = ceil_div_i64(M, gridDim.y);
+  int64_t lo = (int64_t)blockIdx.y * per;
+  int64_t hi = lo + per < M ? lo + per : M;
+  float sdy = 0.f, sdyx = 0.f;
+  if (c < C) {
+    float mu = mean[c], is = invstd[c];
+    for (int64_t r = lo + row_g; r < hi; r += 4) {
+      int64_t off = r * C + c;
+      float g = bf16_to_f32(dy[off]);
+      if (yrelu != nullptr && bf16_to_f32(yrelu[off]) <= 0.f) g = 0.f;
+      sdy += g;
+      sdyx += g * (bf16_to_f32(x[off]) - mu) * is;
+    }
+  }
+  __shared__ float l0[256], l1[256];
+  l0[threadIdx.x] = sdy;
+  l1[threadIdx.x] = sdyx;
+  __syncthreads();
+  if (row_g == 0 && c < C) {
+    sdy = l0[lane_c] + l0[lane_c + 64] + l0[lane_c + 128] + l0[lane_c + 192];
+    sdyx = l1[lane_c] + l1[lane_c + 64] + l1[lane_c + 128] + l1[lane_c + 192];
+    atomicAdd(&dbeta[c], sdy);
+    atomicAdd(&dgamma[c], sdyx);
+  }
+}
+
+extern "C" hipError_t launch_bn_bwd_reduce_nhwc(const bf16raw* dy, const bf16raw* yrelu,
+                                                const bf16raw* x, const float* mean,
+                                                const float* invstd, float* dbeta,
+                                                float* dgamma, int64_t M, int C, int nsplit,
+                                                hipStream_t stream) {
+  dim3 grid((C + 63) / 64, nsplit);
+  bn_bwd_reduce_nhwc_kernel<<<grid, 256, 0, stream>>>(dy, yrelu, x, mean, invstd, dbeta, dgamma,
+                                                      M, C);
+  HIP_CHECK_LAUNCH();
+  return hipSuccess;
+}
+
+__global__ void bn_bwd_dx_nhwc_kernel(const bf16raw* __restrict__ dy,
+                                      const bf16raw* __restrict__ yrelu,
+                                      const bf16raw* __restrict__ x,
+                                      const float* __restrict__ mean,
+                                      const float* __restrict__ invstd,
+                                      const float* __restrict__ gamma,
+                                      const float* __restrict__ dbeta,
+                                      const float* __restrict__ dgamma, bf16raw* __restrict__ dx,
+                                      int C, int64_t total, float inv_count, int train_stats) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int c = (int)(i % C);
+    float g = bf16_to_f32(dy[i]);
+    if (yrelu != nullptr && bf16_to_f32(yrelu[i]) <= 0.f) g = 0.f;
+    float is = invstd[c];
+    float out;
+    if (train_stats) {
+      float xh = (bf16_to_f32(x[i]) - mean[c]) * is;
+      out = gamma[c] * is * (g - dbeta[c] * inv_count - xh * dgamma[c] * inv_count);
+    } else {
+      out = gamma[c] * is * g;
+    }
+    dx[i] = f32_to_bf16(out);
+  }
+}
+
+extern "C" hipError_t launch_bn_bwd_dx_nhwc(const bf16raw* dy, const bf16raw* yrelu,
+                                            const bf16raw* x, const float* mean,
+                                            const float* invstd, const float* gamma,
+                                            const float* dbeta, const float* dgamma,
+                                            bf16raw* dx, int C, int64_t total, float inv_count,
+                                            int train_stats, hipStream_t stream) {
+  int64_t g = ceil_div_i64(total, 256);
+  int grid = (int)(g > 8192 ? 8192 : (g < 1 ? 1 : g));
+  bn_bwd_dx_nhwc_kernel<<<grid, 256, 0, stream>>>(dy, yrelu, x, mean, invstd, gamma, dbeta,
+                                                  dgamma, dx, C, total, inv_count, train_stats);
+  HIP_CHECK_LAUNCH();
+  return hipSuccess;
+}
+
+// ---------------------------------------------------------------------------
 // BN apply: y = gamma*(x-mean)*invstd + beta [+ residual] [relu].
 // Elementwise over B*C*HW; residual may be null.
 // ---------------------------------------------------------------------------

@@ -168,6 +168,181 @@ def hip_conv2d(x, weight, bias=None, stride=(1, 1), padding=(0, 0), relu=False):
     return F.relu(y) if relu else y
 
 
+class _Conv2dNHWCFn(torch.autograd.Function):
+    """Channels-last Conv2d as implicit GEMM.  The profile of the NCHW path
+    (profiles/) showed im2col/col2im gather at 54% of the ResNet-18 step;
+    in NHWC both kernels move CI-contiguous shortx8 vectors and the GEMM
+    output [B*HO*WO, CO] IS the NHWC activation — zero permute kernels.
+    Weights stay [CO,CI,KH,KW] (state_dict compatible); the [CO, KH*KW*CI]
+    GEMM view is built per forward (small)."""
+
+    @staticmethod
+    def forward(ctx, x, w, b, stride, padding, relu):
+        ext = ops.ext()
+        B, H, W, CI = x.shape
+        CO, _, KH, KW = w.shape
+        sh, sw = stride
+        ph, pw = padding
+        HO = (H + 2 * ph - KH) // sh + 1
+        WO = (W + 2 * pw - KW) // sw + 1
+        col = ext.im2col_nhwc(x, KH, KW, sh, sw, ph, pw)
+        w2d = w.permute(0, 2, 3, 1).reshape(CO, KH * KW * CI).contiguous()
+        y2d = ext.linear_fwd(col, w2d, b, relu)  # [B*HO*WO, CO] == NHWC
+        ctx.save_for_backward(col, w2d, y2d)
+        ctx.meta = (B, CI, H, W, CO, KH, KW, sh, sw, ph, pw, HO, WO, relu, b is not None)
+        ctx.b_ref = b
+        return y2d.view(B, HO, WO, CO)
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = ops.ext()
+        col, w2d, y2d = ctx.saved_tensors
+        B, CI, H, W, CO, KH, KW, sh, sw, ph, pw, HO, WO, relu, has_bias = ctx.meta
+        dy2d = dy.reshape(B * HO * WO, CO).contiguous()
+        if dy2d.dtype != torch.bfloat16:
+            dy2d = dy2d.to(torch.bfloat16)
+        dz = ext.relu_bwd(dy2d, y2d) if relu else dy2d
+
+        dw = db = None
+        sk = _choose_splitk(dz.shape[0], CO, col.shape[1])
+        if ctx.needs_input_grad[1]:
+            # (kh,kw,ci)-ordered wgrad -> param layout; autograd adds into the
+            # flat-bucket grad view, so overlap hooks still fire.
+            dw = ext.linear_wgrad(dz, col, sk).view(CO, KH, KW, CI).permute(0, 3, 1, 2)
+        if has_bias and ctx.needs_input_grad[2]:
+            db = ext.bias_grad(dz)
+
+        dx = None
+        if ctx.needs_input_grad[0]:
+            dcol = ext.linear_dgrad(dz, w2d)
+            dx = ext.col2im_nhwc(dcol, B, CI, H, W, KH, KW, sh, sw, ph, pw)
+        return dx, dw, db, None, None, None
+
+
+def hip_conv2d_nhwc(x, weight, bias=None, stride=(1, 1), padding=(0, 0), relu=False):
+    """x is [B,H,W,CI] contiguous; returns [B,HO,WO,CO]."""
+    if x.is_cuda:
+        if x.dtype != torch.bfloat16:
+            x = x.to(torch.bfloat16)
+        return _Conv2dNHWCFn.apply(x.contiguous(), weight, bias, stride, padding, relu)
+    y = F.conv2d(x.permute(0, 3, 1, 2), weight, bias, stride=stride, padding=padding)
+    y = F.relu(y) if relu else y
+    return y.permute(0, 2, 3, 1).contiguous()
+
+
+class _BatchNorm2dNHWCFn(torch.autograd.Function):
+    """NHWC BatchNorm2d: per-channel column reductions (coalesced 64-channel
+    tiles), elementwise apply with optional fused ReLU; same direct-grad
+    bucket convention as the NCHW variant."""
+
+    @staticmethod
+    def forward(ctx, x, gamma, beta, running_mean, running_var, training, momentum, eps, relu):
+        ext = ops.ext()
+        if training:
+            mean, invstd = ext.bn_stats_nhwc(x, running_mean, running_var, momentum, eps)
+        else:
+            mean = running_mean.contiguous()
+            invstd = (running_var + eps).rsqrt().contiguous()
+        y = ext.bn_apply_nhwc(x, None, mean, invstd, gamma, beta, relu)
+        ctx.save_for_backward(x, y if relu else None, mean, invstd, gamma)
+        ctx.relu = relu
+        ctx.train_stats = training
+        ctx.gamma_ref, ctx.beta_ref = gamma, beta
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = ops.ext()
+        x, yrelu, mean, invstd, gamma = ctx.saved_tensors
+        dy = dy.contiguous()
+        if dy.dtype != torch.bfloat16:
+            dy = dy.to(torch.bfloat16)
+        g, b = ctx.gamma_ref, ctx.beta_ref
+        g_notify = getattr(g, "_bucket_notify", None)
+        b_notify = getattr(b, "_bucket_notify", None)
+        direct = (
+            g_notify is not None and g.grad is not None
+            and b_notify is not None and b.grad is not None
+        )
+        if direct:
+            dgamma_buf, dbeta_buf = g.grad, b.grad
+        else:
+            dgamma_buf = torch.zeros_like(mean)
+            dbeta_buf = torch.zeros_like(mean)
+        ext.bn_bwd_reduce_nhwc(dy, yrelu, x, mean, invstd, dbeta_buf, dgamma_buf)
+        dx = None
+        if ctx.needs_input_grad[0]:
+            dx = ext.bn_bwd_dx_nhwc(dy, yrelu, x, mean, invstd, gamma, dbeta_buf, dgamma_buf,
+                                    ctx.train_stats)
+        if direct:
+            g_notify()
+            b_notify()
+            return dx, None, None, None, None, None, None, None, None
+        return dx, dgamma_buf, dbeta_buf, None, None, None, None, None, None
+
+
+def hip_batch_norm2d_nhwc(x, gamma, beta, running_mean, running_var, training: bool,
+                          momentum: float = 0.1, eps: float = 1e-5, relu: bool = False):
+    if x.is_cuda:
+        if x.dtype != torch.bfloat16:
+            x = x.to(torch.bfloat16)
+        return _BatchNorm2dNHWCFn.apply(x.contiguous(), gamma, beta, running_mean, running_var,
+                                        training, momentum, eps, relu)
+    y = F.batch_norm(x.permute(0, 3, 1, 2), running_mean, running_var, gamma, beta, training,
+                     momentum, eps)
+    y = F.relu(y) if relu else y
+    return y.permute(0, 2, 3, 1).contiguous()
+
+
+class _MaxPool2dNHWCFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, ks, stride, pad):
+        y, arg = ops.ext().maxpool_nhwc_fwd(x, ks, stride, pad)
+        ctx.save_for_backward(arg)
+        ctx.meta = (x.shape[1], x.shape[2], ks, stride, pad)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (arg,) = ctx.saved_tensors
+        H, W, ks, stride, pad = ctx.meta
+        dx = ops.ext().maxpool_nhwc_bwd(dy.contiguous().to(torch.bfloat16), arg, H, W, ks,
+                                        stride, pad)
+        return dx, None, None, None
+
+
+def hip_max_pool2d_nhwc(x, kernel_size: int, stride: Optional[int] = None, padding: int = 0):
+    if stride is None:
+        stride = kernel_size
+    if x.is_cuda:
+        if x.dtype != torch.bfloat16:
+            x = x.to(torch.bfloat16)
+        return _MaxPool2dNHWCFn.apply(x.contiguous(), kernel_size, stride, padding)
+    y = F.max_pool2d(x.permute(0, 3, 1, 2), kernel_size, stride=stride, padding=padding)
+    return y.permute(0, 2, 3, 1).contiguous()
+
+
+class _GlobalAvgPoolNHWCFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        ctx.meta = (x.shape[1], x.shape[2])
+        return ops.ext().gap_nhwc_fwd(x)
+
+    @staticmethod
+    def backward(ctx, dy):
+        H, W = ctx.meta
+        return ops.ext().gap_nhwc_bwd(dy.contiguous().to(torch.bfloat16), H, W)
+
+
+def hip_global_avg_pool_nhwc(x):
+    """[B,H,W,C] -> [B,C] channel means."""
+    if x.is_cuda:
+        if x.dtype != torch.bfloat16:
+            x = x.to(torch.bfloat16)
+        return _GlobalAvgPoolNHWCFn.apply(x.contiguous())
+    return x.mean(dim=(1, 2))
+
+
 class _BatchNorm2dFn(torch.autograd.Function):
     """BatchNorm2d on the native kernels: training stats (channel×slice grid
     + atomics, fp32), normalize+affine with optional fused ReLU; backward is

@@ -18,12 +18,16 @@ import torch.nn as nn
 from sparktorch_amd.ops.functional import (
     hip_add_relu,
     hip_batch_norm2d,
+    hip_batch_norm2d_nhwc,
     hip_conv2d,
+    hip_conv2d_nhwc,
     hip_cross_entropy,
     hip_dropout,
     hip_global_avg_pool,
+    hip_global_avg_pool_nhwc,
     hip_linear,
     hip_max_pool2d,
+    hip_max_pool2d_nhwc,
     hip_mse,
 )
 
@@ -64,7 +68,8 @@ class HipConv2d(nn.Module):
     dilation=1, groups=1), optional fused ReLU."""
 
     def __init__(self, in_ch: int, out_ch: int, kernel_size: int, stride: int = 1,
-                 padding: int = 0, bias: bool = True, activation: Optional[str] = None):
+                 padding: int = 0, bias: bool = True, activation: Optional[str] = None,
+                 layout: str = "nchw"):
         super().__init__()
         ks = (kernel_size, kernel_size) if isinstance(kernel_size, int) else kernel_size
         self.stride = (stride, stride) if isinstance(stride, int) else stride
@@ -72,6 +77,7 @@ class HipConv2d(nn.Module):
         self.weight = nn.Parameter(torch.empty(out_ch, in_ch, *ks))
         self.bias = nn.Parameter(torch.zeros(out_ch)) if bias else None
         self.activation = activation
+        self.layout = layout
         nn.init.kaiming_uniform_(self.weight, a=5 ** 0.5)
 
     @classmethod
@@ -85,23 +91,28 @@ class HipConv2d(nn.Module):
         m.weight = conv.weight  # SAME Parameter objects
         m.bias = conv.bias
         m.activation = activation
+        m.layout = "nchw"
         return m
 
     def forward(self, x):
-        return hip_conv2d(
+        fn = hip_conv2d_nhwc if self.layout == "nhwc" else hip_conv2d
+        return fn(
             x, self.weight, self.bias, self.stride, self.padding, relu=self.activation == "relu"
         )
 
 
 class HipMaxPool2d(nn.Module):
-    def __init__(self, kernel_size: int, stride: Optional[int] = None, padding: int = 0):
+    def __init__(self, kernel_size: int, stride: Optional[int] = None, padding: int = 0,
+                 layout: str = "nchw"):
         super().__init__()
         self.kernel_size = kernel_size
         self.stride = kernel_size if stride is None else stride
         self.padding = padding
+        self.layout = layout
 
     def forward(self, x):
-        return hip_max_pool2d(x, self.kernel_size, self.stride, self.padding)
+        fn = hip_max_pool2d_nhwc if self.layout == "nhwc" else hip_max_pool2d
+        return fn(x, self.kernel_size, self.stride, self.padding)
 
 
 class HipBatchNorm2d(nn.Module):
@@ -109,12 +120,13 @@ class HipBatchNorm2d(nn.Module):
     with optional fused ReLU.  state_dict-compatible with nn.BatchNorm2d."""
 
     def __init__(self, num_features: int, eps: float = 1e-5, momentum: float = 0.1,
-                 activation: Optional[str] = None):
+                 activation: Optional[str] = None, layout: str = "nchw"):
         super().__init__()
         self.num_features = num_features
         self.eps = eps
         self.momentum = momentum
         self.activation = activation
+        self.layout = layout
         self.weight = nn.Parameter(torch.ones(num_features))
         self.bias = nn.Parameter(torch.zeros(num_features))
         self.register_buffer("running_mean", torch.zeros(num_features))
@@ -131,6 +143,7 @@ class HipBatchNorm2d(nn.Module):
         m.eps = bn.eps
         m.momentum = bn.momentum if bn.momentum is not None else 0.1
         m.activation = activation
+        m.layout = "nchw"
         m.weight = bn.weight  # SAME Parameter objects
         m.bias = bn.bias
         m.register_buffer("running_mean", bn.running_mean)
@@ -141,7 +154,8 @@ class HipBatchNorm2d(nn.Module):
     def forward(self, x):
         if self.training:
             self.num_batches_tracked += 1
-        return hip_batch_norm2d(
+        fn = hip_batch_norm2d_nhwc if self.layout == "nhwc" else hip_batch_norm2d
+        return fn(
             x, self.weight, self.bias, self.running_mean, self.running_var,
             self.training, self.momentum, self.eps, relu=self.activation == "relu",
         )
@@ -153,8 +167,13 @@ class HipBatchNorm2d(nn.Module):
 class HipGlobalAvgPool(nn.Module):
     """adaptive_avg_pool2d(x, 1).flatten(1) as one reduction kernel."""
 
+    def __init__(self, layout: str = "nchw"):
+        super().__init__()
+        self.layout = layout
+
     def forward(self, x):
-        return hip_global_avg_pool(x)
+        fn = hip_global_avg_pool_nhwc if self.layout == "nhwc" else hip_global_avg_pool
+        return fn(x)
 
 
 class HipDropout(nn.Module):
@@ -196,17 +215,19 @@ class FusedBasicBlock(nn.Module):
     fused ReLU, and a fused residual add+relu join.  state_dict-compatible
     with models.resnet.BasicBlock."""
 
-    def __init__(self, in_ch: int, out_ch: int, stride: int = 1):
+    def __init__(self, in_ch: int, out_ch: int, stride: int = 1, layout: str = "nhwc"):
         super().__init__()
-        self.conv1 = HipConv2d(in_ch, out_ch, 3, stride=stride, padding=1, bias=False)
-        self.bn1 = HipBatchNorm2d(out_ch, activation="relu")
-        self.conv2 = HipConv2d(out_ch, out_ch, 3, stride=1, padding=1, bias=False)
-        self.bn2 = HipBatchNorm2d(out_ch)
+        self.conv1 = HipConv2d(in_ch, out_ch, 3, stride=stride, padding=1, bias=False,
+                               layout=layout)
+        self.bn1 = HipBatchNorm2d(out_ch, activation="relu", layout=layout)
+        self.conv2 = HipConv2d(out_ch, out_ch, 3, stride=1, padding=1, bias=False,
+                               layout=layout)
+        self.bn2 = HipBatchNorm2d(out_ch, layout=layout)
         self.down = None
         if stride != 1 or in_ch != out_ch:
             self.down = nn.Sequential(
-                HipConv2d(in_ch, out_ch, 1, stride=stride, bias=False),
-                HipBatchNorm2d(out_ch),
+                HipConv2d(in_ch, out_ch, 1, stride=stride, bias=False, layout=layout),
+                HipBatchNorm2d(out_ch, layout=layout),
             )
 
     def forward(self, x):
@@ -219,24 +240,29 @@ class FusedBasicBlock(nn.Module):
 
 class ResNet18Fused(nn.Module):
     """ResNet-18 (BASELINE config 4: synthetic 3x224x224 Vector rows) fully
-    on the hand-written CDNA4 kernels: 7x7/s2 stem conv, overlapping
-    3x3/s2/p1 maxpool, 8 basic blocks, global-avg-pool reduction, MFMA fc.
-    state_dict-compatible with models.resnet.ResNet18."""
+    on the hand-written CDNA4 kernels, channels-last (NHWC) end to end:
+    7x7/s2 stem conv, overlapping 3x3/s2/p1 maxpool, 8 basic blocks,
+    global-avg-pool reduction, MFMA fc.  NHWC makes every im2col/col2im run
+    CI-contiguous (shortx8 vector moves) and the implicit-GEMM output
+    [B*HO*WO, CO] IS the activation — no layout permutes anywhere (the NCHW
+    gathers were 54%% of the step in the first rocprof capture, profiles/).
+    state_dict-compatible with models.resnet.ResNet18 (weights [CO,CI,KH,KW])."""
 
-    def __init__(self, num_classes: int = 1000, in_ch: int = 3):
+    def __init__(self, num_classes: int = 1000, in_ch: int = 3, layout: str = "nhwc"):
         super().__init__()
-        self.conv1 = HipConv2d(in_ch, 64, 7, stride=2, padding=3, bias=False)
-        self.bn1 = HipBatchNorm2d(64, activation="relu")
-        self.maxpool = HipMaxPool2d(3, stride=2, padding=1)
+        self.layout = layout
+        self.conv1 = HipConv2d(in_ch, 64, 7, stride=2, padding=3, bias=False, layout=layout)
+        self.bn1 = HipBatchNorm2d(64, activation="relu", layout=layout)
+        self.maxpool = HipMaxPool2d(3, stride=2, padding=1, layout=layout)
         layers = []
         cfg = [(64, 1), (128, 2), (256, 2), (512, 2)]
         ch = 64
         for out_ch, stride in cfg:
-            layers.append(FusedBasicBlock(ch, out_ch, stride))
-            layers.append(FusedBasicBlock(out_ch, out_ch, 1))
+            layers.append(FusedBasicBlock(ch, out_ch, stride, layout=layout))
+            layers.append(FusedBasicBlock(out_ch, out_ch, 1, layout=layout))
             ch = out_ch
         self.layers = nn.Sequential(*layers)
-        self.gap = HipGlobalAvgPool()
+        self.gap = HipGlobalAvgPool(layout=layout)
         self.fc = HipLinear(512, num_classes)
 
     def forward(self, x):
@@ -244,6 +270,8 @@ class ResNet18Fused(nn.Module):
             x = x.to(torch.bfloat16)
         if x.dim() == 2:  # flattened Vector rows, like the CNN unflatten idiom
             x = x.view(-1, 3, 224, 224)
+        if self.layout == "nhwc":  # one entry permute; everything after is NHWC
+            x = x.permute(0, 2, 3, 1).contiguous()
         x = self.bn1(self.conv1(x))
         x = self.maxpool(x)
         x = self.layers(x)
