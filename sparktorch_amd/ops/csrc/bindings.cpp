@@ -176,8 +176,8 @@ at::Tensor linear_fwd(at::Tensor x, at::Tensor w, c10::optional<at::Tensor> bias
     check_gpu_contig(*bias, at::kFloat, "bias");
     bptr = bias->data_ptr<float>();
     epi = relu ? 3 : 2;
-  } else {
-    TORCH_CHECK(!relu, "relu without bias not wired");
+  } else if (relu) {
+    epi = 4;  // EPI_RELU
   }
   CHECK_HIP(launch_gemm_bf16(x.data_ptr(), w.data_ptr(), w.scalar_type() == at::kFloat ? 1 : 0,
                              nullptr, (bf16raw*)y.data_ptr(), bptr, (int)M, (int)N, (int)K,

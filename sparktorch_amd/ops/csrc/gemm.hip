@@ -34,6 +34,7 @@
 #define EPI_BF16 1       // bf16 store
 #define EPI_BIAS 2       // bf16 store, + bias[n]
 #define EPI_BIAS_RELU 3  // bf16 store, + bias[n], relu
+#define EPI_RELU 4       // bf16 store, relu (no bias)
 
 typedef shortx8 frag_t;  // 8 bf16 (4 VGPRs)
 
@@ -243,8 +244,8 @@ __global__ __launch_bounds__(256) void gemm_kernel(const void* __restrict__ Ap,
           }
         } else {
           int64_t off = (int64_t)m * N + n;
-          if (EPI >= EPI_BIAS) val += bias[n];
-          if (EPI == EPI_BIAS_RELU) val = fmaxf(val, 0.f);
+          if (EPI == EPI_BIAS || EPI == EPI_BIAS_RELU) val += bias[n];
+          if (EPI == EPI_BIAS_RELU || EPI == EPI_RELU) val = fmaxf(val, 0.f);
           Cb[off] = f32_to_bf16(val);
         }
       }
@@ -284,6 +285,7 @@ extern "C" hipError_t launch_gemm_bf16(const void* A, const void* B, int b_is_f3
     else if (epi == EPI_BF16) DISPATCH(true, EPI_BF16, false);
     else if (epi == EPI_BIAS) DISPATCH(true, EPI_BIAS, false);
     else if (epi == EPI_BIAS_RELU) DISPATCH(true, EPI_BIAS_RELU, false);
+    else if (epi == EPI_RELU) DISPATCH(true, EPI_RELU, false);
     else return hipErrorInvalidValue;
   } else {
     if (atomic && epi == EPI_F32) DISPATCH(false, EPI_F32, true);
@@ -291,6 +293,7 @@ extern "C" hipError_t launch_gemm_bf16(const void* A, const void* B, int b_is_f3
     else if (epi == EPI_BF16) DISPATCH(false, EPI_BF16, false);
     else if (epi == EPI_BIAS) DISPATCH(false, EPI_BIAS, false);
     else if (epi == EPI_BIAS_RELU) DISPATCH(false, EPI_BIAS_RELU, false);
+    else if (epi == EPI_RELU) DISPATCH(false, EPI_RELU, false);
     else return hipErrorInvalidValue;
   }
 #undef DISPATCH

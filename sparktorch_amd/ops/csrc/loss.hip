@@ -47,11 +47,49 @@ __global__ void ce_fused_kernel(const bf16raw* __restrict__ logits,
   if (threadIdx.x == 0 && red[0] != 0.f) atomicAdd(loss_out, red[0]);
 }
 
+// Wave-per-row variant for wide C (e.g. 1000-class ResNet head) or small B:
+// 64 lanes sweep the row with shuffle all-reduce for max / sum-exp, so one
+// block covers 4 rows on 4 SIMDs instead of 256 rows on one CU.
+__global__ void ce_fused_wave_kernel(const bf16raw* __restrict__ logits,
+                                     const int64_t* __restrict__ tgt,
+                                     float* __restrict__ loss_out,
+                                     bf16raw* __restrict__ dlogits, int B, int C, float inv_B) {
+  int wave = threadIdx.x >> 6;
+  int lane = threadIdx.x & 63;
+  int64_t b = (int64_t)blockIdx.x * 4 + wave;
+  if (b >= B) return;
+  const bf16raw* row = logits + b * C;
+  bf16raw* drow = dlogits + b * C;
+  float mx = -1e30f;
+  for (int c = lane; c < C; c += 64) mx = fmaxf(mx, bf16_to_f32(row[c]));
+  for (int off = 32; off; off >>= 1) mx = fmaxf(mx, __shfl_xor(mx, off, 64));
+  float se = 0.f;
+  for (int c = lane; c < C; c += 64) se += __expf(bf16_to_f32(row[c]) - mx);
+  for (int off = 32; off; off >>= 1) se += __shfl_xor(se, off, 64);
+  int64_t t = tgt[b];
+  float inv_se = 1.f / se;
+  for (int c = lane; c < C; c += 64) {
+    float p = __expf(bf16_to_f32(row[c]) - mx) * inv_se;
+    if (c == (int)t) p -= 1.f;
+    drow[c] = f32_to_bf16(p * inv_B);
+  }
+  if (lane == 0) {
+    float my_loss = (__logf(se) + mx - bf16_to_f32(row[t])) * inv_B;
+    atomicAdd(loss_out, my_loss);
+  }
+}
+
 extern "C" hipError_t launch_ce_fused(const bf16raw* logits, const int64_t* tgt, float* loss_out,
                                       bf16raw* dlogits, int B, int C, hipStream_t stream) {
-  int block = 256;
-  int grid = (int)ceil_div_i64(B, block);
-  ce_fused_kernel<<<grid, block, 0, stream>>>(logits, tgt, loss_out, dlogits, B, C, 1.0f / B);
+  if (C >= 128 || B <= 2048) {
+    int grid = (int)ceil_div_i64(B, 4);
+    ce_fused_wave_kernel<<<grid, 256, 0, stream>>>(logits, tgt, loss_out, dlogits, B, C,
+                                                   1.0f / B);
+  } else {
+    int block = 256;
+    int grid = (int)ceil_div_i64(B, block);
+    ce_fused_kernel<<<grid, block, 0, stream>>>(logits, tgt, loss_out, dlogits, B, C, 1.0f / B);
+  }
   HIP_CHECK_LAUNCH();
   return hipSuccess;
 }
