@@ -83,6 +83,44 @@ __global__ void im2col_nhwc_kernel(const bf16raw* __restrict__ x, bf16raw* __res
   }
 }
 
+// Small-CI patch-row copy (the 3-channel stem conv): one thread = one
+// (b,ho,wo,kh) patch row of KW*CI elements — reads and writes are contiguous
+// runs (42 B for 7x7/CI=3), decode cost amortized over the whole row.
+__global__ void im2col_nhwc_rowcopy_kernel(const bf16raw* __restrict__ x,
+                                           bf16raw* __restrict__ col, int B, int CI, int H,
+                                           int W, int KH, int KW, int HO, int WO, int sh,
+                                           int sw, int ph, int pw) {
+  int64_t K = (int64_t)CI * KH * KW;
+  int64_t total = (int64_t)B * HO * WO * KH;
+  for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (int64_t)gridDim.x * blockDim.x) {
+    int kh = (int)(idx % KH);
+    int64_t t = idx / KH;
+    int wo = (int)(t % WO);
+    t /= WO;
+    int ho = (int)(t % HO);
+    int b = (int)(t / HO);
+    int h = ho * sh - ph + kh;
+    int w0 = wo * sw - pw;
+    int64_t m = ((int64_t)b * HO + ho) * WO + wo;
+    bf16raw* dst = col + m * K + (int64_t)kh * KW * CI;
+    if (h < 0 || h >= H) {
+      for (int j = 0; j < KW * CI; ++j) dst[j] = 0;
+      continue;
+    }
+    const bf16raw* src = x + (((int64_t)b * H + h) * W + w0) * CI;
+    if (w0 >= 0 && w0 + KW <= W) {
+      for (int j = 0; j < KW * CI; ++j) dst[j] = src[j];
+    } else {
+      for (int kw = 0; kw < KW; ++kw) {
+        int w = w0 + kw;
+        for (int ci = 0; ci < CI; ++ci)
+          dst[kw * CI + ci] = (w >= 0 && w < W) ? src[kw * CI + ci] : (bf16raw)0;
+      }
+    }
+  }
+}
+
 extern "C" hipError_t launch_im2col_nhwc(const bf16raw* x, bf16raw* col, int B, int CI, int H,
                                          int W, int KH, int KW, int HO, int WO, int sh, int sw,
                                          int ph, int pw, hipStream_t stream) {
@@ -91,9 +129,10 @@ extern "C" hipError_t launch_im2col_nhwc(const bf16raw* x, bf16raw* col, int B, 
     im2col_nhwc_vec_kernel<<<cgrid8(total, 256), 256, 0, stream>>>(x, col, B, CI, H, W, KH, KW,
                                                                    HO, WO, sh, sw, ph, pw);
   } else {
-    int64_t total = (int64_t)B * HO * WO * CI * KH * KW;
-    im2col_nhwc_kernel<<<cgrid8(total, 256), 256, 0, stream>>>(x, col, B, CI, H, W, KH, KW, HO,
-                                                               WO, sh, sw, ph, pw);
+    int64_t total = (int64_t)B * HO * WO * KH;
+    im2col_nhwc_rowcopy_kernel<<<cgrid8(total, 256), 256, 0, stream>>>(x, col, B, CI, H, W, KH,
+                                                                       KW, HO, WO, sh, sw, ph,
+                                                                       pw);
   }
   HIP_CHECK_LAUNCH();
   return hipSuccess;
@@ -270,12 +309,116 @@ __global__ void maxpool_nhwc_bwd_kernel(const bf16raw* __restrict__ dy,
   }
 }
 
+// Vectorized variants: one thread = 8 channels of one spatial cell (the
+// window geometry is channel-invariant, so bounds math amortizes 8x and all
+// loads/stores are shortx8).
+__global__ void maxpool_nhwc_fwd_vec_kernel(const bf16raw* __restrict__ x,
+                                            bf16raw* __restrict__ y, uint8_t* __restrict__ arg,
+                                            int B, int C, int H, int W, int HO, int WO, int KS,
+                                            int S, int P) {
+  int cv = C >> 3;
+  int64_t total = (int64_t)B * HO * WO * cv;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int c8 = (int)(i % cv);
+    int64_t t = i / cv;
+    int wo = (int)(t % WO);
+    t /= WO;
+    int ho = (int)(t % HO);
+    int b = (int)(t / HO);
+    const bf16raw* xp = x + (int64_t)b * H * W * C + (c8 << 3);
+    float best[8];
+    int bk[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      best[j] = -3.4e38f;
+      bk[j] = 0;
+    }
+    for (int kh = 0; kh < KS; ++kh) {
+      int h = ho * S - P + kh;
+      if (h < 0 || h >= H) continue;
+      for (int kw = 0; kw < KS; ++kw) {
+        int w = wo * S - P + kw;
+        if (w < 0 || w >= W) continue;
+        const shortx8 v = *(const shortx8*)(xp + ((int64_t)h * W + w) * C);
+        int kk = kh * KS + kw;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float f = bf16_to_f32((bf16raw)v[j]);
+          if (f > best[j]) {
+            best[j] = f;
+            bk[j] = kk;
+          }
+        }
+      }
+    }
+    int64_t o = (((int64_t)b * HO + ho) * WO + wo) * C + (c8 << 3);
+    shortx8 out;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      out[j] = (short)f32_to_bf16(best[j]);
+      arg[o + j] = (uint8_t)bk[j];
+    }
+    *(shortx8*)(y + o) = out;
+  }
+}
+
+__global__ void maxpool_nhwc_bwd_vec_kernel(const bf16raw* __restrict__ dy,
+                                            const uint8_t* __restrict__ arg,
+                                            bf16raw* __restrict__ dx, int B, int C, int H,
+                                            int W, int HO, int WO, int KS, int S, int P) {
+  int cv = C >> 3;
+  int64_t total = (int64_t)B * H * W * cv;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int c8 = (int)(i % cv);
+    int64_t t = i / cv;
+    int w = (int)(t % W);
+    t /= W;
+    int h = (int)(t % H);
+    int b = (int)(t / H);
+    float acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+    int ho_lo = (h + P - KS + S) / S;
+    if (ho_lo < 0) ho_lo = 0;
+    int ho_hi = (h + P) / S;
+    if (ho_hi >= HO) ho_hi = HO - 1;
+    int wo_lo = (w + P - KS + S) / S;
+    if (wo_lo < 0) wo_lo = 0;
+    int wo_hi = (w + P) / S;
+    if (wo_hi >= WO) wo_hi = WO - 1;
+    for (int ho = ho_lo; ho <= ho_hi; ++ho) {
+      int kh = h - (ho * S - P);
+      if (kh < 0 || kh >= KS) continue;
+      for (int wo = wo_lo; wo <= wo_hi; ++wo) {
+        int kw = w - (wo * S - P);
+        if (kw < 0 || kw >= KS) continue;
+        int64_t o = (((int64_t)b * HO + ho) * WO + wo) * C + (c8 << 3);
+        const shortx8 g8 = *(const shortx8*)(dy + o);
+        uint8_t kk = (uint8_t)(kh * KS + kw);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          if (arg[o + j] == kk) acc[j] += bf16_to_f32((bf16raw)g8[j]);
+      }
+    }
+    shortx8 out;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) out[j] = (short)f32_to_bf16(acc[j]);
+    *(shortx8*)(dx + ((((int64_t)b * H + h) * W + w) * C) + (c8 << 3)) = out;
+  }
+}
+
 extern "C" hipError_t launch_maxpool_nhwc_fwd(const bf16raw* x, bf16raw* y, uint8_t* arg, int B,
                                               int C, int H, int W, int HO, int WO, int KS, int S,
                                               int P, hipStream_t stream) {
-  int64_t total = (int64_t)B * HO * WO * C;
-  maxpool_nhwc_fwd_kernel<<<cgrid8(total, 256), 256, 0, stream>>>(x, y, arg, B, C, H, W, HO, WO,
-                                                                  KS, S, P);
+  if ((C & 7) == 0) {
+    int64_t total = (int64_t)B * HO * WO * (C >> 3);
+    maxpool_nhwc_fwd_vec_kernel<<<cgrid8(total, 256), 256, 0, stream>>>(x, y, arg, B, C, H, W,
+                                                                        HO, WO, KS, S, P);
+  } else {
+    int64_t total = (int64_t)B * HO * WO * C;
+    maxpool_nhwc_fwd_kernel<<<cgrid8(total, 256), 256, 0, stream>>>(x, y, arg, B, C, H, W, HO,
+                                                                    WO, KS, S, P);
+  }
   HIP_CHECK_LAUNCH();
   return hipSuccess;
 }
@@ -283,9 +426,15 @@ extern "C" hipError_t launch_maxpool_nhwc_fwd(const bf16raw* x, bf16raw* y, uint
 extern "C" hipError_t launch_maxpool_nhwc_bwd(const bf16raw* dy, const uint8_t* arg, bf16raw* dx,
                                               int B, int C, int H, int W, int HO, int WO, int KS,
                                               int S, int P, hipStream_t stream) {
-  int64_t total = (int64_t)B * H * W * C;
-  maxpool_nhwc_bwd_kernel<<<cgrid8(total, 256), 256, 0, stream>>>(dy, arg, dx, B, C, H, W, HO,
-                                                                  WO, KS, S, P);
+  if ((C & 7) == 0) {
+    int64_t total = (int64_t)B * H * W * (C >> 3);
+    maxpool_nhwc_bwd_vec_kernel<<<cgrid8(total, 256), 256, 0, stream>>>(dy, arg, dx, B, C, H, W,
+                                                                        HO, WO, KS, S, P);
+  } else {
+    int64_t total = (int64_t)B * H * W * C;
+    maxpool_nhwc_bwd_kernel<<<cgrid8(total, 256), 256, 0, stream>>>(dy, arg, dx, B, C, H, W, HO,
+                                                                    WO, KS, S, P);
+  }
   HIP_CHECK_LAUNCH();
   return hipSuccess;
 }

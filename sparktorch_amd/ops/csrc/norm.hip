@@ -122,12 +122,65 @@ __global__ void bn_stats_partial_nhwc_kernel(const bf16raw* __restrict__ x,
   }
 }
 
+// Vectorized stats: thread owns 8 consecutive channels (shortx8 row loads),
+// 256/(C/8) row-lanes per block; requires C in {8,16,...,2048} dividing the
+// block evenly (ResNet channels are powers of two).
+__global__ void bn_stats_partial_nhwc_vec_kernel(const bf16raw* __restrict__ x,
+                                                 float* __restrict__ sum,
+                                                 float* __restrict__ sumsq, int64_t M, int C) {
+  int groups = C >> 3;
+  int rpg = 256 / groups;             // row-lanes per block
+  int cg = threadIdx.x % groups;
+  int rg = threadIdx.x / groups;
+  int c0 = cg << 3;
+  int64_t per = ceil_div_i64(M, gridDim.x);
+  int64_t lo = (int64_t)blockIdx.x * per;
+  int64_t hi = lo + per < M ? lo + per : M;
+  float s[8] = {0, 0, 0, 0, 0, 0, 0, 0}, ss[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  for (int64_t r = lo + rg; r < hi; r += rpg) {
+    const shortx8 v = *(const shortx8*)(x + r * C + c0);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = bf16_to_f32((bf16raw)v[j]);
+      s[j] += f;
+      ss[j] += f * f;
+    }
+  }
+  __shared__ float l0[256][8], l1[256][8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    l0[threadIdx.x][j] = s[j];
+    l1[threadIdx.x][j] = ss[j];
+  }
+  __syncthreads();
+  if (rg == 0) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float a = 0.f, b = 0.f;
+      for (int q = 0; q < rpg; ++q) {
+        a += l0[q * groups + cg][j];
+        b += l1[q * groups + cg][j];
+      }
+      atomicAdd(&sum[c0 + j], a);
+      atomicAdd(&sumsq[c0 + j], b);
+    }
+  }
+}
+
+static inline bool bn_vec_ok(int C) {
+  return C >= 8 && C <= 2048 && (C & 7) == 0 && (256 % (C >> 3)) == 0;
+}
+
 extern "C" hipError_t launch_bn_stats_nhwc(const bf16raw* x, float* sum, float* sumsq,
                                            float* mean, float* invstd, float* running_mean,
                                            float* running_var, int64_t M, int C, float momentum,
                                            float eps, int nsplit, hipStream_t stream) {
-  dim3 grid((C + 63) / 64, nsplit);
-  bn_stats_partial_nhwc_kernel<<<grid, 256, 0, stream>>>(x, sum, sumsq, M, C);
+  if (bn_vec_ok(C)) {
+    bn_stats_partial_nhwc_vec_kernel<<<2048, 256, 0, stream>>>(x, sum, sumsq, M, C);
+  } else {
+    dim3 grid((C + 63) / 64, nsplit);
+    bn_stats_partial_nhwc_kernel<<<grid, 256, 0, stream>>>(x, sum, sumsq, M, C);
+  }
   HIP_CHECK_LAUNCH();
   int blocks = (int)ceil_div_i64(C, 256);
   bn_stats_finalize_kernel<<<blocks, 256, 0, stream>>>(sum, sumsq, mean, invstd, running_mean,
@@ -154,14 +207,48 @@ __global__ void bn_apply_nhwc_kernel(const bf16raw* __restrict__ x,
   }
 }
 
+// 8 elements per thread (C % 8 == 0 keeps the vector inside one channel run)
+__global__ void bn_apply_nhwc_vec_kernel(const bf16raw* __restrict__ x,
+                                         const bf16raw* __restrict__ res,
+                                         bf16raw* __restrict__ y, const float* __restrict__ mean,
+                                         const float* __restrict__ invstd,
+                                         const float* __restrict__ gamma,
+                                         const float* __restrict__ beta, int C, int64_t total8,
+                                         int do_relu) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total8;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t off = i << 3;
+    int c0 = (int)(off % C);
+    const shortx8 v = *(const shortx8*)(x + off);
+    shortx8 out;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int c = c0 + j;
+      float f = (bf16_to_f32((bf16raw)v[j]) - mean[c]) * invstd[c] * gamma[c] + beta[c];
+      if (res != nullptr) f += bf16_to_f32(res[off + j]);
+      if (do_relu && f < 0.f) f = 0.f;
+      out[j] = (short)f32_to_bf16(f);
+    }
+    *(shortx8*)(y + off) = out;
+  }
+}
+
 extern "C" hipError_t launch_bn_apply_nhwc(const bf16raw* x, const bf16raw* res, bf16raw* y,
                                            const float* mean, const float* invstd,
                                            const float* gamma, const float* beta, int C,
                                            int64_t total, int do_relu, hipStream_t stream) {
-  int64_t g = ceil_div_i64(total, 256);
-  int grid = (int)(g > 8192 ? 8192 : (g < 1 ? 1 : g));
-  bn_apply_nhwc_kernel<<<grid, 256, 0, stream>>>(x, res, y, mean, invstd, gamma, beta, C, total,
-                                                 do_relu);
+  if ((C & 7) == 0) {
+    int64_t total8 = total >> 3;
+    int64_t g = ceil_div_i64(total8, 256);
+    int grid = (int)(g > 8192 ? 8192 : (g < 1 ? 1 : g));
+    bn_apply_nhwc_vec_kernel<<<grid, 256, 0, stream>>>(x, res, y, mean, invstd, gamma, beta, C,
+                                                       total8, do_relu);
+  } else {
+    int64_t g = ceil_div_i64(total, 256);
+    int grid = (int)(g > 8192 ? 8192 : (g < 1 ? 1 : g));
+    bn_apply_nhwc_kernel<<<grid, 256, 0, stream>>>(x, res, y, mean, invstd, gamma, beta, C,
+                                                   total, do_relu);
+  }
   HIP_CHECK_LAUNCH();
   return hipSuccess;
 }
@@ -204,14 +291,83 @@ __global__ void bn_bwd_reduce_nhwc_kernel(const bf16raw* __restrict__ dy,
   }
 }
 
+__global__ void bn_bwd_reduce_nhwc_vec_kernel(const bf16raw* __restrict__ dy,
+                                              const bf16raw* __restrict__ yrelu,
+                                              const bf16raw* __restrict__ x,
+                                              const float* __restrict__ mean,
+                                              const float* __restrict__ invstd,
+                                              float* __restrict__ dbeta,
+                                              float* __restrict__ dgamma, int64_t M, int C) {
+  int groups = C >> 3;
+  int rpg = 256 / groups;
+  int cg = threadIdx.x % groups;
+  int rg = threadIdx.x / groups;
+  int c0 = cg << 3;
+  int64_t per = ceil_div_i64(M, gridDim.x);
+  int64_t lo = (int64_t)blockIdx.x * per;
+  int64_t hi = lo + per < M ? lo + per : M;
+  float mu[8], is[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    mu[j] = mean[c0 + j];
+    is[j] = invstd[c0 + j];
+  }
+  float sdy[8] = {0, 0, 0, 0, 0, 0, 0, 0}, sdyx[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  for (int64_t r = lo + rg; r < hi; r += rpg) {
+    int64_t off = r * C + c0;
+    const shortx8 g8 = *(const shortx8*)(dy + off);
+    const shortx8 x8 = *(const shortx8*)(x + off);
+    if (yrelu != nullptr) {
+      const shortx8 y8 = *(const shortx8*)(yrelu + off);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float g = bf16_to_f32((bf16raw)y8[j]) > 0.f ? bf16_to_f32((bf16raw)g8[j]) : 0.f;
+        sdy[j] += g;
+        sdyx[j] += g * (bf16_to_f32((bf16raw)x8[j]) - mu[j]) * is[j];
+      }
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float g = bf16_to_f32((bf16raw)g8[j]);
+        sdy[j] += g;
+        sdyx[j] += g * (bf16_to_f32((bf16raw)x8[j]) - mu[j]) * is[j];
+      }
+    }
+  }
+  __shared__ float l0[256][8], l1[256][8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    l0[threadIdx.x][j] = sdy[j];
+    l1[threadIdx.x][j] = sdyx[j];
+  }
+  __syncthreads();
+  if (rg == 0) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float a = 0.f, b = 0.f;
+      for (int q = 0; q < rpg; ++q) {
+        a += l0[q * groups + cg][j];
+        b += l1[q * groups + cg][j];
+      }
+      atomicAdd(&dbeta[c0 + j], a);
+      atomicAdd(&dgamma[c0 + j], b);
+    }
+  }
+}
+
 extern "C" hipError_t launch_bn_bwd_reduce_nhwc(const bf16raw* dy, const bf16raw* yrelu,
                                                 const bf16raw* x, const float* mean,
                                                 const float* invstd, float* dbeta,
                                                 float* dgamma, int64_t M, int C, int nsplit,
                                                 hipStream_t stream) {
-  dim3 grid((C + 63) / 64, nsplit);
-  bn_bwd_reduce_nhwc_kernel<<<grid, 256, 0, stream>>>(dy, yrelu, x, mean, invstd, dbeta, dgamma,
-                                                      M, C);
+  if (bn_vec_ok(C)) {
+    bn_bwd_reduce_nhwc_vec_kernel<<<2048, 256, 0, stream>>>(dy, yrelu, x, mean, invstd, dbeta,
+                                                            dgamma, M, C);
+  } else {
+    dim3 grid((C + 63) / 64, nsplit);
+    bn_bwd_reduce_nhwc_kernel<<<grid, 256, 0, stream>>>(dy, yrelu, x, mean, invstd, dbeta,
+                                                        dgamma, M, C);
+  }
   HIP_CHECK_LAUNCH();
   return hipSuccess;
 }
@@ -242,16 +398,61 @@ __global__ void bn_bwd_dx_nhwc_kernel(const bf16raw* __restrict__ dy,
   }
 }
 
+__global__ void bn_bwd_dx_nhwc_vec_kernel(const bf16raw* __restrict__ dy,
+                                          const bf16raw* __restrict__ yrelu,
+                                          const bf16raw* __restrict__ x,
+                                          const float* __restrict__ mean,
+                                          const float* __restrict__ invstd,
+                                          const float* __restrict__ gamma,
+                                          const float* __restrict__ dbeta,
+                                          const float* __restrict__ dgamma,
+                                          bf16raw* __restrict__ dx, int C, int64_t total8,
+                                          float inv_count, int train_stats) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total8;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t off = i << 3;
+    int c0 = (int)(off % C);
+    const shortx8 g8 = *(const shortx8*)(dy + off);
+    shortx8 out;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int c = c0 + j;
+      float g = bf16_to_f32((bf16raw)g8[j]);
+      if (yrelu != nullptr && bf16_to_f32(yrelu[off + j]) <= 0.f) g = 0.f;
+      float is = invstd[c];
+      float o;
+      if (train_stats) {
+        float xh = (bf16_to_f32(x[off + j]) - mean[c]) * is;
+        o = gamma[c] * is * (g - dbeta[c] * inv_count - xh * dgamma[c] * inv_count);
+      } else {
+        o = gamma[c] * is * g;
+      }
+      out[j] = (short)f32_to_bf16(o);
+    }
+    *(shortx8*)(dx + off) = out;
+  }
+}
+
 extern "C" hipError_t launch_bn_bwd_dx_nhwc(const bf16raw* dy, const bf16raw* yrelu,
                                             const bf16raw* x, const float* mean,
                                             const float* invstd, const float* gamma,
                                             const float* dbeta, const float* dgamma,
                                             bf16raw* dx, int C, int64_t total, float inv_count,
                                             int train_stats, hipStream_t stream) {
-  int64_t g = ceil_div_i64(total, 256);
-  int grid = (int)(g > 8192 ? 8192 : (g < 1 ? 1 : g));
-  bn_bwd_dx_nhwc_kernel<<<grid, 256, 0, stream>>>(dy, yrelu, x, mean, invstd, gamma, dbeta,
-                                                  dgamma, dx, C, total, inv_count, train_stats);
+  if ((C & 7) == 0) {
+    int64_t total8 = total >> 3;
+    int64_t g = ceil_div_i64(total8, 256);
+    int grid = (int)(g > 8192 ? 8192 : (g < 1 ? 1 : g));
+    bn_bwd_dx_nhwc_vec_kernel<<<grid, 256, 0, stream>>>(dy, yrelu, x, mean, invstd, gamma,
+                                                        dbeta, dgamma, dx, C, total8, inv_count,
+                                                        train_stats);
+  } else {
+    int64_t g = ceil_div_i64(total, 256);
+    int grid = (int)(g > 8192 ? 8192 : (g < 1 ? 1 : g));
+    bn_bwd_dx_nhwc_kernel<<<grid, 256, 0, stream>>>(dy, yrelu, x, mean, invstd, gamma, dbeta,
+                                                    dgamma, dx, C, total, inv_count,
+                                                    train_stats);
+  }
   HIP_CHECK_LAUNCH();
   return hipSuccess;
 }
