@@ -160,7 +160,12 @@ __device__ __forceinline__ void stage_tile(const void* __restrict__ src, bf16raw
   }
 }
 
-template <bool B_IS_F32, int EPI, bool SPLITK>
+// Wave arrangement WR x WC (each wave always owns a 64x64 sub-tile = 4x4
+// MFMA fragments): <2,2> = 128x128 block tile (the default), <4,1> = 256x64
+// for narrow-N shapes (64-channel convs, small classifier heads) where half
+// of a 128-wide N tile would be dead.  K-loop is double-buffered: the next
+// BK slab's global loads are issued before the MFMAs on the current slab.
+template <bool B_IS_F32, int EPI, bool SPLITK, int WR, int WC>
 __global__ __launch_bounds__(256) void gemm_kernel(const void* __restrict__ Ap,
                                                    const void* __restrict__ Bp,
                                                    float* __restrict__ Cf,
@@ -169,8 +174,11 @@ __global__ __launch_bounds__(256) void gemm_kernel(const void* __restrict__ Ap,
                                                    int K, int64_t sam, int64_t sak, int64_t sbk,
                                                    int64_t sbn, int k_per_split,
                                                    float* __restrict__ Db, int ones_row) {
-  __shared__ bf16raw As[BM * LDSK];
-  __shared__ bf16raw Bs[BN * LDSK];
+  constexpr int BMt = WR * 64;
+  constexpr int BNt = WC * 64;
+  constexpr int BROWS = BNt < 128 ? 128 : BNt;  // stage_tile writes 128 LDS rows
+  __shared__ bf16raw As[2][BMt * LDSK];
+  __shared__ bf16raw Bs[2][BROWS * LDSK];
 
   // bijective XCD-aware swizzle of the flattened block id (guide §5.5 T1):
   // consecutive output tiles land on one XCD so shared operand rows stay in
@@ -181,8 +189,8 @@ __global__ __launch_bounds__(256) void gemm_kernel(const void* __restrict__ Ap,
   int q = nwg >> 3, rr = nwg & 7;
   int wg = ((orig & 7) < rr ? (orig & 7) * (q + 1) : rr * (q + 1) + ((orig & 7) - rr) * q) +
            (orig >> 3);
-  const int m0 = (wg % gx) * BM;
-  const int n0 = (wg / gx) * BN;
+  const int m0 = (wg % gx) * BMt;
+  const int n0 = (wg / gx) * BNt;
 
   int k_begin = 0, k_end = K;
   if (SPLITK) {
@@ -192,24 +200,34 @@ __global__ __launch_bounds__(256) void gemm_kernel(const void* __restrict__ Ap,
 
   const int wid = threadIdx.x / WAVE;
   const int lane = threadIdx.x % WAVE;
-  const int wr = wid >> 1, wc = wid & 1;
+  const int wr = WC == 1 ? wid : (wid >> 1);
+  const int wc = WC == 1 ? 0 : (wid & 1);
   const int l15 = lane & 15, kg = lane >> 4;
 
   floatx4 acc[4][4] = {};
 
+#define STAGE_SLAB(bufi, kt)                                                              \
+  do {                                                                                    \
+    _Pragma("unroll") for (int ro = 0; ro < BMt; ro += 128)                               \
+        stage_tile<false>(Ap, As[bufi] + ro * LDSK, m0 + ro, M, kt, k_end, sam, sak);     \
+    /* B staged TRANSPOSED: LDS row = n, col = k -> srow := sbn, skol := sbk */           \
+    stage_tile<B_IS_F32>(Bp, Bs[bufi], n0, N, kt, k_end, sbn, sbk, ones_row);             \
+  } while (0)
+
+  int buf = 0;
+  if (k_begin < k_end) STAGE_SLAB(0, k_begin);
+  __syncthreads();
+
   for (int kt = k_begin; kt < k_end; kt += BK) {
-    stage_tile<false>(Ap, As, m0, M, kt, k_end, sam, sak);
-    // B staged TRANSPOSED: LDS row = n, LDS col = k -> srow := sbn, skol := sbk
-    stage_tile<B_IS_F32>(Bp, Bs, n0, N, kt, k_end, sbn, sbk, ones_row);
-    __syncthreads();
+    if (kt + BK < k_end) STAGE_SLAB(buf ^ 1, kt + BK);  // prefetch next slab
 
     frag_t a[4], b[4];
 #pragma unroll
     for (int mi = 0; mi < 4; ++mi)
-      a[mi] = *(const frag_t*)&As[(wr * 64 + mi * 16 + l15) * LDSK + kg * 8];
+      a[mi] = *(const frag_t*)&As[buf][(wr * 64 + mi * 16 + l15) * LDSK + kg * 8];
 #pragma unroll
     for (int ni = 0; ni < 4; ++ni)
-      b[ni] = *(const frag_t*)&Bs[(wc * 64 + ni * 16 + l15) * LDSK + kg * 8];
+      b[ni] = *(const frag_t*)&Bs[buf][(wc * 64 + ni * 16 + l15) * LDSK + kg * 8];
 
 #pragma unroll
     for (int mi = 0; mi < 4; ++mi)
@@ -218,7 +236,9 @@ __global__ __launch_bounds__(256) void gemm_kernel(const void* __restrict__ Ap,
         acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[mi], b[ni], acc[mi][ni], 0, 0, 0);
 
     __syncthreads();
+    buf ^= 1;
   }
+#undef STAGE_SLAB
 
   // epilogue: C/D fragment layout col = lane&15, row = 4*(lane>>4) + reg
   const int m_base = m0 + wr * 64;
@@ -273,11 +293,20 @@ extern "C" hipError_t launch_gemm_bf16(const void* A, const void* B, int b_is_f3
   }
   if (splitk == 1) kps = K;
   const bool atomic = (splitk > 1 || force_atomic);
-  dim3 grid((unsigned)ceil_div_i64(M, BM), (unsigned)ceil_div_i64(N, BN), (unsigned)splitk);
+  // narrow-N shapes: 256x64 tiles (4x1 waves) instead of half-dead 128x128
+  const bool narrow = (N <= 64 && M > 128);
+  const int bm = narrow ? 256 : BM, bn = narrow ? 64 : BN;
+  dim3 grid((unsigned)ceil_div_i64(M, bm), (unsigned)ceil_div_i64(N, bn), (unsigned)splitk);
 
-#define DISPATCH(BF32, EPIC, SPK)                                                          \
-  gemm_kernel<BF32, EPIC, SPK><<<grid, block, 0, stream>>>(A, B, Cf, Cb, bias, M, N, K, sam, \
-                                                           sak, sbk, sbn, kps, Db, ones_row)
+#define DISPATCH(BF32, EPIC, SPK)                                                            \
+  do {                                                                                       \
+    if (narrow)                                                                              \
+      gemm_kernel<BF32, EPIC, SPK, 4, 1><<<grid, block, 0, stream>>>(                        \
+          A, B, Cf, Cb, bias, M, N, K, sam, sak, sbk, sbn, kps, Db, ones_row);               \
+    else                                                                                     \
+      gemm_kernel<BF32, EPIC, SPK, 2, 2><<<grid, block, 0, stream>>>(                        \
+          A, B, Cf, Cb, bias, M, N, K, sam, sak, sbk, sbn, kps, Db, ones_row);               \
+  } while (0)
 
   if (b_is_f32) {
     if (atomic && epi == EPI_F32) DISPATCH(true, EPI_F32, true);
