@@ -55,13 +55,13 @@ hipError_t launch_maxpool_nhwc_bwd(const bf16raw*, const uint8_t*, bf16raw*, int
 hipError_t launch_gap_nhwc_fwd(const bf16raw*, bf16raw*, int, int, int64_t, hipStream_t);
 hipError_t launch_gap_nhwc_bwd(const bf16raw*, bf16raw*, int, int, int64_t, hipStream_t);
 hipError_t launch_bn_stats_nhwc(const bf16raw*, float*, float*, float*, float*, float*, float*,
-                                int64_t, int, float, float, int, hipStream_t);
+                                int64_t, int, float, float, int, float*, int, hipStream_t);
 hipError_t launch_bn_apply_nhwc(const bf16raw*, const bf16raw*, bf16raw*, const float*,
                                 const float*, const float*, const float*, int, int64_t, int,
                                 hipStream_t);
 hipError_t launch_bn_bwd_reduce_nhwc(const bf16raw*, const bf16raw*, const bf16raw*,
                                      const float*, const float*, float*, float*, int64_t, int,
-                                     int, hipStream_t);
+                                     int, float*, int, hipStream_t);
 hipError_t launch_bn_bwd_dx_nhwc(const bf16raw*, const bf16raw*, const bf16raw*, const float*,
                                  const float*, const float*, const float*, const float*,
                                  bf16raw*, int, int64_t, float, int, hipStream_t);
@@ -576,6 +576,15 @@ at::Tensor gap_nhwc_bwd(at::Tensor dy, int64_t H, int64_t W) {
   return dx;
 }
 
+
+// vec reduction path applies when C is a power-of-two in [8, 2048]; S slices
+// sized so each covers ~512 KB (few partial rows, no atomics)
+static bool bn_vec_path(int C) { return C >= 8 && C <= 2048 && (2048 % C) == 0; }
+static int bn_slices(int64_t M, int C) {
+  int64_t s = (M * C * 2) / (256 * 1024);  // ~256 KB per slice
+  return (int)(s < 64 ? 64 : (s > 512 ? 512 : s));
+}
+
 std::tuple<at::Tensor, at::Tensor> bn_stats_nhwc(at::Tensor x,
                                                  c10::optional<at::Tensor> running_mean,
                                                  c10::optional<at::Tensor> running_var,
@@ -598,10 +607,18 @@ std::tuple<at::Tensor, at::Tensor> bn_stats_nhwc(at::Tensor x,
   int tiles = (C + 63) / 64;
   int nsplit = 2048 / (tiles > 0 ? tiles : 1);
   nsplit = nsplit < 1 ? 1 : (nsplit > 256 ? 256 : nsplit);
+  at::Tensor scratch;
+  float* sp = nullptr;
+  int S = 0;
+  if (bn_vec_path(C)) {
+    S = bn_slices(M, C);
+    scratch = at::empty({2LL * S * C}, fopt);
+    sp = scratch.data_ptr<float>();
+  }
   CHECK_HIP(launch_bn_stats_nhwc((const bf16raw*)x.data_ptr(), sum.data_ptr<float>(),
                                  sumsq.data_ptr<float>(), mean.data_ptr<float>(),
                                  invstd.data_ptr<float>(), rm, rv, M, C, (float)momentum,
-                                 (float)eps, nsplit, cur_stream()));
+                                 (float)eps, nsplit, sp, S, cur_stream()));
   return {mean, invstd};
 }
 
@@ -637,10 +654,19 @@ void bn_bwd_reduce_nhwc(at::Tensor dy, c10::optional<at::Tensor> yrelu, at::Tens
   int tiles = (C + 63) / 64;
   int nsplit = 2048 / (tiles > 0 ? tiles : 1);
   nsplit = nsplit < 1 ? 1 : (nsplit > 256 ? 256 : nsplit);
+  at::Tensor scratch;
+  float* sp = nullptr;
+  int S = 0;
+  if (bn_vec_path(C)) {
+    S = bn_slices(M, C);
+    scratch = at::empty({2LL * S * C}, x.options().dtype(at::kFloat));
+    sp = scratch.data_ptr<float>();
+  }
   CHECK_HIP(launch_bn_bwd_reduce_nhwc((const bf16raw*)dy.data_ptr(), yp,
                                       (const bf16raw*)x.data_ptr(), mean.data_ptr<float>(),
                                       invstd.data_ptr<float>(), dbeta.data_ptr<float>(),
-                                      dgamma.data_ptr<float>(), M, C, nsplit, cur_stream()));
+                                      dgamma.data_ptr<float>(), M, C, nsplit, sp, S,
+                                      cur_stream()));
 }
 
 at::Tensor bn_bwd_dx_nhwc(at::Tensor dy, c10::optional<at::Tensor> yrelu, at::Tensor x,
