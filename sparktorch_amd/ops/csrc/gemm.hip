@@ -166,7 +166,7 @@ __device__ __forceinline__ void stage_tile(const void* __restrict__ src, bf16raw
 // of a 128-wide N tile would be dead.  K-loop is double-buffered: the next
 // BK slab's global loads are issued before the MFMAs on the current slab.
 template <bool B_IS_F32, int EPI, bool SPLITK, int WR, int WC>
-__global__ __launch_bounds__(256) void gemm_kernel(const void* __restrict__ Ap,
+__global__ __launch_bounds__(256, 2) void gemm_kernel(const void* __restrict__ Ap,
                                                    const void* __restrict__ Bp,
                                                    float* __restrict__ Cf,
                                                    bf16raw* __restrict__ Cb,
@@ -293,8 +293,10 @@ extern "C" hipError_t launch_gemm_bf16(const void* A, const void* B, int b_is_f3
   }
   if (splitk == 1) kps = K;
   const bool atomic = (splitk > 1 || force_atomic);
-  // narrow-N shapes: 256x64 tiles (4x1 waves) instead of half-dead 128x128
-  const bool narrow = (N <= 64 && M > 128);
+  // 256x64 (4x1-wave) tiles win for narrow N once the kernel is held to
+  // 2 waves/SIMD (tools/gemm_bench.hip: stem fwd 686us vs 754us, l1 conv
+  // 348us vs 388us)
+  const bool narrow = (N <= 64 && M > 256);
   const int bm = narrow ? 256 : BM, bn = narrow ? 64 : BN;
   dim3 grid((unsigned)ceil_div_i64(M, bm), (unsigned)ceil_div_i64(N, bn), (unsigned)splitk);
 
