@@ -170,19 +170,34 @@ __global__ void bn_stats_partial_nhwc_vec_kernel(const bf16raw* __restrict__ x,
   }
 }
 
-// reduce [S, C] partials and finish mean/invstd + running update
+// reduce [S, C] partials and finish mean/invstd + running update.
+// One block per channel: 256 threads stride the S partial rows, LDS tree.
 __global__ void bn_stats_finalize2_kernel(const float* __restrict__ psum,
                                           const float* __restrict__ psumsq, int S,
                                           float* __restrict__ mean, float* __restrict__ invstd,
                                           float* __restrict__ running_mean,
                                           float* __restrict__ running_var, int64_t count,
                                           float momentum, float eps, int C) {
-  for (int c = blockIdx.x * blockDim.x + threadIdx.x; c < C; c += gridDim.x * blockDim.x) {
-    float sum = 0.f, sumsq = 0.f;
-    for (int sidx = 0; sidx < S; ++sidx) {
-      sum += psum[(int64_t)sidx * C + c];
-      sumsq += psumsq[(int64_t)sidx * C + c];
+  int c = blockIdx.x;
+  {
+    float a = 0.f, b = 0.f;
+    for (int sidx = threadIdx.x; sidx < S; sidx += blockDim.x) {
+      a += psum[(int64_t)sidx * C + c];
+      b += psumsq[(int64_t)sidx * C + c];
     }
+    __shared__ float l0[256], l1[256];
+    l0[threadIdx.x] = a;
+    l1[threadIdx.x] = b;
+    __syncthreads();
+    for (int off = blockDim.x >> 1; off > 0; off >>= 1) {
+      if (threadIdx.x < off) {
+        l0[threadIdx.x] += l0[threadIdx.x + off];
+        l1[threadIdx.x] += l1[threadIdx.x + off];
+      }
+      __syncthreads();
+    }
+    if (threadIdx.x != 0) return;
+    float sum = l0[0], sumsq = l1[0];
     float mu = sum / (float)count;
     float var = sumsq / (float)count - mu * mu;
     var = var > 0.f ? var : 0.f;
@@ -210,10 +225,9 @@ extern "C" hipError_t launch_bn_stats_nhwc(const bf16raw* x, float* sum, float* 
     bn_stats_partial_nhwc_vec_kernel<<<S, 256, 0, stream>>>(x, scratch, scratch + (int64_t)S * C,
                                                             M, C);
     HIP_CHECK_LAUNCH();
-    int blocks = (int)ceil_div_i64(C, 256);
-    bn_stats_finalize2_kernel<<<blocks, 256, 0, stream>>>(scratch, scratch + (int64_t)S * C, S,
-                                                          mean, invstd, running_mean,
-                                                          running_var, M, momentum, eps, C);
+    bn_stats_finalize2_kernel<<<C, 256, 0, stream>>>(scratch, scratch + (int64_t)S * C, S,
+                                                     mean, invstd, running_mean, running_var, M,
+                                                     momentum, eps, C);
     HIP_CHECK_LAUNCH();
     return hipSuccess;
   }
@@ -422,14 +436,26 @@ __global__ void bn_bwd_reduce_finalize_kernel(const float* __restrict__ pdb,
                                               const float* __restrict__ pdg, int S,
                                               float* __restrict__ dbeta,
                                               float* __restrict__ dgamma, int C) {
-  for (int c = blockIdx.x * blockDim.x + threadIdx.x; c < C; c += gridDim.x * blockDim.x) {
-    float a = 0.f, b = 0.f;
-    for (int sidx = 0; sidx < S; ++sidx) {
-      a += pdb[(int64_t)sidx * C + c];
-      b += pdg[(int64_t)sidx * C + c];
+  int c = blockIdx.x;
+  float a = 0.f, b = 0.f;
+  for (int sidx = threadIdx.x; sidx < S; sidx += blockDim.x) {
+    a += pdb[(int64_t)sidx * C + c];
+    b += pdg[(int64_t)sidx * C + c];
+  }
+  __shared__ float l0[256], l1[256];
+  l0[threadIdx.x] = a;
+  l1[threadIdx.x] = b;
+  __syncthreads();
+  for (int off = blockDim.x >> 1; off > 0; off >>= 1) {
+    if (threadIdx.x < off) {
+      l0[threadIdx.x] += l0[threadIdx.x + off];
+      l1[threadIdx.x] += l1[threadIdx.x + off];
     }
-    dbeta[c] += a;
-    dgamma[c] += b;
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) {
+    dbeta[c] += l0[0];
+    dgamma[c] += l1[0];
   }
 }
 
@@ -442,10 +468,8 @@ extern "C" hipError_t launch_bn_bwd_reduce_nhwc(const bf16raw* dy, const bf16raw
     bn_bwd_reduce_nhwc_vec_kernel<<<S, 256, 0, stream>>>(dy, yrelu, x, mean, invstd, scratch,
                                                          scratch + (int64_t)S * C, M, C);
     HIP_CHECK_LAUNCH();
-    int blocks = (int)ceil_div_i64(C, 256);
-    bn_bwd_reduce_finalize_kernel<<<blocks, 256, 0, stream>>>(scratch,
-                                                              scratch + (int64_t)S * C, S,
-                                                              dbeta, dgamma, C);
+    bn_bwd_reduce_finalize_kernel<<<C, 256, 0, stream>>>(scratch, scratch + (int64_t)S * C, S,
+                                                         dbeta, dgamma, C);
     HIP_CHECK_LAUNCH();
     return hipSuccess;
   }
