@@ -50,7 +50,7 @@ typedef shortx8 frag_t;  // 8 bf16 (4 VGPRs)
 // ones_row >= 0 marks a VIRTUAL row whose every element is 1.0 (the bias
 // column of dW_ext = dz^T @ [x | 1]); memory is only touched for rows below
 // it.  -1 = no virtual row.
-template <bool SRC_F32>
+template <bool SRC_F32, int LDSTRIDE = LDSK>
 __device__ __forceinline__ void stage_tile(const void* __restrict__ src, bf16raw* __restrict__ lds,
                                            int row0, int rmax, int kt, int kmax, int64_t srow,
                                            int64_t skol, int ones_row = -1) {
@@ -61,7 +61,7 @@ __device__ __forceinline__ void stage_tile(const void* __restrict__ src, bf16raw
     int r = t >> 1;
     int k0 = (t & 1) * 16;
     int gr = row0 + r;
-    bf16raw* dst = lds + r * LDSK + k0;
+    bf16raw* dst = lds + r * LDSTRIDE + k0;
     if (ones_row >= 0 && gr == ones_row) {
 #pragma unroll
       for (int j = 0; j < 16; ++j) dst[j] = (kt + k0 + j < kmax) ? BF16_ONE : (bf16raw)0;
@@ -112,7 +112,7 @@ __device__ __forceinline__ void stage_tile(const void* __restrict__ src, bf16raw
           for (int v = 0; v < 4; ++v) {
             floatx4 x = *(const floatx4*)(s + v * 4);
 #pragma unroll
-            for (int j = 0; j < 4; ++j) lds[(r0 + v * 4 + j) * LDSK + k] = f32_to_bf16(x[j]);
+            for (int j = 0; j < 4; ++j) lds[(r0 + v * 4 + j) * LDSTRIDE + k] = f32_to_bf16(x[j]);
           }
         } else {
           const bf16raw* s = (const bf16raw*)base + row0 + r0;
@@ -120,7 +120,7 @@ __device__ __forceinline__ void stage_tile(const void* __restrict__ src, bf16raw
           for (int v = 0; v < 2; ++v) {
             shortx8 x = *(const shortx8*)(s + v * 8);
 #pragma unroll
-            for (int j = 0; j < 8; ++j) lds[(r0 + v * 8 + j) * LDSK + k] = (bf16raw)x[j];
+            for (int j = 0; j < 8; ++j) lds[(r0 + v * 8 + j) * LDSTRIDE + k] = (bf16raw)x[j];
           }
         }
       } else {
@@ -133,12 +133,12 @@ __device__ __forceinline__ void stage_tile(const void* __restrict__ src, bf16raw
           else if (j < mem_rem)  // row0 + r < mem_rows
             v = SRC_F32 ? f32_to_bf16(((const float*)base)[row0 + r])
                         : ((const bf16raw*)base)[row0 + r];
-          lds[r * LDSK + k] = v;
+          lds[r * LDSTRIDE + k] = v;
         }
       }
     } else {
 #pragma unroll
-      for (int j = 0; j < 16; ++j) lds[(r0 + j) * LDSK + k] = 0;
+      for (int j = 0; j < 16; ++j) lds[(r0 + j) * LDSTRIDE + k] = 0;
     }
   } else {
     // generic scalar path
@@ -155,17 +155,71 @@ __device__ __forceinline__ void stage_tile(const void* __restrict__ src, bf16raw
         const char* p = (const char*)src + ((int64_t)gr * srow + (int64_t)gk * skol) * (SRC_F32 ? 4 : 2);
         v = SRC_F32 ? f32_to_bf16(*(const float*)p) : *(const bf16raw*)p;
       }
-      lds[r * LDSK + k] = v;
+      lds[r * LDSTRIDE + k] = v;
     }
   }
 }
 
-// Wave arrangement WR x WC (each wave always owns a 64x64 sub-tile = 4x4
-// MFMA fragments): <2,2> = 128x128 block tile (the default), <4,1> = 256x64
-// for narrow-N shapes (64-channel convs, small classifier heads) where half
-// of a 128-wide N tile would be dead.  K-loop is double-buffered: the next
-// BK slab's global loads are issued before the MFMAs on the current slab.
-template <bool B_IS_F32, int EPI, bool SPLITK, int WR, int WC>
+// ---------------------------------------------------------------------------
+// Direct global->LDS staging (glds) for the k-contiguous A operand.  The LDS
+// image is pad-free [ROWS][BKT] bf16 with the 16-byte k-slot of each row
+// XOR-swizzled by (row & (SLOTS-1)) — lane-linear for the DMA (dest is
+// wave-uniform base + lane*16; guide §5 "Async global->LDS"), bank-spread
+// for the ds_read_b128 fragment reads.  The swizzle permutes 16B pieces
+// within one 128B line of the source row, so global coalescing is kept.
+// ---------------------------------------------------------------------------
+
+template <int ROWS, int BKT>
+__device__ __forceinline__ void stage_glds(const bf16raw* __restrict__ src,
+                                           bf16raw* __restrict__ lds, int row0, int64_t sam,
+                                           int kt) {
+  constexpr int SLOTS = BKT / 8;       // 16B slots per LDS row
+  constexpr int CH_ROWS = 64 / SLOTS;  // rows per 1KB wave chunk
+  constexpr int NCHUNK = ROWS / CH_ROWS;
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int r_in = lane / SLOTS;
+  const int slot = lane % SLOTS;
+#pragma unroll
+  for (int c = wid; c < NCHUNK; c += 4) {
+    int row = c * CH_ROWS + r_in;
+    int sslot = slot ^ (row & (SLOTS - 1));
+    const bf16raw* g = src + (int64_t)(row0 + row) * sam + kt + sslot * 8;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)g,
+        (__attribute__((address_space(3))) unsigned int*)(lds + c * CH_ROWS * BKT), 16, 0, 0);
+  }
+}
+
+// boundary-slab fallback writing the SAME swizzled image with zero fill
+template <int ROWS, int BKT>
+__device__ __forceinline__ void stage_swz_fallback(const bf16raw* __restrict__ src,
+                                                   bf16raw* __restrict__ lds, int row0, int rmax,
+                                                   int kt, int kmax, int64_t sam) {
+  constexpr int SLOTS = BKT / 8;
+  constexpr int E = ROWS * BKT / 256;
+  int base = threadIdx.x * E;
+#pragma unroll
+  for (int j = 0; j < E; ++j) {
+    int idx = base + j;
+    int r = idx / BKT, k = idx % BKT;
+    bf16raw v = 0;
+    int gr = row0 + r, gk = kt + k;
+    if (gr < rmax && gk < kmax) v = src[(int64_t)gr * sam + gk];
+    lds[r * BKT + (((k >> 3) ^ (r & (SLOTS - 1))) << 3) + (k & 7)] = v;
+  }
+}
+
+// Wave arrangement WR x WC (each wave owns a 64x64 sub-tile = 4x4 MFMA
+// fragments): <2,2> = 128x128 block tile with BK=64, <4,1> = 256x64 with
+// BK=32 for narrow-N shapes.  The A operand (the big streamed matrix) is
+// staged by async global_load_lds DMA into a pad-free XOR-swizzled image
+// when it is k-contiguous (AG) — the guide's ladder measured 517->874 TF
+// from exactly this change on the 128^2-tile structure — and by the padded
+// register path otherwise (wgrad, where the reduction runs over the outer
+// stride of both operands).  B (small, L2-hot, possibly fp32 master
+// weights) always uses the register path.  K-loop is double-buffered.
+template <bool B_IS_F32, int EPI, bool SPLITK, int WR, int WC, bool AG>
 __global__ __launch_bounds__(256, 2) void gemm_kernel(const void* __restrict__ Ap,
                                                    const void* __restrict__ Bp,
                                                    float* __restrict__ Cf,
@@ -176,9 +230,13 @@ __global__ __launch_bounds__(256, 2) void gemm_kernel(const void* __restrict__ A
                                                    float* __restrict__ Db, int ones_row) {
   constexpr int BMt = WR * 64;
   constexpr int BNt = WC * 64;
+  constexpr int BKT = (WC == 1) ? 32 : 64;  // narrow tile keeps BK=32 (LDS budget)
+  constexpr int SUBS = BKT / 32;            // MFMA k-steps per slab
+  constexpr int SLOTS = BKT / 8;
+  constexpr int LP = BKT + LDS_PAD;         // padded stride (register-staged images)
   constexpr int BROWS = BNt < 128 ? 128 : BNt;  // stage_tile writes 128 LDS rows
-  __shared__ bf16raw As[2][BMt * LDSK];
-  __shared__ bf16raw Bs[2][BROWS * LDSK];
+  __shared__ bf16raw As[2][AG ? (BMt * BKT) : (BMt * LP)];
+  __shared__ bf16raw Bs[2][BROWS * LP];
 
   // bijective XCD-aware swizzle of the flattened block id (guide §5.5 T1):
   // consecutive output tiles land on one XCD so shared operand rows stay in
@@ -204,36 +262,61 @@ __global__ __launch_bounds__(256, 2) void gemm_kernel(const void* __restrict__ A
   const int wc = WC == 1 ? 0 : (wid & 1);
   const int l15 = lane & 15, kg = lane >> 4;
 
+  const bool a_rows_full = (m0 + BMt <= M);
+
   floatx4 acc[4][4] = {};
 
-#define STAGE_SLAB(bufi, kt)                                                              \
-  do {                                                                                    \
-    _Pragma("unroll") for (int ro = 0; ro < BMt; ro += 128)                               \
-        stage_tile<false>(Ap, As[bufi] + ro * LDSK, m0 + ro, M, kt, k_end, sam, sak);     \
-    /* B staged TRANSPOSED: LDS row = n, col = k -> srow := sbn, skol := sbk */           \
-    stage_tile<B_IS_F32>(Bp, Bs[bufi], n0, N, kt, k_end, sbn, sbk, ones_row);             \
+#define STAGE_SLAB(bufi, kt)                                                               \
+  do {                                                                                     \
+    if (AG) {                                                                              \
+      if (a_rows_full && (kt) + BKT <= k_end)                                              \
+        stage_glds<BMt, BKT>((const bf16raw*)Ap, As[bufi], m0, sam, kt);                   \
+      else                                                                                 \
+        stage_swz_fallback<BMt, BKT>((const bf16raw*)Ap, As[bufi], m0, M, kt, k_end, sam); \
+    } else {                                                                               \
+      _Pragma("unroll") for (int ro = 0; ro < BMt; ro += 128) {                            \
+        _Pragma("unroll") for (int kh = 0; kh < BKT; kh += 32)                             \
+            stage_tile<false, LP>(Ap, As[bufi] + ro * LP + kh, m0 + ro, M, (kt) + kh,      \
+                                  k_end, sam, sak);                                        \
+      }                                                                                    \
+    }                                                                                      \
+    /* B staged TRANSPOSED: LDS row = n, col = k -> srow := sbn, skol := sbk */            \
+    _Pragma("unroll") for (int kh = 0; kh < BKT; kh += 32)                                 \
+        stage_tile<B_IS_F32, LP>(Bp, Bs[bufi] + kh, n0, N, (kt) + kh, k_end, sbn, sbk,     \
+                                 ones_row);                                                \
   } while (0)
 
   int buf = 0;
   if (k_begin < k_end) STAGE_SLAB(0, k_begin);
   __syncthreads();
 
-  for (int kt = k_begin; kt < k_end; kt += BK) {
-    if (kt + BK < k_end) STAGE_SLAB(buf ^ 1, kt + BK);  // prefetch next slab
-
-    frag_t a[4], b[4];
-#pragma unroll
-    for (int mi = 0; mi < 4; ++mi)
-      a[mi] = *(const frag_t*)&As[buf][(wr * 64 + mi * 16 + l15) * LDSK + kg * 8];
-#pragma unroll
-    for (int ni = 0; ni < 4; ++ni)
-      b[ni] = *(const frag_t*)&Bs[buf][(wc * 64 + ni * 16 + l15) * LDSK + kg * 8];
+  for (int kt = k_begin; kt < k_end; kt += BKT) {
+    if (kt + BKT < k_end) STAGE_SLAB(buf ^ 1, kt + BKT);  // prefetch next slab
 
 #pragma unroll
-    for (int mi = 0; mi < 4; ++mi)
+    for (int sub = 0; sub < SUBS; ++sub) {
+      frag_t a[4], b[4];
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi) {
+        int row = wr * 64 + mi * 16 + l15;
+        if (AG) {
+          int kq = kg + sub * 4;
+          a[mi] = *(const frag_t*)&As[buf][row * BKT + (((kq) ^ (row & (SLOTS - 1))) << 3)];
+        } else {
+          a[mi] = *(const frag_t*)&As[buf][row * LP + kg * 8 + sub * 32];
+        }
+      }
 #pragma unroll
       for (int ni = 0; ni < 4; ++ni)
-        acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[mi], b[ni], acc[mi][ni], 0, 0, 0);
+        b[ni] = *(const frag_t*)&Bs[buf][(wc * 64 + ni * 16 + l15) * LP + kg * 8 + sub * 32];
+
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          acc[mi][ni] =
+              __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[mi], b[ni], acc[mi][ni], 0, 0, 0);
+    }
 
     __syncthreads();
     buf ^= 1;
@@ -288,7 +371,8 @@ extern "C" hipError_t launch_gemm_bf16(const void* A, const void* B, int b_is_f3
   int kps = 0;
   if (splitk < 1) splitk = 1;
   if (splitk > 1) {
-    kps = (int)ceil_div_i64(ceil_div_i64(K, splitk), BK) * BK;
+    // slices are multiples of 64 so both BK=64 and BK=32 variants align
+    kps = (int)ceil_div_i64(ceil_div_i64(K, splitk), 64) * 64;
     splitk = (int)ceil_div_i64(K, kps);
   }
   if (splitk == 1) kps = K;
@@ -300,13 +384,22 @@ extern "C" hipError_t launch_gemm_bf16(const void* A, const void* B, int b_is_f3
   const int bm = narrow ? 256 : BM, bn = narrow ? 64 : BN;
   dim3 grid((unsigned)ceil_div_i64(M, bm), (unsigned)ceil_div_i64(N, bn), (unsigned)splitk);
 
+  // A is DMA-staged (glds) when k-contiguous with 16B-aligned rows
+  const bool ag = (sak == 1) && (sam % 8 == 0);
+
 #define DISPATCH(BF32, EPIC, SPK)                                                            \
   do {                                                                                       \
-    if (narrow)                                                                              \
-      gemm_kernel<BF32, EPIC, SPK, 4, 1><<<grid, block, 0, stream>>>(                        \
+    if (narrow && ag)                                                                        \
+      gemm_kernel<BF32, EPIC, SPK, 4, 1, true><<<grid, block, 0, stream>>>(                  \
+          A, B, Cf, Cb, bias, M, N, K, sam, sak, sbk, sbn, kps, Db, ones_row);               \
+    else if (narrow)                                                                         \
+      gemm_kernel<BF32, EPIC, SPK, 4, 1, false><<<grid, block, 0, stream>>>(                 \
+          A, B, Cf, Cb, bias, M, N, K, sam, sak, sbk, sbn, kps, Db, ones_row);               \
+    else if (ag)                                                                             \
+      gemm_kernel<BF32, EPIC, SPK, 2, 2, true><<<grid, block, 0, stream>>>(                  \
           A, B, Cf, Cb, bias, M, N, K, sam, sak, sbk, sbn, kps, Db, ones_row);               \
     else                                                                                     \
-      gemm_kernel<BF32, EPIC, SPK, 2, 2><<<grid, block, 0, stream>>>(                        \
+      gemm_kernel<BF32, EPIC, SPK, 2, 2, false><<<grid, block, 0, stream>>>(                 \
           A, B, Cf, Cb, bias, M, N, K, sam, sak, sbk, sbn, kps, Db, ones_row);               \
   } while (0)
 

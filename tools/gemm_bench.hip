@@ -17,7 +17,7 @@
     }                                                                          \
   } while (0)
 
-template <int WR, int WC>
+template <int WR, int WC, bool AG>
 double run_fwd(const bf16raw* A, const bf16raw* B, bf16raw* C, int M, int N, int K, int iters) {
   dim3 grid((unsigned)((M + WR * 64 - 1) / (WR * 64)), (unsigned)((N + WC * 64 - 1) / (WC * 64)),
             1);
@@ -25,7 +25,7 @@ double run_fwd(const bf16raw* A, const bf16raw* B, bf16raw* C, int M, int N, int
   CK(hipEventCreate(&a));
   CK(hipEventCreate(&b));
   auto launch = [&]() {
-    gemm_kernel<false, EPI_BF16, false, WR, WC><<<grid, 256>>>(
+    gemm_kernel<false, EPI_BF16, false, WR, WC, AG><<<grid, 256>>>(
         A, B, nullptr, C, nullptr, M, N, K, /*sam*/ K, /*sak*/ 1, /*sbk*/ 1, /*sbn*/ K, K,
         nullptr, -1);
   };
@@ -52,7 +52,7 @@ double run_wgrad(const bf16raw* dz, const bf16raw* col, float* dw, int CO, int K
   auto launch = [&]() {
     // A[m=co, k=row] = dz[row*CO+co] -> sam=1, sak=CO ; B[k=row, n=kc] = col[row*Kcol+kc]
     // staged transposed: srow := sbn=1, skol := sbk=Kcol
-    gemm_kernel<false, EPI_F32, true, 2, 2><<<grid, 256>>>(
+    gemm_kernel<false, EPI_F32, true, 2, 2, false><<<grid, 256>>>(
         dz, col, dw, nullptr, nullptr, CO, Kcol, (int)R, /*sam*/ 1, /*sak*/ CO, /*sbk*/ Kcol,
         /*sbn*/ 1, kps, nullptr, -1);
   };
@@ -87,8 +87,12 @@ int main() {
     CK(hipMalloc(&C, cn * 2));
     CK(hipMemset(A, 0x3c, an * 2));
     CK(hipMemset(B, 0x3c, bn * 2));
-    double t22 = run_fwd<2, 2>(A, B, C, s.M, s.N, s.K, 10);
-    double t41 = s.N <= 64 ? run_fwd<4, 1>(A, B, C, s.M, s.N, s.K, 10) : -1;
+    bool ag = (s.K % 8) == 0;
+    double t22 = ag ? run_fwd<2, 2, true>(A, B, C, s.M, s.N, s.K, 10)
+                    : run_fwd<2, 2, false>(A, B, C, s.M, s.N, s.K, 10);
+    double t41 = s.N <= 64 ? (ag ? run_fwd<4, 1, true>(A, B, C, s.M, s.N, s.K, 10)
+                                 : run_fwd<4, 1, false>(A, B, C, s.M, s.N, s.K, 10))
+                           : -1;
     double fl = 2.0 * s.M * s.N * s.K;
     printf("%s M=%8d N=%4d K=%5d | 2x2 %8.1fus (%5.0f TF)  4x1 %8.1fus (%5.0f TF)\n", s.name,
            s.M, s.N, s.K, t22, fl / t22 / 1e6, t41, t41 > 0 ? fl / t41 / 1e6 : 0);
