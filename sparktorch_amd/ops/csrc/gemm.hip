@@ -160,6 +160,137 @@ __device__ __forceinline__ void stage_tile(const void* __restrict__ src, bf16raw
   }
 }
 
+// Split staging (guide T14): stage_load fills 16 per-thread registers for a
+// 32-k chunk; stage_write stores them to the padded LDS image.  Holding the
+// slab in registers lets the global loads for slab t+1 be issued right
+// after the barrier and span the whole MFMA section of slab t, instead of
+// stalling on vmcnt just before it — this is the latency fix for the wgrad
+// shapes where neither operand is k-contiguous (no glds possible).
+template <bool SRC_F32>
+__device__ __forceinline__ void stage_load(const void* __restrict__ src, int row0, int rmax,
+                                           int kt, int kmax, int64_t srow, int64_t skol,
+                                           int ones_row, bf16raw* __restrict__ regs) {
+  const int t = threadIdx.x;
+  const int mem_rows = ones_row >= 0 ? ones_row : rmax;
+  if (skol == 1) {
+    int r = t >> 1;
+    int k0 = (t & 1) * 16;
+    int gr = row0 + r;
+    if (ones_row >= 0 && gr == ones_row) {
+#pragma unroll
+      for (int j = 0; j < 16; ++j) regs[j] = (kt + k0 + j < kmax) ? BF16_ONE : (bf16raw)0;
+    } else if (gr < mem_rows) {
+      const char* base = (const char*)src + (int64_t)gr * srow * (SRC_F32 ? 4 : 2);
+      int krem = kmax - kt - k0;
+      if (krem >= 16) {
+        if (SRC_F32) {
+          const float* sp = (const float*)base + kt + k0;
+#pragma unroll
+          for (int v = 0; v < 4; ++v) {
+            floatx4 x = *(const floatx4*)(sp + v * 4);
+#pragma unroll
+            for (int j = 0; j < 4; ++j) regs[v * 4 + j] = f32_to_bf16(x[j]);
+          }
+        } else {
+          const bf16raw* sp = (const bf16raw*)base + kt + k0;
+          *(shortx8*)regs = *(const shortx8*)sp;
+          *(shortx8*)(regs + 8) = *(const shortx8*)(sp + 8);
+        }
+      } else {
+#pragma unroll
+        for (int j = 0; j < 16; ++j) {
+          int k = kt + k0 + j;
+          regs[j] = (k < kmax)
+                        ? (SRC_F32 ? f32_to_bf16(((const float*)base)[k]) : ((const bf16raw*)base)[k])
+                        : (bf16raw)0;
+        }
+      }
+    } else {
+#pragma unroll
+      for (int j = 0; j < 16; ++j) regs[j] = 0;
+    }
+  } else if (srow == 1) {
+    int k = t >> 3;
+    int r0 = (t & 7) * 16;
+    int gk = kt + k;
+    if (gk < kmax) {
+      const char* base = (const char*)src + (int64_t)gk * skol * (SRC_F32 ? 4 : 2);
+      int mem_rem = mem_rows - row0 - r0;
+      if (mem_rem >= 16) {
+        if (SRC_F32) {
+          const float* sp = (const float*)base + row0 + r0;
+#pragma unroll
+          for (int v = 0; v < 4; ++v) {
+            floatx4 x = *(const floatx4*)(sp + v * 4);
+#pragma unroll
+            for (int j = 0; j < 4; ++j) regs[v * 4 + j] = f32_to_bf16(x[j]);
+          }
+        } else {
+          const bf16raw* sp = (const bf16raw*)base + row0 + r0;
+          *(shortx8*)regs = *(const shortx8*)sp;
+          *(shortx8*)(regs + 8) = *(const shortx8*)(sp + 8);
+        }
+      } else {
+#pragma unroll
+        for (int j = 0; j < 16; ++j) {
+          bf16raw v = 0;
+          if (ones_row >= 0 && row0 + r0 + j == ones_row)
+            v = BF16_ONE;
+          else if (j < mem_rem)
+            v = SRC_F32 ? f32_to_bf16(((const float*)base)[row0 + r0 + j])
+                        : ((const bf16raw*)base)[row0 + r0 + j];
+          regs[j] = v;
+        }
+      }
+    } else {
+#pragma unroll
+      for (int j = 0; j < 16; ++j) regs[j] = 0;
+    }
+  } else {
+    int idx0 = t * 16;
+#pragma unroll
+    for (int j = 0; j < 16; ++j) {
+      int idx = idx0 + j;
+      int r = idx / BK, k = idx % BK;
+      int gr = row0 + r, gk = kt + k;
+      bf16raw v = 0;
+      if (ones_row >= 0 && gr == ones_row && gk < kmax) {
+        v = BF16_ONE;
+      } else if (gr < mem_rows && gk < kmax) {
+        const char* p =
+            (const char*)src + ((int64_t)gr * srow + (int64_t)gk * skol) * (SRC_F32 ? 4 : 2);
+        v = SRC_F32 ? f32_to_bf16(*(const float*)p) : *(const bf16raw*)p;
+      }
+      regs[j] = v;
+    }
+  }
+}
+
+template <int LDSTRIDE>
+__device__ __forceinline__ void stage_write(bf16raw* __restrict__ lds, int64_t srow,
+                                            int64_t skol, const bf16raw* __restrict__ regs) {
+  const int t = threadIdx.x;
+  if (skol == 1) {
+    int r = t >> 1;
+    int k0 = (t & 1) * 16;
+    bf16raw* dst = lds + r * LDSTRIDE + k0;
+    *(shortx8*)dst = *(const shortx8*)regs;
+    *(shortx8*)(dst + 8) = *(const shortx8*)(regs + 8);
+  } else if (srow == 1) {
+    int k = t >> 3;
+    int r0 = (t & 7) * 16;
+#pragma unroll
+    for (int j = 0; j < 16; ++j) lds[(r0 + j) * LDSTRIDE + k] = regs[j];
+  } else {
+    int idx0 = t * 16;
+#pragma unroll
+    for (int j = 0; j < 16; ++j) {
+      int idx = idx0 + j;
+      lds[(idx / BK) * LDSTRIDE + (idx % BK)] = regs[j];
+    }
+  }
+}
+
 // ---------------------------------------------------------------------------
 // Direct global->LDS staging (glds) for the k-contiguous A operand.  The LDS
 // image is pad-free [ROWS][BKT] bf16 with the 16-byte k-slot of each row
@@ -235,8 +366,9 @@ __global__ __launch_bounds__(256, 2) void gemm_kernel(const void* __restrict__ A
   constexpr int SLOTS = BKT / 8;
   constexpr int LP = BKT + LDS_PAD;         // padded stride (register-staged images)
   constexpr int BROWS = BNt < 128 ? 128 : BNt;  // stage_tile writes 128 LDS rows
-  __shared__ bf16raw As[2][AG ? (BMt * BKT) : (BMt * LP)];
-  __shared__ bf16raw Bs[2][BROWS * LP];
+  constexpr int NBUF = AG ? 2 : 1;  // T14 register pipeline needs one buffer
+  __shared__ bf16raw As[NBUF][AG ? (BMt * BKT) : (BMt * LP)];
+  __shared__ bf16raw Bs[NBUF][BROWS * LP];
 
   // bijective XCD-aware swizzle of the flattened block id (guide §5.5 T1):
   // consecutive output tiles land on one XCD so shared operand rows stay in
@@ -266,62 +398,111 @@ __global__ __launch_bounds__(256, 2) void gemm_kernel(const void* __restrict__ A
 
   floatx4 acc[4][4] = {};
 
+  if (AG) {
+    // A by async DMA (double-buffered prefetch), B by direct register stage.
 #define STAGE_SLAB(bufi, kt)                                                               \
   do {                                                                                     \
-    if (AG) {                                                                              \
-      if (a_rows_full && (kt) + BKT <= k_end)                                              \
-        stage_glds<BMt, BKT>((const bf16raw*)Ap, As[bufi], m0, sam, kt);                   \
-      else                                                                                 \
-        stage_swz_fallback<BMt, BKT>((const bf16raw*)Ap, As[bufi], m0, M, kt, k_end, sam); \
-    } else {                                                                               \
-      _Pragma("unroll") for (int ro = 0; ro < BMt; ro += 128) {                            \
-        _Pragma("unroll") for (int kh = 0; kh < BKT; kh += 32)                             \
-            stage_tile<false, LP>(Ap, As[bufi] + ro * LP + kh, m0 + ro, M, (kt) + kh,      \
-                                  k_end, sam, sak);                                        \
-      }                                                                                    \
-    }                                                                                      \
+    if (a_rows_full && (kt) + BKT <= k_end)                                                \
+      stage_glds<BMt, BKT>((const bf16raw*)Ap, As[bufi], m0, sam, kt);                     \
+    else                                                                                   \
+      stage_swz_fallback<BMt, BKT>((const bf16raw*)Ap, As[bufi], m0, M, kt, k_end, sam);   \
     /* B staged TRANSPOSED: LDS row = n, col = k -> srow := sbn, skol := sbk */            \
     _Pragma("unroll") for (int kh = 0; kh < BKT; kh += 32)                                 \
-        stage_tile<B_IS_F32, LP>(Bp, Bs[bufi] + kh, n0, N, (kt) + kh, k_end, sbn, sbk,     \
-                                 ones_row);                                                \
+        stage_tile<B_IS_F32, LP>(Bp, Bs[1 && (bufi)] + kh, n0, N, (kt) + kh, k_end, sbn,   \
+                                 sbk, ones_row);                                           \
   } while (0)
 
-  int buf = 0;
-  if (k_begin < k_end) STAGE_SLAB(0, k_begin);
-  __syncthreads();
+    int buf = 0;
+    if (k_begin < k_end) STAGE_SLAB(0, k_begin);
+    __syncthreads();
 
-  for (int kt = k_begin; kt < k_end; kt += BKT) {
-    if (kt + BKT < k_end) STAGE_SLAB(buf ^ 1, kt + BKT);  // prefetch next slab
+    for (int kt = k_begin; kt < k_end; kt += BKT) {
+      if (kt + BKT < k_end) STAGE_SLAB(buf ^ 1, kt + BKT);  // prefetch next slab
 
 #pragma unroll
-    for (int sub = 0; sub < SUBS; ++sub) {
-      frag_t a[4], b[4];
+      for (int sub = 0; sub < SUBS; ++sub) {
+        frag_t a[4], b[4];
 #pragma unroll
-      for (int mi = 0; mi < 4; ++mi) {
-        int row = wr * 64 + mi * 16 + l15;
-        if (AG) {
+        for (int mi = 0; mi < 4; ++mi) {
+          int row = wr * 64 + mi * 16 + l15;
           int kq = kg + sub * 4;
           a[mi] = *(const frag_t*)&As[buf][row * BKT + (((kq) ^ (row & (SLOTS - 1))) << 3)];
-        } else {
-          a[mi] = *(const frag_t*)&As[buf][row * LP + kg * 8 + sub * 32];
         }
-      }
-#pragma unroll
-      for (int ni = 0; ni < 4; ++ni)
-        b[ni] = *(const frag_t*)&Bs[buf][(wc * 64 + ni * 16 + l15) * LP + kg * 8 + sub * 32];
-
-#pragma unroll
-      for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
         for (int ni = 0; ni < 4; ++ni)
-          acc[mi][ni] =
-              __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[mi], b[ni], acc[mi][ni], 0, 0, 0);
-    }
+          b[ni] = *(const frag_t*)&Bs[buf][(wc * 64 + ni * 16 + l15) * LP + kg * 8 + sub * 32];
+#pragma unroll
+        for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+          for (int ni = 0; ni < 4; ++ni)
+            acc[mi][ni] =
+                __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[mi], b[ni], acc[mi][ni], 0, 0, 0);
+      }
 
-    __syncthreads();
-    buf ^= 1;
-  }
+      __syncthreads();
+      buf ^= 1;
+    }
 #undef STAGE_SLAB
+  } else {
+    // T14 register pipeline, single LDS buffer: hold slab t in registers,
+    // write it to LDS after the barrier, immediately issue slab t+1's
+    // loads, then run slab t's MFMAs while those loads fly.
+    constexpr int AH = (BMt / 128) * (BKT / 32);  // 32-k x 128-row chunks of A
+    constexpr int BH = BKT / 32;
+    alignas(16) bf16raw rA[AH * 16];
+    alignas(16) bf16raw rB[BH * 16];
+
+#define LOAD_SLAB(kt)                                                                      \
+  do {                                                                                     \
+    _Pragma("unroll") for (int ro = 0; ro < BMt / 128; ++ro) {                             \
+      _Pragma("unroll") for (int kh = 0; kh < BKT / 32; ++kh)                              \
+          stage_load<false>(Ap, m0 + ro * 128, M, (kt) + kh * 32, k_end, sam, sak, -1,     \
+                            rA + (ro * (BKT / 32) + kh) * 16);                             \
+    }                                                                                      \
+    _Pragma("unroll") for (int kh = 0; kh < BKT / 32; ++kh)                                \
+        stage_load<B_IS_F32>(Bp, n0, N, (kt) + kh * 32, k_end, sbn, sbk, ones_row,         \
+                             rB + kh * 16);                                                \
+  } while (0)
+
+    if (k_begin < k_end) LOAD_SLAB(k_begin);
+
+    for (int kt = k_begin; kt < k_end; kt += BKT) {
+      if (kt > k_begin) __syncthreads();  // prior slab's MFMA reads done
+#pragma unroll
+      for (int ro = 0; ro < BMt / 128; ++ro) {
+#pragma unroll
+        for (int kh = 0; kh < BKT / 32; ++kh)
+          stage_write<LP>(As[0] + ro * 128 * LP + kh * 32, sam, sak,
+                          rA + (ro * (BKT / 32) + kh) * 16);
+      }
+#pragma unroll
+      for (int kh = 0; kh < BKT / 32; ++kh)
+        stage_write<LP>(Bs[0] + kh * 32, sbn, sbk, rB + kh * 16);
+      __syncthreads();  // publish
+
+      if (kt + BKT < k_end) LOAD_SLAB(kt + BKT);  // loads span the MFMAs below
+
+#pragma unroll
+      for (int sub = 0; sub < SUBS; ++sub) {
+        frag_t a[4], b[4];
+#pragma unroll
+        for (int mi = 0; mi < 4; ++mi) {
+          int row = wr * 64 + mi * 16 + l15;
+          a[mi] = *(const frag_t*)&As[0][row * LP + kg * 8 + sub * 32];
+        }
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          b[ni] = *(const frag_t*)&Bs[0][(wc * 64 + ni * 16 + l15) * LP + kg * 8 + sub * 32];
+#pragma unroll
+        for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+          for (int ni = 0; ni < 4; ++ni)
+            acc[mi][ni] =
+                __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[mi], b[ni], acc[mi][ni], 0, 0, 0);
+      }
+    }
+#undef LOAD_SLAB
+  }
 
   // epilogue: C/D fragment layout col = lane&15, row = 4*(lane>>4) + reg
   const int m_base = m0 + wr * 64;
