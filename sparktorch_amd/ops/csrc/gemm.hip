@@ -47,6 +47,17 @@ typedef shortx8 frag_t;  // 8 bf16 (4 VGPRs)
 //   else: scalar element loop.
 #define BF16_ONE ((bf16raw)0x3F80)
 
+// Granule swizzle for the PADDED LDS images: XOR the 8-element (16 B)
+// granule index of a 32-k half by row bits >= 4.  The srow-staging write
+// pattern (16-row stride between a wave's lanes) lands every lane on the
+// same bank group whatever the row pad (16-row x any-16B-aligned stride is
+// 0 mod the 256 B bank period); the XOR spreads it 8-way -> ~2-way.  Lane
+// groups of a fragment read share (row >> 4), so reads stay 16 B
+// contiguous and conflict-free.
+__device__ __forceinline__ int swz_col(int row, int col) {
+  return col ^ (((row >> 4) & 3) << 3);
+}
+
 // ones_row >= 0 marks a VIRTUAL row whose every element is 1.0 (the bias
 // column of dW_ext = dz^T @ [x | 1]); memory is only touched for rows below
 // it.  -1 = no virtual row.
@@ -61,10 +72,13 @@ __device__ __forceinline__ void stage_tile(const void* __restrict__ src, bf16raw
     int r = t >> 1;
     int k0 = (t & 1) * 16;
     int gr = row0 + r;
-    bf16raw* dst = lds + r * LDSTRIDE + k0;
+    bf16raw* d0 = lds + r * LDSTRIDE + swz_col(r, k0);
+    bf16raw* d1 = lds + r * LDSTRIDE + swz_col(r, k0 + 8);
     if (ones_row >= 0 && gr == ones_row) {
 #pragma unroll
-      for (int j = 0; j < 16; ++j) dst[j] = (kt + k0 + j < kmax) ? BF16_ONE : (bf16raw)0;
+      for (int j = 0; j < 8; ++j) d0[j] = (kt + k0 + j < kmax) ? BF16_ONE : (bf16raw)0;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) d1[j] = (kt + k0 + 8 + j < kmax) ? BF16_ONE : (bf16raw)0;
     } else if (gr < mem_rows) {
       const char* base = (const char*)src + (int64_t)gr * srow * (SRC_F32 ? 4 : 2);
       int krem = kmax - kt - k0;  // how many of our 16 k are in range
@@ -74,27 +88,31 @@ __device__ __forceinline__ void stage_tile(const void* __restrict__ src, bf16raw
 #pragma unroll
           for (int v = 0; v < 4; ++v) {
             floatx4 x = *(const floatx4*)(s + v * 4);
+            bf16raw* dv = v < 2 ? d0 + v * 4 : d1 + (v - 2) * 4;
 #pragma unroll
-            for (int j = 0; j < 4; ++j) dst[v * 4 + j] = f32_to_bf16(x[j]);
+            for (int j = 0; j < 4; ++j) dv[j] = f32_to_bf16(x[j]);
           }
         } else {
           const bf16raw* s = (const bf16raw*)base + kt + k0;
-          *(shortx8*)dst = *(const shortx8*)s;
-          *(shortx8*)(dst + 8) = *(const shortx8*)(s + 8);
+          *(shortx8*)d0 = *(const shortx8*)s;
+          *(shortx8*)d1 = *(const shortx8*)(s + 8);
         }
       } else {
 #pragma unroll
         for (int j = 0; j < 16; ++j) {
           int k = kt + k0 + j;
+          bf16raw* dj = (j < 8 ? d0 + j : d1 + (j - 8));
           if (k < kmax)
-            dst[j] = SRC_F32 ? f32_to_bf16(((const float*)base)[k]) : ((const bf16raw*)base)[k];
+            *dj = SRC_F32 ? f32_to_bf16(((const float*)base)[k]) : ((const bf16raw*)base)[k];
           else
-            dst[j] = 0;
+            *dj = 0;
         }
       }
     } else {
 #pragma unroll
-      for (int j = 0; j < 16; ++j) dst[j] = 0;
+      for (int j = 0; j < 8; ++j) d0[j] = 0;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) d1[j] = 0;
     }
   } else if (srow == 1) {
     // 16 consecutive rows at fixed k; 8 threads per k column
@@ -112,7 +130,7 @@ __device__ __forceinline__ void stage_tile(const void* __restrict__ src, bf16raw
           for (int v = 0; v < 4; ++v) {
             floatx4 x = *(const floatx4*)(s + v * 4);
 #pragma unroll
-            for (int j = 0; j < 4; ++j) lds[(r0 + v * 4 + j) * LDSTRIDE + k] = f32_to_bf16(x[j]);
+            for (int j = 0; j < 4; ++j) lds[(r0 + v * 4 + j) * LDSTRIDE + swz_col(r0 + v * 4 + j, k)] = f32_to_bf16(x[j]);
           }
         } else {
           const bf16raw* s = (const bf16raw*)base + row0 + r0;
@@ -120,7 +138,7 @@ __device__ __forceinline__ void stage_tile(const void* __restrict__ src, bf16raw
           for (int v = 0; v < 2; ++v) {
             shortx8 x = *(const shortx8*)(s + v * 8);
 #pragma unroll
-            for (int j = 0; j < 8; ++j) lds[(r0 + v * 8 + j) * LDSTRIDE + k] = (bf16raw)x[j];
+            for (int j = 0; j < 8; ++j) lds[(r0 + v * 8 + j) * LDSTRIDE + swz_col(r0 + v * 8 + j, k)] = (bf16raw)x[j];
           }
         }
       } else {
@@ -133,12 +151,12 @@ __device__ __forceinline__ void stage_tile(const void* __restrict__ src, bf16raw
           else if (j < mem_rem)  // row0 + r < mem_rows
             v = SRC_F32 ? f32_to_bf16(((const float*)base)[row0 + r])
                         : ((const bf16raw*)base)[row0 + r];
-          lds[r * LDSTRIDE + k] = v;
+          lds[r * LDSTRIDE + swz_col(r, k)] = v;
         }
       }
     } else {
 #pragma unroll
-      for (int j = 0; j < 16; ++j) lds[(r0 + j) * LDSTRIDE + k] = 0;
+      for (int j = 0; j < 16; ++j) lds[(r0 + j) * LDSTRIDE + swz_col(r0 + j, k)] = 0;
     }
   } else {
     // generic scalar path
@@ -155,7 +173,7 @@ __device__ __forceinline__ void stage_tile(const void* __restrict__ src, bf16raw
         const char* p = (const char*)src + ((int64_t)gr * srow + (int64_t)gk * skol) * (SRC_F32 ? 4 : 2);
         v = SRC_F32 ? f32_to_bf16(*(const float*)p) : *(const bf16raw*)p;
       }
-      lds[r * LDSTRIDE + k] = v;
+      lds[r * LDSTRIDE + swz_col(r, k)] = v;
     }
   }
 }
@@ -273,20 +291,19 @@ __device__ __forceinline__ void stage_write(bf16raw* __restrict__ lds, int64_t s
   if (skol == 1) {
     int r = t >> 1;
     int k0 = (t & 1) * 16;
-    bf16raw* dst = lds + r * LDSTRIDE + k0;
-    *(shortx8*)dst = *(const shortx8*)regs;
-    *(shortx8*)(dst + 8) = *(const shortx8*)(regs + 8);
+    *(shortx8*)(lds + r * LDSTRIDE + swz_col(r, k0)) = *(const shortx8*)regs;
+    *(shortx8*)(lds + r * LDSTRIDE + swz_col(r, k0 + 8)) = *(const shortx8*)(regs + 8);
   } else if (srow == 1) {
     int k = t >> 3;
     int r0 = (t & 7) * 16;
 #pragma unroll
-    for (int j = 0; j < 16; ++j) lds[(r0 + j) * LDSTRIDE + k] = regs[j];
+    for (int j = 0; j < 16; ++j) lds[(r0 + j) * LDSTRIDE + swz_col(r0 + j, k)] = regs[j];
   } else {
     int idx0 = t * 16;
 #pragma unroll
     for (int j = 0; j < 16; ++j) {
       int idx = idx0 + j;
-      lds[(idx / BK) * LDSTRIDE + (idx % BK)] = regs[j];
+      lds[(idx / BK) * LDSTRIDE + swz_col(idx / BK, idx % BK)] = regs[j];
     }
   }
 }
@@ -429,8 +446,10 @@ __global__ __launch_bounds__(256, 2) void gemm_kernel(const void* __restrict__ A
           a[mi] = *(const frag_t*)&As[buf][row * BKT + (((kq) ^ (row & (SLOTS - 1))) << 3)];
         }
 #pragma unroll
-        for (int ni = 0; ni < 4; ++ni)
-          b[ni] = *(const frag_t*)&Bs[buf][(wc * 64 + ni * 16 + l15) * LP + kg * 8 + sub * 32];
+        for (int ni = 0; ni < 4; ++ni) {
+          int rowb = wc * 64 + ni * 16 + l15;
+          b[ni] = *(const frag_t*)&Bs[buf][rowb * LP + sub * 32 + swz_col(rowb, kg * 8)];
+        }
 #pragma unroll
         for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
@@ -488,11 +507,13 @@ __global__ __launch_bounds__(256, 2) void gemm_kernel(const void* __restrict__ A
 #pragma unroll
         for (int mi = 0; mi < 4; ++mi) {
           int row = wr * 64 + mi * 16 + l15;
-          a[mi] = *(const frag_t*)&As[0][row * LP + kg * 8 + sub * 32];
+          a[mi] = *(const frag_t*)&As[0][row * LP + sub * 32 + swz_col(row, kg * 8)];
         }
 #pragma unroll
-        for (int ni = 0; ni < 4; ++ni)
-          b[ni] = *(const frag_t*)&Bs[0][(wc * 64 + ni * 16 + l15) * LP + kg * 8 + sub * 32];
+        for (int ni = 0; ni < 4; ++ni) {
+          int rowb = wc * 64 + ni * 16 + l15;
+          b[ni] = *(const frag_t*)&Bs[0][rowb * LP + sub * 32 + swz_col(rowb, kg * 8)];
+        }
 #pragma unroll
         for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
