@@ -27,9 +27,9 @@ def _choose_splitk(batch: int, n_out: int, k_in: int) -> int:
     batch reduction until ~1024 blocks (2 blocks/CU x 2 for tail overlap),
     keeping >=512 rows per slice."""
     base = -(-n_out // 128) * (-(-k_in) // 128)
-    want = max(1, 1024 // max(1, base))
+    want = max(1, 2048 // max(1, base))
     max_by_rows = max(1, batch // 512)
-    return int(min(want, max_by_rows, 256))
+    return int(min(want, max_by_rows, 512))
 
 
 class _LinearFn(torch.autograd.Function):

@@ -117,7 +117,7 @@ int main() {
     CK(hipMemset(col, 0x3c, (size_t)s.R * s.Kcol * 2));
     double fl = 2.0 * s.CO * s.Kcol * (double)s.R;
     printf("%s CO=%d Kcol=%d R=%ld |", s.name, s.CO, s.Kcol, (long)s.R);
-    for (int sk : {32, 64, 128, 256}) {
+    for (int sk : {128, 256, 512, 1024}) {
       CK(hipMemset(dw, 0, (size_t)s.CO * s.Kcol * 4));
       double t = run_wgrad(dz, col, dw, s.CO, s.Kcol, s.R, sk, 5);
       printf("  sk%-3d %7.1fus (%4.0f TF)", sk, t, fl / t / 1e6);
