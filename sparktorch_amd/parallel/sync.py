@@ -39,6 +39,7 @@ from sparktorch_amd.parallel.rendezvous import (
 )
 from sparktorch_amd.utils.data import handle_features
 from sparktorch_amd.utils.early_stopper import EarlyStopping
+from sparktorch_amd.utils.trace import StepMetrics, trace_range
 from sparktorch_amd.utils.serialize import (
     load_base_torch,
     load_torch_model,
@@ -270,6 +271,7 @@ def handle_model(
         # collectives must live on the comm device (RCCL wants GPU tensors)
         comm_dev = dev if dev.startswith("cuda") else "cpu"
         should_stop_t = torch.zeros(1, device=comm_dev)
+        metrics = StepMetrics()
 
         for i in range(iters):
             if 0 < mini_batch < n:
@@ -279,7 +281,10 @@ def handle_model(
             else:
                 xb, yb = x_train, y_train
 
-            loss = trainer.train_step(xb, yb)
+            metrics.start()
+            with trace_range("train_step"):
+                loss = trainer.train_step(xb, yb)
+            metrics.stop(loss)
 
             if es is not None:
                 loss_for_es = (
@@ -295,6 +300,9 @@ def handle_model(
 
             if verbose:
                 print("rank %d iter %d loss %.6f" % (rank, i, loss), flush=True)
+
+        if verbose:
+            print("rank %d metrics %r" % (rank, metrics), flush=True)
 
         # ranks end in identical states (grads synced; same init) —
         # reference collects all and takes [0] (distributed.py:255-261)

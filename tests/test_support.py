@@ -129,3 +129,20 @@ def test_handle_data_autoencoder_mode():
     rows = [{"features": np.ones(3)}]
     mapped = list(handle_data("features", None)(iter(rows)))
     assert mapped[0].y_train is None
+
+
+def test_step_metrics_and_trace_range():
+    from sparktorch_amd.utils.trace import StepMetrics, trace_range
+
+    m = StepMetrics(window=4)
+    for i in range(6):
+        with m.step():
+            pass
+        m.last_losses.append(float(i))
+    s = m.summary()
+    assert s["iters"] == 6
+    assert s["avg_ms"] >= 0.0
+    assert len(m.last_ms) == 4  # windowed
+    with trace_range("cpu-noop"):  # no-op without a GPU
+        x = 1 + 1
+    assert x == 2
