@@ -263,8 +263,8 @@ hipError_t launch_maxpool_fwd(const bf16raw*, bf16raw*, uint8_t*, int, int, int,
                               hipStream_t);
 hipError_t launch_maxpool_bwd(const bf16raw*, const uint8_t*, bf16raw*, int, int, int, int, int,
                               int, hipStream_t);
-hipError_t launch_dropout(const bf16raw*, bf16raw*, int64_t, int64_t, float, uint32_t,
-                          hipStream_t);
+hipError_t launch_dropout(const bf16raw*, bf16raw*, int64_t, int64_t, int64_t, float,
+                          uint32_t, hipStream_t);
 }
 
 at::Tensor im2col(at::Tensor x, int64_t KH, int64_t KW, int64_t sh, int64_t sw, int64_t ph,
@@ -323,11 +323,13 @@ at::Tensor maxpool_bwd(at::Tensor dy, at::Tensor arg, int64_t H, int64_t W, int6
   return dx;
 }
 
-at::Tensor dropout_apply(at::Tensor x, double p, int64_t seed, int64_t units_div) {
+at::Tensor dropout_apply(at::Tensor x, double p, int64_t seed, int64_t units_div,
+                         int64_t cmod) {
   check_gpu_contig(x, at::kBFloat16, "x");
   auto y = at::empty_like(x);
   CHECK_HIP(launch_dropout((const bf16raw*)x.data_ptr(), (bf16raw*)y.data_ptr(), x.numel(),
-                           units_div, (float)p, (uint32_t)(seed & 0xffffffff), cur_stream()));
+                           units_div, cmod, (float)p, (uint32_t)(seed & 0xffffffff),
+                           cur_stream()));
   return y;
 }
 

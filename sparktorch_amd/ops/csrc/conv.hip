@@ -180,23 +180,31 @@ __device__ __forceinline__ uint32_t hash_u32(uint32_t a, uint32_t seed) {
   return h;
 }
 
+// unit id: element-wise (units_div=1), NCHW channel-wise (units_div=HW:
+// unit = idx/HW = b*C+c), or NHWC channel-wise (units_div=HWC, cmod=C:
+// unit = (idx/HWC)*C + idx%C = b*C+c)
 __global__ void dropout_kernel(const bf16raw* __restrict__ in, bf16raw* __restrict__ out,
-                               int64_t n, int64_t units_div, uint32_t thresh, float scale,
-                               uint32_t seed) {
+                               int64_t n, int64_t units_div, int64_t cmod, uint32_t thresh,
+                               float scale, uint32_t seed) {
   for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; idx < n;
        idx += (int64_t)gridDim.x * blockDim.x) {
-    uint32_t unit = (uint32_t)(units_div > 1 ? idx / units_div : idx);
+    uint32_t unit;
+    if (cmod > 0)
+      unit = (uint32_t)((idx / units_div) * cmod + idx % cmod);
+    else
+      unit = (uint32_t)(units_div > 1 ? idx / units_div : idx);
     bool keep = hash_u32(unit, seed) >= thresh;
     out[idx] = keep ? f32_to_bf16(bf16_to_f32(in[idx]) * scale) : (bf16raw)0;
   }
 }
 
 extern "C" hipError_t launch_dropout(const bf16raw* in, bf16raw* out, int64_t n,
-                                     int64_t units_div, float p, uint32_t seed,
+                                     int64_t units_div, int64_t cmod, float p, uint32_t seed,
                                      hipStream_t stream) {
   float scale = 1.0f / (1.0f - p);
   uint32_t thresh = (uint32_t)(p * 4294967296.0);
-  dropout_kernel<<<cgrid(n, 256), 256, 0, stream>>>(in, out, n, units_div, thresh, scale, seed);
+  dropout_kernel<<<cgrid(n, 256), 256, 0, stream>>>(in, out, n, units_div, cmod, thresh, scale,
+                                                    seed);
   HIP_CHECK_LAUNCH();
   return hipSuccess;
 }

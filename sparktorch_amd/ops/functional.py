@@ -506,27 +506,40 @@ def hip_max_pool2d(x, kernel_size: int, stride: Optional[int] = None, padding: i
 
 class _DropoutFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, p, seed, units_div):
-        ctx.meta = (p, seed, units_div)
-        return ops.ext().dropout_apply(x, p, seed, units_div)
+    def forward(ctx, x, p, seed, units_div, cmod):
+        ctx.meta = (p, seed, units_div, cmod)
+        return ops.ext().dropout_apply(x, p, seed, units_div, cmod)
 
     @staticmethod
     def backward(ctx, dy):
-        p, seed, units_div = ctx.meta
+        p, seed, units_div, cmod = ctx.meta
         # same (seed, unit) hash -> same mask; scale applies to dy too
-        return ops.ext().dropout_apply(dy.contiguous().to(torch.bfloat16), p, seed, units_div), None, None, None
+        return (ops.ext().dropout_apply(dy.contiguous().to(torch.bfloat16), p, seed, units_div,
+                                        cmod), None, None, None, None)
 
 
-def hip_dropout(x, p: float, training: bool = True, channel_wise: bool = False):
+def hip_dropout(x, p: float, training: bool = True, channel_wise: bool = False,
+                layout: str = "nchw"):
     if not training or p <= 0.0:
         return x
     if x.is_cuda:
         if x.dtype != torch.bfloat16:
             x = x.to(torch.bfloat16)
-        units_div = (x.shape[2] * x.shape[3]) if (channel_wise and x.dim() == 4) else 1
+        cmod = 0
+        units_div = 1
+        if channel_wise and x.dim() == 4:
+            if layout == "nhwc":  # unit (b, c) from [B,H,W,C]: idx/HWC*C + idx%C
+                units_div = x.shape[1] * x.shape[2] * x.shape[3]
+                cmod = x.shape[3]
+            else:
+                units_div = x.shape[2] * x.shape[3]
         seed = int(torch.randint(0, 2**31 - 1, (1,)).item())
-        return _DropoutFn.apply(x.contiguous(), p, seed, units_div)
-    return F.dropout2d(x, p, training) if channel_wise else F.dropout(x, p, training)
+        return _DropoutFn.apply(x.contiguous(), p, seed, units_div, cmod)
+    if channel_wise:
+        if layout == "nhwc":
+            return F.dropout2d(x.permute(0, 3, 1, 2), p, training).permute(0, 2, 3, 1).contiguous()
+        return F.dropout2d(x, p, training)
+    return F.dropout(x, p, training)
 
 
 class _CEFn(torch.autograd.Function):

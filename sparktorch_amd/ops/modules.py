@@ -177,37 +177,45 @@ class HipGlobalAvgPool(nn.Module):
 
 
 class HipDropout(nn.Module):
-    def __init__(self, p: float = 0.5, channel_wise: bool = False):
+    def __init__(self, p: float = 0.5, channel_wise: bool = False, layout: str = "nchw"):
         super().__init__()
         self.p = p
         self.channel_wise = channel_wise
+        self.layout = layout
 
     def forward(self, x):
-        return hip_dropout(x, self.p, training=self.training, channel_wise=self.channel_wise)
+        return hip_dropout(x, self.p, training=self.training,
+                           channel_wise=self.channel_wise, layout=self.layout)
 
 
 class MnistCNNFused(nn.Module):
     """The reference example CNN (examples/cnn_network.py:6-24) on the fully
-    native path: implicit-GEMM convs with fused ReLU, native maxpool,
-    counter-based Dropout2d.  state_dict-compatible with models.mnist.MnistCNN."""
+    native path, channels-last end to end: NHWC implicit-GEMM convs with
+    fused ReLU, NHWC maxpool, counter-based NHWC Dropout2d.  The fc layer
+    expects the reference's NCHW flatten order, so its (tiny) weight is
+    viewed in NHWC order per forward instead of permuting the (huge)
+    activation.  state_dict-compatible with models.mnist.MnistCNN."""
 
     def __init__(self):
         super().__init__()
-        self.conv1 = HipConv2d(1, 16, kernel_size=5, activation="relu")
-        self.conv2 = HipConv2d(16, 32, kernel_size=3, activation="relu")
-        self.dropout = HipDropout(p=0.25, channel_wise=True)
+        self.conv1 = HipConv2d(1, 16, kernel_size=5, activation="relu", layout="nhwc")
+        self.conv2 = HipConv2d(16, 32, kernel_size=3, activation="relu", layout="nhwc")
+        self.dropout = HipDropout(p=0.25, channel_wise=True, layout="nhwc")
         self.fc = HipLinear(3872, 10)
 
     def forward(self, x):
         if x.is_cuda and x.dtype != torch.bfloat16:
             x = x.to(torch.bfloat16)
-        x = x.view(-1, 1, 28, 28)
+        x = x.view(-1, 1, 28, 28).permute(0, 2, 3, 1).contiguous()  # NHWC entry
         x = self.conv1(x)
         x = self.conv2(x)
-        x = hip_max_pool2d(x, 2)
+        x = hip_max_pool2d_nhwc(x, 2)
         x = self.dropout(x)
-        x = torch.flatten(x, 1)
-        return self.fc(x)
+        B, H, W, C = x.shape
+        x = x.reshape(B, H * W * C)
+        # fc weight reordered [10, C*H*W] -> [10, H*W*C] to match NHWC flatten
+        w = self.fc.weight.view(-1, C, H, W).permute(0, 2, 3, 1).reshape(-1, H * W * C)
+        return hip_linear(x, w.contiguous(), self.fc.bias)
 
 
 class FusedBasicBlock(nn.Module):
