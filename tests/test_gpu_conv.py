@@ -90,13 +90,13 @@ def test_dropout_statistics_and_backward_consistency():
     torch.manual_seed(3)
     x = torch.ones(100_000, device=DEV, dtype=torch.bfloat16)
     p = 0.25
-    out = _DropoutFn.apply(x, p, 12345, 1)
+    out = _DropoutFn.apply(x, p, 12345, 1, 0)
     kept = (out != 0).float().mean().item()
     assert abs(kept - 0.75) < 0.02
     # kept values scaled by 1/(1-p)
     assert abs(out.float().max().item() - 1.0 / 0.75) < 0.01
     # backward must reproduce the SAME mask
-    out2 = _DropoutFn.apply(x, p, 12345, 1)
+    out2 = _DropoutFn.apply(x, p, 12345, 1, 0)
     assert torch.equal(out, out2)
 
 
@@ -104,9 +104,19 @@ def test_dropout2d_channelwise():
     from sparktorch_amd.ops.functional import _DropoutFn
 
     x = torch.ones(8, 64, 11, 11, device=DEV, dtype=torch.bfloat16)
-    out = _DropoutFn.apply(x, 0.5, 99, 11 * 11)
+    out = _DropoutFn.apply(x, 0.5, 99, 11 * 11, 0)
     per_channel = out.float().sum(dim=(2, 3)).flatten()
     # each channel entirely zero or entirely kept
+    assert ((per_channel == 0) | (per_channel > 100)).all()
+
+
+def test_dropout2d_channelwise_nhwc():
+    from sparktorch_amd.ops.functional import _DropoutFn
+
+    x = torch.ones(8, 11, 11, 64, device=DEV, dtype=torch.bfloat16)
+    out = _DropoutFn.apply(x, 0.5, 99, 11 * 11 * 64, 64)
+    per_channel = out.float().sum(dim=(1, 2)).flatten()  # NHWC: sum over H,W
+    # each (b, c) unit entirely zero or entirely kept
     assert ((per_channel == 0) | (per_channel > 100)).all()
 
 
