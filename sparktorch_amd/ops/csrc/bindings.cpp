@@ -17,7 +17,8 @@ hipError_t launch_fused_adam(float*, const float*, float*, float*, int64_t, floa
 hipError_t launch_fused_sgd(float*, const float*, float*, int64_t, float, float, float, float,
                             float, int, int, int, hipStream_t);
 hipError_t launch_relu_bwd(const bf16raw*, const bf16raw*, bf16raw*, int64_t, hipStream_t);
-hipError_t launch_bias_grad(const bf16raw*, float*, int, int, hipStream_t);
+hipError_t launch_bias_grad(const bf16raw*, float*, int, int, float*, int,
+                            hipStream_t);
 hipError_t launch_cast_f64_f32(const double*, float*, int64_t, hipStream_t);
 hipError_t launch_cast_f32_bf16(const float*, bf16raw*, int64_t, hipStream_t);
 hipError_t launch_ce_fused(const bf16raw*, const int64_t*, float*, bf16raw*, int, int,
@@ -111,12 +112,32 @@ at::Tensor relu_bwd(at::Tensor dy, at::Tensor y) {
   return dz;
 }
 
+
+// small-N column-sum path: vectorized 8-col groups, partials in scratch
+static bool bias_small_path(int64_t M, int N) {
+  return N >= 8 && N <= 2048 && (N & 7) == 0 && (256 % (N >> 3)) == 0 && M >= 4096;
+}
+static int bias_slices(int64_t M, int N) {
+  int64_t s = (M * N * 2) / (256 * 1024);
+  return (int)(s < 64 ? 64 : (s > 512 ? 512 : s));
+}
+
 at::Tensor bias_grad(at::Tensor dz) {
   check_gpu_contig(dz, at::kBFloat16, "dz");
   TORCH_CHECK(dz.dim() == 2);
   auto db = at::zeros({dz.size(1)}, dz.options().dtype(at::kFloat));
-  CHECK_HIP(launch_bias_grad((const bf16raw*)dz.data_ptr(), db.data_ptr<float>(),
-                             (int)dz.size(0), (int)dz.size(1), cur_stream()));
+  int64_t M = dz.size(0);
+  int N = (int)dz.size(1);
+  at::Tensor scratch;
+  float* sp = nullptr;
+  int S = 0;
+  if (bias_small_path(M, N)) {
+    S = bias_slices(M, N);
+    scratch = at::empty({(int64_t)S * N}, dz.options().dtype(at::kFloat));
+    sp = scratch.data_ptr<float>();
+  }
+  CHECK_HIP(launch_bias_grad((const bf16raw*)dz.data_ptr(), db.data_ptr<float>(), (int)M, N, sp,
+                             S, cur_stream()));
   return db;
 }
 
@@ -246,8 +267,18 @@ void bias_grad_into(at::Tensor dz, at::Tensor db) {
   check_gpu_contig(dz, at::kBFloat16, "dz");
   check_gpu_contig(db, at::kFloat, "db");
   TORCH_CHECK(dz.dim() == 2 && db.numel() == dz.size(1));
-  CHECK_HIP(launch_bias_grad((const bf16raw*)dz.data_ptr(), db.data_ptr<float>(),
-                             (int)dz.size(0), (int)dz.size(1), cur_stream()));
+  int64_t M = dz.size(0);
+  int N = (int)dz.size(1);
+  at::Tensor scratch;
+  float* sp = nullptr;
+  int S = 0;
+  if (bias_small_path(M, N)) {
+    S = bias_slices(M, N);
+    scratch = at::empty({(int64_t)S * N}, dz.options().dtype(at::kFloat));
+    sp = scratch.data_ptr<float>();
+  }
+  CHECK_HIP(launch_bias_grad((const bf16raw*)dz.data_ptr(), db.data_ptr<float>(), (int)M, N, sp,
+                             S, cur_stream()));
 }
 
 // ---------------------------------------------------------------------------

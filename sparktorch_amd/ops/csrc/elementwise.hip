@@ -131,38 +131,39 @@ extern "C" hipError_t launch_fused_sgd(float* p, const float* g, float* buf, int
 // ReLU backward mask: dz = dy * (y > 0); bf16 in/out, 8-wide vectorized
 // ---------------------------------------------------------------------------
 
-__global__ void relu_bwd_kernel(const bf16raw* __restrict__ dy, const bf16raw* __restrict__ y,
-                                bf16raw* __restrict__ dz, int64_t n) {
-  int64_t nvec = n >> 3;
-  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  const shortx8* dy8 = (const shortx8*)dy;
-  const shortx8* y8 = (const shortx8*)y;
-  shortx8* dz8 = (shortx8*)dz;
-  for (int64_t k = i; k < nvec; k += stride) {
-    shortx8 d = dy8[k];
-    shortx8 yv = y8[k];
-    shortx8 o;
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      // y > 0 in bf16 bits: sign bit clear and magnitude nonzero
-      uint16_t yb = (uint16_t)yv[j];
-      bool pos = (yb & 0x8000u) == 0 && yb != 0;
-      o[j] = pos ? d[j] : (short)0;
-    }
-    dz8[k] = o;
+__global__ void relu_bwd_kernel(const bf16raw* __restrict__ dy,
+                                const bf16raw* __restrict__ y, bf16raw* __restrict__ dz,
+                                int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    dz[i] = bf16_to_f32(y[i]) > 0.f ? dy[i] : (bf16raw)0;
   }
-  for (int64_t k = (nvec << 3) + i; k < n; k += stride) {
-    uint16_t yb = y[k];
-    bool pos = (yb & 0x8000u) == 0 && yb != 0;
-    dz[k] = pos ? dy[k] : (bf16raw)0;
+}
+
+__global__ void relu_bwd_vec_kernel(const bf16raw* __restrict__ dy,
+                                    const bf16raw* __restrict__ y, bf16raw* __restrict__ dz,
+                                    int64_t n8) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n8;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t off = i << 3;
+    const shortx8 g = *(const shortx8*)(dy + off);
+    const shortx8 v = *(const shortx8*)(y + off);
+    shortx8 out;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      out[j] = bf16_to_f32((bf16raw)v[j]) > 0.f ? g[j] : (short)0;
+    *(shortx8*)(dz + off) = out;
   }
 }
 
 extern "C" hipError_t launch_relu_bwd(const bf16raw* dy, const bf16raw* y, bf16raw* dz, int64_t n,
                                       hipStream_t stream) {
   int block = 256;
-  relu_bwd_kernel<<<ew_grid(n / 8 + 1, block), block, 0, stream>>>(dy, y, dz, n);
+  if ((n & 7) == 0 && n > 0) {
+    relu_bwd_vec_kernel<<<ew_grid(n / 8, block), block, 0, stream>>>(dy, y, dz, n >> 3);
+  } else {
+    relu_bwd_kernel<<<ew_grid(n / 8 + 1, block), block, 0, stream>>>(dy, y, dz, n);
+  }
   HIP_CHECK_LAUNCH();
   return hipSuccess;
 }
@@ -214,8 +215,63 @@ __global__ void bias_grad_kernel(const bf16raw* __restrict__ dz, float* __restri
   }
 }
 
+// Small-N variant (conv bias over huge M): thread owns an 8-column group,
+// 256/(N/8) row lanes per block; block partials go to scratch by plain
+// stores (no atomic chains), then a tiny tree finalize ACCUMULATES into db.
+__global__ void bias_grad_small_kernel(const bf16raw* __restrict__ dz,
+                                       float* __restrict__ partial, int64_t M, int N) {
+  int groups = N >> 3;
+  int rpg = 256 / groups;
+  int cg = threadIdx.x % groups;
+  int rg = threadIdx.x / groups;
+  int c0 = cg << 3;
+  int64_t per = ceil_div_i64(M, gridDim.x);
+  int64_t lo = (int64_t)blockIdx.x * per;
+  int64_t hi = lo + per < M ? lo + per : M;
+  float s[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  for (int64_t r = lo + rg; r < hi; r += rpg) {
+    const shortx8 v = *(const shortx8*)(dz + r * N + c0);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) s[j] += bf16_to_f32((bf16raw)v[j]);
+  }
+  __shared__ float l0[256][8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) l0[threadIdx.x][j] = s[j];
+  __syncthreads();
+  if (rg == 0) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float a = 0.f;
+      for (int q = 0; q < rpg; ++q) a += l0[q * groups + cg][j];
+      partial[(int64_t)blockIdx.x * N + c0 + j] = a;
+    }
+  }
+}
+
+__global__ void bias_grad_small_finalize_kernel(const float* __restrict__ partial, int S,
+                                                float* __restrict__ db, int N) {
+  int c = blockIdx.x;
+  float a = 0.f;
+  for (int sidx = threadIdx.x; sidx < S; sidx += blockDim.x) a += partial[(int64_t)sidx * N + c];
+  __shared__ float l[256];
+  l[threadIdx.x] = a;
+  __syncthreads();
+  for (int off = blockDim.x >> 1; off > 0; off >>= 1) {
+    if (threadIdx.x < off) l[threadIdx.x] += l[threadIdx.x + off];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) db[c] += l[0];
+}
+
 extern "C" hipError_t launch_bias_grad(const bf16raw* dz, float* db, int M, int N,
-                                       hipStream_t stream) {
+                                       float* scratch, int S, hipStream_t stream) {
+  if (scratch != nullptr) {
+    bias_grad_small_kernel<<<S, 256, 0, stream>>>(dz, scratch, M, N);
+    HIP_CHECK_LAUNCH();
+    bias_grad_small_finalize_kernel<<<N, 256, 0, stream>>>(scratch, S, db, N);
+    HIP_CHECK_LAUNCH();
+    return hipSuccess;
+  }
   dim3 block(64, 4);
   // ~512 row chunks: 2048 waves keeps 256 CUs latency-hidden; per-column
   // atomic chains stay ~512 deep (few us, parallel across columns)

@@ -339,22 +339,33 @@ __device__ __forceinline__ void stage_glds(const bf16raw* __restrict__ src,
   }
 }
 
-// boundary-slab fallback writing the SAME swizzled image with zero fill
+// boundary-slab fallback writing the SAME swizzled image with zero fill.
+// Vectorized: AG implies a k-contiguous source, so full granules move as
+// shortx8 (the K<BKT degenerate shapes — e.g. a K=32 dgrad — run this every
+// slab, and the original per-element version was the whole kernel's cost).
 template <int ROWS, int BKT>
 __device__ __forceinline__ void stage_swz_fallback(const bf16raw* __restrict__ src,
                                                    bf16raw* __restrict__ lds, int row0, int rmax,
                                                    int kt, int kmax, int64_t sam) {
   constexpr int SLOTS = BKT / 8;
-  constexpr int E = ROWS * BKT / 256;
-  int base = threadIdx.x * E;
+  constexpr int TPR = 256 / ROWS;    // threads per row (1 or 2)
+  constexpr int KSPAN = BKT / TPR;   // k elements per thread
+  const int t = threadIdx.x;
+  const int r = t / TPR;
+  const int kb = (t % TPR) * KSPAN;
+  const int gr = row0 + r;
+  const bf16raw* base = src + (int64_t)gr * sam + kt;
 #pragma unroll
-  for (int j = 0; j < E; ++j) {
-    int idx = base + j;
-    int r = idx / BKT, k = idx % BKT;
-    bf16raw v = 0;
-    int gr = row0 + r, gk = kt + k;
-    if (gr < rmax && gk < kmax) v = src[(int64_t)gr * sam + gk];
-    lds[r * BKT + (((k >> 3) ^ (r & (SLOTS - 1))) << 3) + (k & 7)] = v;
+  for (int g = 0; g < KSPAN / 8; ++g) {
+    int k = kb + g * 8;
+    bf16raw* dst = lds + r * BKT + (((k >> 3) ^ (r & (SLOTS - 1))) << 3);
+    if (gr < rmax && kt + k + 8 <= kmax) {
+      *(shortx8*)dst = *(const shortx8*)(base + k);
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        dst[j] = (gr < rmax && kt + k + j < kmax) ? base[k + j] : (bf16raw)0;
+    }
   }
 }
 
