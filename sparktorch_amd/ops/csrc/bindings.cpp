@@ -19,6 +19,8 @@ hipError_t launch_fused_sgd(float*, const float*, float*, int64_t, float, float,
 hipError_t launch_relu_bwd(const bf16raw*, const bf16raw*, bf16raw*, int64_t, hipStream_t);
 hipError_t launch_bias_grad(const bf16raw*, float*, int, int, float*, int,
                             hipStream_t);
+hipError_t launch_small_wgrad(const bf16raw*, const bf16raw*, float*, float*, int, int64_t, int,
+                              int, hipStream_t);
 hipError_t launch_cast_f64_f32(const double*, float*, int64_t, hipStream_t);
 hipError_t launch_cast_f32_bf16(const float*, bf16raw*, int64_t, hipStream_t);
 hipError_t launch_ce_fused(const bf16raw*, const int64_t*, float*, bf16raw*, int, int,
@@ -218,11 +220,30 @@ at::Tensor linear_dgrad(at::Tensor dz, at::Tensor w) {
   return dx;
 }
 
+
+// tiny-output wgrad: batched outer product instead of a ~98%-dead MFMA tile
+static bool small_wgrad_path(int64_t M, int64_t N, int64_t K) {
+  return N * K <= 1024 && N <= 64 && K <= 256 && M >= 65536;
+}
+static int small_wgrad_slices(int64_t M) {
+  int64_t s = M / 4096;
+  return (int)(s < 256 ? 256 : (s > 2048 ? 2048 : s));
+}
+
 at::Tensor linear_wgrad(at::Tensor dz, at::Tensor x, int64_t splitk) {
   check_gpu_contig(dz, at::kBFloat16, "dz");
   check_gpu_contig(x, at::kBFloat16, "x");
   int64_t B = dz.size(0), N = dz.size(1), K = x.size(1);
   TORCH_CHECK(x.size(0) == B, "shape mismatch");
+  if (small_wgrad_path(B, N, K)) {
+    auto dw = at::zeros({N, K}, x.options().dtype(at::kFloat));
+    int S = small_wgrad_slices(B);
+    auto scratch = at::empty({(int64_t)S * N * K}, x.options().dtype(at::kFloat));
+    CHECK_HIP(launch_small_wgrad((const bf16raw*)dz.data_ptr(), (const bf16raw*)x.data_ptr(),
+                                 dw.data_ptr<float>(), scratch.data_ptr<float>(), S, B, (int)N,
+                                 (int)K, cur_stream()));
+    return dw;
+  }
   at::Tensor dw = splitk > 1 ? at::zeros({N, K}, x.options().dtype(at::kFloat))
                              : at::empty({N, K}, x.options().dtype(at::kFloat));
   CHECK_HIP(launch_gemm_bf16(dz.data_ptr(), x.data_ptr(), /*b_is_f32*/ 0, dw.data_ptr<float>(),
@@ -241,6 +262,14 @@ void linear_wgrad_into(at::Tensor dz, at::Tensor x, at::Tensor dw, int64_t split
   check_gpu_contig(dw, at::kFloat, "dw");
   int64_t B = dz.size(0), N = dz.size(1), K = x.size(1);
   TORCH_CHECK(x.size(0) == B && dw.size(0) == N && dw.size(1) == K, "shape mismatch");
+  if (small_wgrad_path(B, N, K)) {
+    int S = small_wgrad_slices(B);
+    auto scratch = at::empty({(int64_t)S * N * K}, x.options().dtype(at::kFloat));
+    CHECK_HIP(launch_small_wgrad((const bf16raw*)dz.data_ptr(), (const bf16raw*)x.data_ptr(),
+                                 dw.data_ptr<float>(), scratch.data_ptr<float>(), S, B, (int)N,
+                                 (int)K, cur_stream()));
+    return;
+  }
   if (splitk < 1) splitk = 1;
   CHECK_HIP(launch_gemm_bf16(dz.data_ptr(), x.data_ptr(), 0, dw.data_ptr<float>(), nullptr,
                              nullptr, (int)N, (int)K, (int)B, 1, N, K, 1, 0, -(int)splitk,

@@ -284,6 +284,81 @@ extern "C" hipError_t launch_bias_grad(const bf16raw* dz, float* db, int M, int 
 }
 
 // ---------------------------------------------------------------------------
+// Tiny-output weight gradient: dW[CO,K] = dz^T @ col with CO*K <= 1024 and a
+// huge row reduction (e.g. a 16x25 first-layer conv over 37M image rows).
+// A 128-row MFMA tile is ~98% dead at M=16, so this is a batched outer
+// product instead: stage a row tile of both operands in LDS (contiguous
+// copies), every thread accumulates its own output cells over the tile
+// (dz reads broadcast within a wave), partials to scratch, tree finalize.
+// ---------------------------------------------------------------------------
+
+#define SWG_ROWS 64
+
+__global__ void small_wgrad_kernel(const bf16raw* __restrict__ dz, const bf16raw* __restrict__ col,
+                                   float* __restrict__ partial, int64_t M, int CO, int K) {
+  const int OUT = CO * K;
+  const int per_thread = (OUT + 255) / 256;
+  float acc[4] = {0, 0, 0, 0};  // per_thread <= 4 (OUT <= 1024)
+  int o0 = threadIdx.x * per_thread;
+
+  __shared__ bf16raw zs[SWG_ROWS * 64];   // CO <= 64
+  __shared__ bf16raw cs[SWG_ROWS * 256];  // K <= 256
+
+  int64_t per = ceil_div_i64(ceil_div_i64(M, gridDim.x), SWG_ROWS) * SWG_ROWS;
+  int64_t lo = (int64_t)blockIdx.x * per;
+  int64_t hi = lo + per < M ? lo + per : M;
+
+  for (int64_t r0 = lo; r0 < hi; r0 += SWG_ROWS) {
+    int nrows = (int)((hi - r0) < SWG_ROWS ? (hi - r0) : SWG_ROWS);
+    __syncthreads();
+    // stage dz rows [r0, r0+nrows) x CO and col rows x K (both contiguous)
+    for (int i = threadIdx.x; i < nrows * CO; i += 256) zs[i] = dz[r0 * CO + i];
+    for (int i = threadIdx.x; i < nrows * K; i += 256) cs[i] = col[r0 * K + i];
+    __syncthreads();
+    for (int r = 0; r < nrows; ++r) {
+      const bf16raw* zrow = zs + r * CO;
+      const bf16raw* crow = cs + r * K;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        int o = o0 + j;
+        if (j < per_thread && o < OUT)
+          acc[j] += bf16_to_f32(zrow[o / K]) * bf16_to_f32(crow[o % K]);
+      }
+    }
+  }
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    int o = o0 + j;
+    if (j < per_thread && o < OUT) partial[(int64_t)blockIdx.x * OUT + o] = acc[j];
+  }
+}
+
+__global__ void small_wgrad_finalize_kernel(const float* __restrict__ partial, int S,
+                                            float* __restrict__ dw, int OUT) {
+  int o = blockIdx.x;
+  float a = 0.f;
+  for (int s = threadIdx.x; s < S; s += blockDim.x) a += partial[(int64_t)s * OUT + o];
+  __shared__ float l[256];
+  l[threadIdx.x] = a;
+  __syncthreads();
+  for (int off = blockDim.x >> 1; off > 0; off >>= 1) {
+    if (threadIdx.x < off) l[threadIdx.x] += l[threadIdx.x + off];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) dw[o] += l[0];
+}
+
+extern "C" hipError_t launch_small_wgrad(const bf16raw* dz, const bf16raw* col, float* dw,
+                                         float* scratch, int S, int64_t M, int CO, int K,
+                                         hipStream_t stream) {
+  small_wgrad_kernel<<<S, 256, 0, stream>>>(dz, col, scratch, M, CO, K);
+  HIP_CHECK_LAUNCH();
+  small_wgrad_finalize_kernel<<<CO * K, 256, 0, stream>>>(scratch, S, dw, CO * K);
+  HIP_CHECK_LAUNCH();
+  return hipSuccess;
+}
+
+// ---------------------------------------------------------------------------
 // fp64 -> fp32 cast (Spark DenseVector pack)
 // ---------------------------------------------------------------------------
 
