@@ -300,6 +300,17 @@ __global__ void small_wgrad_kernel(const bf16raw* __restrict__ dz, const bf16raw
   const int per_thread = (OUT + 255) / 256;
   float acc[4] = {0, 0, 0, 0};  // per_thread <= 4 (OUT <= 1024)
   int o0 = threadIdx.x * per_thread;
+  // hoist the (co, k) decode of each owned cell out of the row loop (the
+  // integer divides dominated the kernel when recomputed per row)
+  int zi[4], ci[4];
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    int o = o0 + j;
+    bool ok = (j < per_thread && o < OUT);
+    zi[j] = ok ? o / K : 0;
+    ci[j] = ok ? o % K : 0;
+    if (!ok) zi[j] = -1;
+  }
 
   __shared__ bf16raw zs[SWG_ROWS * 64];   // CO <= 64
   __shared__ bf16raw cs[SWG_ROWS * 256];  // K <= 256
@@ -319,11 +330,8 @@ __global__ void small_wgrad_kernel(const bf16raw* __restrict__ dz, const bf16raw
       const bf16raw* zrow = zs + r * CO;
       const bf16raw* crow = cs + r * K;
 #pragma unroll
-      for (int j = 0; j < 4; ++j) {
-        int o = o0 + j;
-        if (j < per_thread && o < OUT)
-          acc[j] += bf16_to_f32(zrow[o / K]) * bf16_to_f32(crow[o % K]);
-      }
+      for (int j = 0; j < 4; ++j)
+        if (zi[j] >= 0) acc[j] += bf16_to_f32(zrow[zi[j]]) * bf16_to_f32(crow[ci[j]]);
     }
   }
 #pragma unroll
