@@ -313,7 +313,8 @@ __global__ void small_wgrad_kernel(const bf16raw* __restrict__ dz, const bf16raw
   }
 
   __shared__ bf16raw zs[SWG_ROWS * 64];   // CO <= 64
-  __shared__ bf16raw cs[SWG_ROWS * 256];  // K <= 256
+  __shared__ bf16raw cs[SWG_ROWS * 256];  // K (padded to x4) <= 256
+  const int Kp = (K + 3) & ~3;  // 8B-aligned LDS rows for the shortx4 path
 
   int64_t per = ceil_div_i64(ceil_div_i64(M, gridDim.x), SWG_ROWS) * SWG_ROWS;
   int64_t lo = (int64_t)blockIdx.x * per;
@@ -324,14 +325,30 @@ __global__ void small_wgrad_kernel(const bf16raw* __restrict__ dz, const bf16raw
     __syncthreads();
     // stage dz rows [r0, r0+nrows) x CO and col rows x K (both contiguous)
     for (int i = threadIdx.x; i < nrows * CO; i += 256) zs[i] = dz[r0 * CO + i];
-    for (int i = threadIdx.x; i < nrows * K; i += 256) cs[i] = col[r0 * K + i];
+    for (int i = threadIdx.x; i < nrows * K; i += 256)
+      cs[(i / K) * Kp + (i % K)] = col[r0 * K + i];
     __syncthreads();
-    for (int r = 0; r < nrows; ++r) {
-      const bf16raw* zrow = zs + r * CO;
-      const bf16raw* crow = cs + r * K;
+    // fast path: the thread's 4 cells share one dz element and 4
+    // consecutive col elements (true for almost every thread when K >= 4):
+    // 1 scalar + 1 vector LDS read per row instead of 8 scalar reads
+    if (per_thread == 4 && zi[0] >= 0 && zi[3] == zi[0] && ci[3] == ci[0] + 3 &&
+        (ci[0] & 3) == 0) {
+      const bf16raw* zp = zs + zi[0];
+      const bf16raw* cp = cs + ci[0];
+      for (int r = 0; r < nrows; ++r) {
+        float z = bf16_to_f32(zp[r * CO]);
+        const shortx4 c4 = *(const shortx4*)(cp + r * Kp);
 #pragma unroll
-      for (int j = 0; j < 4; ++j)
-        if (zi[j] >= 0) acc[j] += bf16_to_f32(zrow[zi[j]]) * bf16_to_f32(crow[ci[j]]);
+        for (int j = 0; j < 4; ++j) acc[j] += z * bf16_to_f32((bf16raw)c4[j]);
+      }
+    } else {
+      for (int r = 0; r < nrows; ++r) {
+        const bf16raw* zrow = zs + r * CO;
+        const bf16raw* crow = cs + r * Kp;
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          if (zi[j] >= 0) acc[j] += bf16_to_f32(zrow[zi[j]]) * bf16_to_f32(crow[ci[j]]);
+      }
     }
   }
 #pragma unroll
