@@ -48,9 +48,9 @@ hipError_t launch_maxpool_gen_bwd(const bf16raw*, const uint8_t*, bf16raw*, int6
 hipError_t launch_gap_fwd(const bf16raw*, bf16raw*, int64_t, int64_t, hipStream_t);
 hipError_t launch_gap_bwd(const bf16raw*, bf16raw*, int64_t, int64_t, hipStream_t);
 hipError_t launch_im2col_nhwc(const bf16raw*, bf16raw*, int, int, int, int, int, int, int, int,
-                              int, int, int, int, hipStream_t);
+                              int, int, int, int, int, hipStream_t);
 hipError_t launch_col2im_nhwc(const bf16raw*, bf16raw*, int, int, int, int, int, int, int, int,
-                              int, int, int, int, hipStream_t);
+                              int, int, int, int, int, hipStream_t);
 hipError_t launch_maxpool_nhwc_fwd(const bf16raw*, bf16raw*, uint8_t*, int, int, int, int, int,
                                    int, int, int, int, hipStream_t);
 hipError_t launch_maxpool_nhwc_bwd(const bf16raw*, const uint8_t*, bf16raw*, int, int, int, int,
@@ -223,7 +223,8 @@ at::Tensor linear_dgrad(at::Tensor dz, at::Tensor w) {
 
 // tiny-output wgrad: batched outer product instead of a ~98%-dead MFMA tile
 static bool small_wgrad_path(int64_t M, int64_t N, int64_t K) {
-  return N * K <= 1024 && N <= 64 && K <= 256 && M >= 65536;
+  // slot kernel: one lane per (co, 8k) cell group; needs aligned K
+  return (K & 7) == 0 && N * K <= 2048 && N <= 64 && K <= 512 && M >= 65536;
 }
 static int small_wgrad_slices(int64_t M) {
   int64_t s = M / 4096;
@@ -572,10 +573,15 @@ at::Tensor im2col_nhwc(at::Tensor x, int64_t KH, int64_t KW, int64_t sh, int64_t
   int B = (int)x.size(0), H = (int)x.size(1), W = (int)x.size(2), CI = (int)x.size(3);
   int HO = (int)((H + 2 * ph - KH) / sh + 1);
   int WO = (int)((W + 2 * pw - KW) / sw + 1);
-  auto col = at::empty({(int64_t)B * HO * WO, KH * KW * CI}, x.options());
+  // K padded to a multiple of 8: keeps every consumer 16B-aligned (glds GEMM
+  // staging, vector wgrad) at the cost of <= 7 zero columns
+  int64_t K = KH * KW * CI;
+  int64_t Kp = (K + 7) & ~7LL;
+  auto col = Kp == K ? at::empty({(int64_t)B * HO * WO, Kp}, x.options())
+                     : at::zeros({(int64_t)B * HO * WO, Kp}, x.options());
   CHECK_HIP(launch_im2col_nhwc((const bf16raw*)x.data_ptr(), (bf16raw*)col.data_ptr(), B, CI, H,
                                W, (int)KH, (int)KW, HO, WO, (int)sh, (int)sw, (int)ph, (int)pw,
-                               cur_stream()));
+                               (int)Kp, cur_stream()));
   return col;
 }
 
@@ -584,10 +590,12 @@ at::Tensor col2im_nhwc(at::Tensor dcol, int64_t B, int64_t CI, int64_t H, int64_
   check_gpu_contig(dcol, at::kBFloat16, "dcol");
   int HO = (int)((H + 2 * ph - KH) / sh + 1);
   int WO = (int)((W + 2 * pw - KW) / sw + 1);
+  int Kp = (int)dcol.size(1);  // row stride (K padded to x8)
+  TORCH_CHECK(Kp >= KH * KW * CI, "dcol narrower than K");
   auto dx = at::empty({B, H, W, CI}, dcol.options());
   CHECK_HIP(launch_col2im_nhwc((const bf16raw*)dcol.data_ptr(), (bf16raw*)dx.data_ptr(), (int)B,
                                (int)CI, (int)H, (int)W, (int)KH, (int)KW, HO, WO, (int)sh,
-                               (int)sw, (int)ph, (int)pw, cur_stream()));
+                               (int)sw, (int)ph, (int)pw, Kp, cur_stream()));
   return dx;
 }
 

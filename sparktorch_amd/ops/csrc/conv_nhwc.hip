@@ -25,9 +25,9 @@ static inline int cgrid8(int64_t n, int block) {
 
 __global__ void im2col_nhwc_vec_kernel(const bf16raw* __restrict__ x, bf16raw* __restrict__ col,
                                        int B, int CI, int H, int W, int KH, int KW, int HO,
-                                       int WO, int sh, int sw, int ph, int pw) {
+                                       int WO, int sh, int sw, int ph, int pw, int Kp) {
   int civ8 = CI >> 3;
-  int64_t K = (int64_t)CI * KH * KW;
+  int64_t K = Kp;  // row stride (logical K padded to a multiple of 8)
   int64_t total = (int64_t)B * HO * WO * KH * KW * civ8;
   for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
        idx += (int64_t)gridDim.x * blockDim.x) {
@@ -60,8 +60,8 @@ __global__ void im2col_nhwc_vec_kernel(const bf16raw* __restrict__ x, bf16raw* _
 // KW*CI stretch of reads and writes is contiguous.
 __global__ void im2col_nhwc_kernel(const bf16raw* __restrict__ x, bf16raw* __restrict__ col,
                                    int B, int CI, int H, int W, int KH, int KW, int HO, int WO,
-                                   int sh, int sw, int ph, int pw) {
-  int64_t K = (int64_t)CI * KH * KW;
+                                   int sh, int sw, int ph, int pw, int Kp) {
+  int64_t K = (int64_t)CI * KH * KW;  // iteration space (row stride is Kp)
   int64_t total = (int64_t)B * HO * WO * K;
   for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
        idx += (int64_t)gridDim.x * blockDim.x) {
@@ -79,7 +79,7 @@ __global__ void im2col_nhwc_kernel(const bf16raw* __restrict__ x, bf16raw* __res
     bf16raw v = 0;
     if (h >= 0 && h < H && w >= 0 && w < W)
       v = x[(((int64_t)b * H + h) * W + w) * CI + ci];
-    col[idx] = v;
+    col[m * Kp + k] = v;
   }
 }
 
@@ -89,8 +89,8 @@ __global__ void im2col_nhwc_kernel(const bf16raw* __restrict__ x, bf16raw* __res
 __global__ void im2col_nhwc_rowcopy_kernel(const bf16raw* __restrict__ x,
                                            bf16raw* __restrict__ col, int B, int CI, int H,
                                            int W, int KH, int KW, int HO, int WO, int sh,
-                                           int sw, int ph, int pw) {
-  int64_t K = (int64_t)CI * KH * KW;
+                                           int sw, int ph, int pw, int Kp) {
+  int64_t K = Kp;
   int64_t total = (int64_t)B * HO * WO * KH;
   for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
        idx += (int64_t)gridDim.x * blockDim.x) {
@@ -123,16 +123,16 @@ __global__ void im2col_nhwc_rowcopy_kernel(const bf16raw* __restrict__ x,
 
 extern "C" hipError_t launch_im2col_nhwc(const bf16raw* x, bf16raw* col, int B, int CI, int H,
                                          int W, int KH, int KW, int HO, int WO, int sh, int sw,
-                                         int ph, int pw, hipStream_t stream) {
+                                         int ph, int pw, int Kp, hipStream_t stream) {
   if ((CI & 7) == 0) {
     int64_t total = (int64_t)B * HO * WO * KH * KW * (CI >> 3);
     im2col_nhwc_vec_kernel<<<cgrid8(total, 256), 256, 0, stream>>>(x, col, B, CI, H, W, KH, KW,
-                                                                   HO, WO, sh, sw, ph, pw);
+                                                                   HO, WO, sh, sw, ph, pw, Kp);
   } else {
     int64_t total = (int64_t)B * HO * WO * KH;
     im2col_nhwc_rowcopy_kernel<<<cgrid8(total, 256), 256, 0, stream>>>(x, col, B, CI, H, W, KH,
                                                                        KW, HO, WO, sh, sw, ph,
-                                                                       pw);
+                                                                       pw, Kp);
   }
   HIP_CHECK_LAUNCH();
   return hipSuccess;
@@ -147,9 +147,9 @@ extern "C" hipError_t launch_im2col_nhwc(const bf16raw* x, bf16raw* col, int B, 
 __global__ void col2im_nhwc_vec_kernel(const bf16raw* __restrict__ dcol,
                                        bf16raw* __restrict__ dx, int B, int CI, int H, int W,
                                        int KH, int KW, int HO, int WO, int sh, int sw, int ph,
-                                       int pw) {
+                                       int pw, int Kp) {
   int civ8 = CI >> 3;
-  int64_t K = (int64_t)CI * KH * KW;
+  int64_t K = Kp;
   int64_t total = (int64_t)B * H * W * civ8;
   for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
        idx += (int64_t)gridDim.x * blockDim.x) {
@@ -188,8 +188,8 @@ __global__ void col2im_nhwc_vec_kernel(const bf16raw* __restrict__ dcol,
 
 __global__ void col2im_nhwc_kernel(const bf16raw* __restrict__ dcol, bf16raw* __restrict__ dx,
                                    int B, int CI, int H, int W, int KH, int KW, int HO, int WO,
-                                   int sh, int sw, int ph, int pw) {
-  int64_t K = (int64_t)CI * KH * KW;
+                                   int sh, int sw, int ph, int pw, int Kp) {
+  int64_t K = Kp;
   int64_t total = (int64_t)B * H * W * CI;
   for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
        idx += (int64_t)gridDim.x * blockDim.x) {
@@ -220,15 +220,15 @@ __global__ void col2im_nhwc_kernel(const bf16raw* __restrict__ dcol, bf16raw* __
 
 extern "C" hipError_t launch_col2im_nhwc(const bf16raw* dcol, bf16raw* dx, int B, int CI, int H,
                                          int W, int KH, int KW, int HO, int WO, int sh, int sw,
-                                         int ph, int pw, hipStream_t stream) {
+                                         int ph, int pw, int Kp, hipStream_t stream) {
   if ((CI & 7) == 0) {
     int64_t total = (int64_t)B * H * W * (CI >> 3);
     col2im_nhwc_vec_kernel<<<cgrid8(total, 256), 256, 0, stream>>>(dcol, dx, B, CI, H, W, KH, KW,
-                                                                   HO, WO, sh, sw, ph, pw);
+                                                                   HO, WO, sh, sw, ph, pw, Kp);
   } else {
     int64_t total = (int64_t)B * H * W * CI;
     col2im_nhwc_kernel<<<cgrid8(total, 256), 256, 0, stream>>>(dcol, dx, B, CI, H, W, KH, KW, HO,
-                                                               WO, sh, sw, ph, pw);
+                                                               WO, sh, sw, ph, pw, Kp);
   }
   HIP_CHECK_LAUNCH();
   return hipSuccess;

@@ -292,69 +292,64 @@ extern "C" hipError_t launch_bias_grad(const bf16raw* dz, float* db, int M, int 
 // (dz reads broadcast within a wave), partials to scratch, tree finalize.
 // ---------------------------------------------------------------------------
 
-#define SWG_ROWS 64
-
+// One lane owns (co, 8-k-granule); K % 8 == 0 (the conv path pads its col
+// matrix), so every c-read is one aligned dwordx4 straight from global (L1
+// serves the row re-reads across co lanes) and the z-read broadcasts.  A
+// wave covers ceil(CO*K/8/64) cell-groups; remaining waves stripe rows.
 __global__ void small_wgrad_kernel(const bf16raw* __restrict__ dz, const bf16raw* __restrict__ col,
                                    float* __restrict__ partial, int64_t M, int CO, int K) {
-  const int OUT = CO * K;
-  const int per_thread = (OUT + 255) / 256;
-  float acc[4] = {0, 0, 0, 0};  // per_thread <= 4 (OUT <= 1024)
-  int o0 = threadIdx.x * per_thread;
-  // hoist the (co, k) decode of each owned cell out of the row loop (the
-  // integer divides dominated the kernel when recomputed per row)
-  int zi[4], ci[4];
-#pragma unroll
-  for (int j = 0; j < 4; ++j) {
-    int o = o0 + j;
-    bool ok = (j < per_thread && o < OUT);
-    zi[j] = ok ? o / K : 0;
-    ci[j] = ok ? o % K : 0;
-    if (!ok) zi[j] = -1;
-  }
+  const int KG8 = K >> 3;
+  const int SLOTS = CO * KG8;               // <= 256
+  const int wpc = (SLOTS + 63) >> 6;        // waves covering the cells (1/2/4)
+  const int rgs = 4 / wpc;                  // row stripes
+  const int wv = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int slot = (wv % wpc) * 64 + lane;
+  const int rg = wv / wpc;
+  const bool act = slot < SLOTS && rg < rgs;
+  const int co = act ? slot / KG8 : 0;
+  const int kg = act ? slot % KG8 : 0;
 
-  __shared__ bf16raw zs[SWG_ROWS * 64];   // CO <= 64
-  __shared__ bf16raw cs[SWG_ROWS * 256];  // K (padded to x4) <= 256
-  const int Kp = (K + 3) & ~3;  // 8B-aligned LDS rows for the shortx4 path
-
-  int64_t per = ceil_div_i64(ceil_div_i64(M, gridDim.x), SWG_ROWS) * SWG_ROWS;
+  int64_t per = ceil_div_i64(M, gridDim.x);
   int64_t lo = (int64_t)blockIdx.x * per;
   int64_t hi = lo + per < M ? lo + per : M;
 
-  for (int64_t r0 = lo; r0 < hi; r0 += SWG_ROWS) {
-    int nrows = (int)((hi - r0) < SWG_ROWS ? (hi - r0) : SWG_ROWS);
-    __syncthreads();
-    // stage dz rows [r0, r0+nrows) x CO and col rows x K (both contiguous)
-    for (int i = threadIdx.x; i < nrows * CO; i += 256) zs[i] = dz[r0 * CO + i];
-    for (int i = threadIdx.x; i < nrows * K; i += 256)
-      cs[(i / K) * Kp + (i % K)] = col[r0 * K + i];
-    __syncthreads();
-    // fast path: the thread's 4 cells share one dz element and 4
-    // consecutive col elements (true for almost every thread when K >= 4):
-    // 1 scalar + 1 vector LDS read per row instead of 8 scalar reads
-    if (per_thread == 4 && zi[0] >= 0 && zi[3] == zi[0] && ci[3] == ci[0] + 3 &&
-        (ci[0] & 3) == 0) {
-      const bf16raw* zp = zs + zi[0];
-      const bf16raw* cp = cs + ci[0];
-      for (int r = 0; r < nrows; ++r) {
-        float z = bf16_to_f32(zp[r * CO]);
-        const shortx4 c4 = *(const shortx4*)(cp + r * Kp);
+  float acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  if (act) {
+    const bf16raw* zp = dz + co;
+    const bf16raw* cp = col + kg * 8;
+    int64_t r = lo + rg;
+    for (; r + rgs * 2 <= hi; r += rgs * 2) {  // 2-deep unroll for load overlap
+      float z0 = bf16_to_f32(zp[r * CO]);
+      const shortx8 c0 = *(const shortx8*)(cp + r * K);
+      int64_t r1 = r + rgs;
+      float z1 = bf16_to_f32(zp[r1 * CO]);
+      const shortx8 c1 = *(const shortx8*)(cp + r1 * K);
 #pragma unroll
-        for (int j = 0; j < 4; ++j) acc[j] += z * bf16_to_f32((bf16raw)c4[j]);
-      }
-    } else {
-      for (int r = 0; r < nrows; ++r) {
-        const bf16raw* zrow = zs + r * CO;
-        const bf16raw* crow = cs + r * Kp;
+      for (int jj = 0; jj < 8; ++jj)
+        acc[jj] += z0 * bf16_to_f32((bf16raw)c0[jj]) + z1 * bf16_to_f32((bf16raw)c1[jj]);
+    }
+    for (; r < hi; r += rgs) {
+      float z = bf16_to_f32(zp[r * CO]);
+      const shortx8 c8 = *(const shortx8*)(cp + r * K);
 #pragma unroll
-        for (int j = 0; j < 4; ++j)
-          if (zi[j] >= 0) acc[j] += bf16_to_f32(zrow[zi[j]]) * bf16_to_f32(crow[ci[j]]);
-      }
+      for (int jj = 0; jj < 8; ++jj) acc[jj] += z * bf16_to_f32((bf16raw)c8[jj]);
     }
   }
+  // combine the rgs row-stripes of each cell through LDS, then store one
+  // partial row per block
+  __shared__ float red[256][8];
 #pragma unroll
-  for (int j = 0; j < 4; ++j) {
-    int o = o0 + j;
-    if (j < per_thread && o < OUT) partial[(int64_t)blockIdx.x * OUT + o] = acc[j];
+  for (int jj = 0; jj < 8; ++jj) red[threadIdx.x][jj] = acc[jj];
+  __syncthreads();
+  if (rg == 0 && act) {
+#pragma unroll
+    for (int jj = 0; jj < 8; ++jj) {
+      float a = acc[jj];
+      for (int q = 1; q < rgs; ++q) a += red[(wv + q * wpc) * 64 + lane][jj];
+      int k = kg * 8 + jj;
+      partial[(int64_t)blockIdx.x * CO * K + co * K + k] = a;
+    }
   }
 }
 

@@ -186,8 +186,12 @@ class _Conv2dNHWCFn(torch.autograd.Function):
         ph, pw = padding
         HO = (H + 2 * ph - KH) // sh + 1
         WO = (W + 2 * pw - KW) // sw + 1
-        col = ext.im2col_nhwc(x, KH, KW, sh, sw, ph, pw)
-        w2d = w.permute(0, 2, 3, 1).reshape(CO, KH * KW * CI).contiguous()
+        col = ext.im2col_nhwc(x, KH, KW, sh, sw, ph, pw)  # [M, Kp] (K padded to x8)
+        K = KH * KW * CI
+        w2d = w.permute(0, 2, 3, 1).reshape(CO, K)
+        if col.shape[1] != K:  # zero-pad weights to the padded K
+            w2d = F.pad(w2d, (0, col.shape[1] - K))
+        w2d = w2d.contiguous()
         y2d = ext.linear_fwd(col, w2d, b, relu)  # [B*HO*WO, CO] == NHWC
         ctx.save_for_backward(col, w2d, y2d)
         ctx.meta = (B, CI, H, W, CO, KH, KW, sh, sw, ph, pw, HO, WO, relu, b is not None)
@@ -208,8 +212,13 @@ class _Conv2dNHWCFn(torch.autograd.Function):
         sk = _choose_splitk(dz.shape[0], CO, col.shape[1])
         if ctx.needs_input_grad[1]:
             # (kh,kw,ci)-ordered wgrad -> param layout; autograd adds into the
-            # flat-bucket grad view, so overlap hooks still fire.
-            dw = ext.linear_wgrad(dz, col, sk).view(CO, KH, KW, CI).permute(0, 3, 1, 2)
+            # flat-bucket grad view, so overlap hooks still fire.  The col
+            # matrix is K-padded, so slice the pad columns off first.
+            dwp = ext.linear_wgrad(dz, col, sk)
+            K = KH * KW * CI
+            if dwp.shape[1] != K:
+                dwp = dwp[:, :K]
+            dw = dwp.reshape(CO, KH, KW, CI).permute(0, 3, 1, 2)
         if has_bias and ctx.needs_input_grad[2]:
             db = ext.bias_grad(dz)
 
