@@ -34,7 +34,7 @@ def main() -> int:
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=30)
     ap.add_argument("--warmup", type=int, default=10)
-    ap.add_argument("--batch", type=int, default=65536, help="per-GPU batch (weak scaling)")
+    ap.add_argument("--batch", type=int, default=131072, help="per-GPU batch (weak scaling)")
     ap.add_argument("--device", type=str, default=None)
     ap.add_argument("--hipgraph", action="store_true", help="capture the train step in a HIP graph")
     ap.add_argument("--model", type=str, default="mnist_mlp",
@@ -72,7 +72,7 @@ def main() -> int:
 
             model = MnistCNN()
     elif args.model == "resnet18":
-        if args.batch == 65536:  # default was sized for the MLP
+        if args.batch == 131072:  # default was sized for the MLP
             args.batch = 256
         if on_gpu:
             from sparktorch_amd.ops.modules import ResNet18Fused

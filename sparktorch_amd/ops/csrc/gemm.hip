@@ -187,8 +187,8 @@ __device__ __forceinline__ void stage_tile(const void* __restrict__ src, bf16raw
 template <bool SRC_F32>
 __device__ __forceinline__ void stage_load(const void* __restrict__ src, int row0, int rmax,
                                            int kt, int kmax, int64_t srow, int64_t skol,
-                                           int ones_row, bf16raw* __restrict__ regs) {
-  const int t = threadIdx.x;
+                                           int ones_row, bf16raw* __restrict__ regs,
+                                           int t) {
   const int mem_rows = ones_row >= 0 ? ones_row : rmax;
   if (skol == 1) {
     int r = t >> 1;
@@ -286,8 +286,8 @@ __device__ __forceinline__ void stage_load(const void* __restrict__ src, int row
 
 template <int LDSTRIDE>
 __device__ __forceinline__ void stage_write(bf16raw* __restrict__ lds, int64_t srow,
-                                            int64_t skol, const bf16raw* __restrict__ regs) {
-  const int t = threadIdx.x;
+                                            int64_t skol, const bf16raw* __restrict__ regs,
+                                            int t) {
   if (skol == 1) {
     int r = t >> 1;
     int k0 = (t & 1) * 16;
@@ -379,7 +379,7 @@ __device__ __forceinline__ void stage_swz_fallback(const bf16raw* __restrict__ s
 // stride of both operands).  B (small, L2-hot, possibly fp32 master
 // weights) always uses the register path.  K-loop is double-buffered.
 template <bool B_IS_F32, int EPI, bool SPLITK, int WR, int WC, bool AG>
-__global__ __launch_bounds__(256, 2) void gemm_kernel(const void* __restrict__ Ap,
+__global__ __launch_bounds__(WR * WC * 64, 2) void gemm_kernel(const void* __restrict__ Ap,
                                                    const void* __restrict__ Bp,
                                                    float* __restrict__ Cf,
                                                    bf16raw* __restrict__ Cb,
@@ -416,17 +416,18 @@ __global__ __launch_bounds__(256, 2) void gemm_kernel(const void* __restrict__ A
     k_end = min(K, k_begin + k_per_split);
   }
 
+  constexpr int THREADS = WR * WC * 64;
   const int wid = threadIdx.x / WAVE;
   const int lane = threadIdx.x % WAVE;
-  const int wr = WC == 1 ? wid : (wid >> 1);
-  const int wc = WC == 1 ? 0 : (wid & 1);
+  const int wr = wid / WC;
+  const int wc = wid % WC;
   const int l15 = lane & 15, kg = lane >> 4;
 
   const bool a_rows_full = (m0 + BMt <= M);
 
   floatx4 acc[4][4] = {};
 
-  if (AG) {
+  if constexpr (AG && THREADS == 256) {
     // A by async DMA (double-buffered prefetch), B by direct register stage.
 #define STAGE_SLAB(bufi, kt)                                                               \
   do {                                                                                     \
@@ -476,38 +477,58 @@ __global__ __launch_bounds__(256, 2) void gemm_kernel(const void* __restrict__ A
   } else {
     // T14 register pipeline, single LDS buffer: hold slab t in registers,
     // write it to LDS after the barrier, immediately issue slab t+1's
-    // loads, then run slab t's MFMAs while those loads fly.
-    constexpr int AH = (BMt / 128) * (BKT / 32);  // 32-k x 128-row chunks of A
-    constexpr int BH = BKT / 32;
+    // loads, then run slab t's MFMAs while those loads fly.  At 512
+    // threads (the 128x256 wgrad tile) the two thread-halves split the
+    // work: half g stages A k-half g and B row-half g (both k-halves).
+    constexpr int NG = THREADS / 256;                  // 1 or 2
+    constexpr int AH = (BMt / 128) * (BKT / 32) / NG;  // A chunks per group
+    constexpr int BH = (BNt < 128 ? 128 : BNt) / 128 * (BKT / 32) / NG;
     alignas(16) bf16raw rA[AH * 16];
     alignas(16) bf16raw rB[BH * 16];
+    const int grp = NG == 1 ? 0 : (threadIdx.x >> 8);
+    const int gt = NG == 1 ? threadIdx.x : (threadIdx.x & 255);
 
 #define LOAD_SLAB(kt)                                                                      \
   do {                                                                                     \
-    _Pragma("unroll") for (int ro = 0; ro < BMt / 128; ++ro) {                             \
+    if (NG == 1) {                                                                         \
+      _Pragma("unroll") for (int ro = 0; ro < BMt / 128; ++ro) {                           \
+        _Pragma("unroll") for (int kh = 0; kh < BKT / 32; ++kh)                            \
+            stage_load<false>(Ap, m0 + ro * 128, M, (kt) + kh * 32, k_end, sam, sak, -1,   \
+                              rA + (ro * (BKT / 32) + kh) * 16, gt);                       \
+      }                                                                                    \
       _Pragma("unroll") for (int kh = 0; kh < BKT / 32; ++kh)                              \
-          stage_load<false>(Ap, m0 + ro * 128, M, (kt) + kh * 32, k_end, sam, sak, -1,     \
-                            rA + (ro * (BKT / 32) + kh) * 16);                             \
+          stage_load<B_IS_F32>(Bp, n0, N, (kt) + kh * 32, k_end, sbn, sbk, ones_row,       \
+                               rB + kh * 16, gt);                                          \
+    } else {                                                                               \
+      /* group g: A 128x32 k-half g; B row-half (n0+g*128) both k-halves */                \
+      stage_load<false>(Ap, m0, M, (kt) + grp * 32, k_end, sam, sak, -1, rA, gt);          \
+      _Pragma("unroll") for (int kh = 0; kh < BKT / 32; ++kh)                              \
+          stage_load<B_IS_F32>(Bp, n0 + grp * 128, N, (kt) + kh * 32, k_end, sbn, sbk,     \
+                               ones_row, rB + kh * 16, gt);                                \
     }                                                                                      \
-    _Pragma("unroll") for (int kh = 0; kh < BKT / 32; ++kh)                                \
-        stage_load<B_IS_F32>(Bp, n0, N, (kt) + kh * 32, k_end, sbn, sbk, ones_row,         \
-                             rB + kh * 16);                                                \
   } while (0)
 
     if (k_begin < k_end) LOAD_SLAB(k_begin);
 
     for (int kt = k_begin; kt < k_end; kt += BKT) {
       if (kt > k_begin) __syncthreads();  // prior slab's MFMA reads done
+      if (NG == 1) {
 #pragma unroll
-      for (int ro = 0; ro < BMt / 128; ++ro) {
+        for (int ro = 0; ro < BMt / 128; ++ro) {
+#pragma unroll
+          for (int kh = 0; kh < BKT / 32; ++kh)
+            stage_write<LP>(As[0] + ro * 128 * LP + kh * 32, sam, sak,
+                            rA + (ro * (BKT / 32) + kh) * 16, gt);
+        }
 #pragma unroll
         for (int kh = 0; kh < BKT / 32; ++kh)
-          stage_write<LP>(As[0] + ro * 128 * LP + kh * 32, sam, sak,
-                          rA + (ro * (BKT / 32) + kh) * 16);
-      }
+          stage_write<LP>(Bs[0] + kh * 32, sbn, sbk, rB + kh * 16, gt);
+      } else {
+        stage_write<LP>(As[0] + grp * 32, sam, sak, rA, gt);
 #pragma unroll
-      for (int kh = 0; kh < BKT / 32; ++kh)
-        stage_write<LP>(Bs[0] + kh * 32, sbn, sbk, rB + kh * 16);
+        for (int kh = 0; kh < BKT / 32; ++kh)
+          stage_write<LP>(Bs[0] + grp * 128 * LP + kh * 32, sbn, sbk, rB + kh * 16, gt);
+      }
       __syncthreads();  // publish
 
       if (kt + BKT < k_end) LOAD_SLAB(kt + BKT);  // loads span the MFMAs below
@@ -594,15 +615,23 @@ extern "C" hipError_t launch_gemm_bf16(const void* A, const void* B, int b_is_f3
   // 2 waves/SIMD (tools/gemm_bench.hip: stem fwd 686us vs 754us, l1 conv
   // 348us vs 388us)
   const bool narrow = (N <= 64 && M > 256);
-  const int bm = narrow ? 256 : BM, bn = narrow ? 64 : BN;
-  dim3 grid((unsigned)ceil_div_i64(M, bm), (unsigned)ceil_div_i64(N, bn), (unsigned)splitk);
-
   // A is DMA-staged (glds) when k-contiguous with 16B-aligned rows
-  const bool ag = (sak == 1) && (sam % 8 == 0);
+  const bool ag0 = (sak == 1) && (sam % 8 == 0);
+  // 128x256 (2x4-wave, 512-thread) tiles for the register-staged big-N
+  // shapes (wgrad): halves the A re-reads, doubles MACs per staged byte
+  const bool wide = !narrow && !ag0 && N >= 256 && M > 64;
+  const int bm = narrow ? 256 : BM, bn = narrow ? 64 : (wide ? 256 : BN);
+  dim3 grid((unsigned)ceil_div_i64(M, bm), (unsigned)ceil_div_i64(N, bn), (unsigned)splitk);
+  if (wide) block = dim3(512);
+
+  const bool ag = ag0;
 
 #define DISPATCH(BF32, EPIC, SPK)                                                            \
   do {                                                                                       \
-    if (narrow && ag)                                                                        \
+    if (wide)                                                                                \
+      gemm_kernel<BF32, EPIC, SPK, 2, 4, false><<<grid, block, 0, stream>>>(                 \
+          A, B, Cf, Cb, bias, M, N, K, sam, sak, sbk, sbn, kps, Db, ones_row);               \
+    else if (narrow && ag)                                                                   \
       gemm_kernel<BF32, EPIC, SPK, 4, 1, true><<<grid, block, 0, stream>>>(                  \
           A, B, Cf, Cb, bias, M, N, K, sam, sak, sbk, sbn, kps, Db, ones_row);               \
     else if (narrow)                                                                         \

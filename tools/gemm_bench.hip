@@ -50,11 +50,10 @@ double run_wgrad(const bf16raw* dz, const bf16raw* col, float* dw, int CO, int K
   CK(hipEventCreate(&a));
   CK(hipEventCreate(&b));
   auto launch = [&]() {
-    // A[m=co, k=row] = dz[row*CO+co] -> sam=1, sak=CO ; B[k=row, n=kc] = col[row*Kcol+kc]
-    // staged transposed: srow := sbn=1, skol := sbk=Kcol
-    gemm_kernel<false, EPI_F32, true, 2, 2, false><<<grid, 256>>>(
-        dz, col, dw, nullptr, nullptr, CO, Kcol, (int)R, /*sam*/ 1, /*sak*/ CO, /*sbk*/ Kcol,
-        /*sbn*/ 1, kps, nullptr, -1);
+    // through the production dispatch (selects the 128x256 wide tile for
+    // N >= 256 register-staged shapes)
+    launch_gemm_bf16(dz, col, 0, dw, nullptr, nullptr, CO, Kcol, (int)R, 1, CO, Kcol, 1,
+                     EPI_F32, splitk, nullptr, -1, 0);
   };
   launch();
   CK(hipDeviceSynchronize());
