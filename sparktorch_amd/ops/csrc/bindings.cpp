@@ -577,8 +577,8 @@ at::Tensor im2col_nhwc(at::Tensor x, int64_t KH, int64_t KW, int64_t sh, int64_t
   // staging, vector wgrad) at the cost of <= 7 zero columns
   int64_t K = KH * KW * CI;
   int64_t Kp = (K + 7) & ~7LL;
-  auto col = Kp == K ? at::empty({(int64_t)B * HO * WO, Kp}, x.options())
-                     : at::zeros({(int64_t)B * HO * WO, Kp}, x.options());
+  // pad columns are zeroed by the im2col kernels themselves
+  auto col = at::empty({(int64_t)B * HO * WO, Kp}, x.options());
   CHECK_HIP(launch_im2col_nhwc((const bf16raw*)x.data_ptr(), (bf16raw*)col.data_ptr(), B, CI, H,
                                W, (int)KH, (int)KW, HO, WO, (int)sh, (int)sw, (int)ph, (int)pw,
                                (int)Kp, cur_stream()));

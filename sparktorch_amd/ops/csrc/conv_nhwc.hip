@@ -104,6 +104,9 @@ __global__ void im2col_nhwc_rowcopy_kernel(const bf16raw* __restrict__ x,
     int w0 = wo * sw - pw;
     int64_t m = ((int64_t)b * HO + ho) * WO + wo;
     bf16raw* dst = col + m * K + (int64_t)kh * KW * CI;
+    if (kh == 0) {  // this (m, kh=0) thread also zeros the row's pad columns
+      for (int j = KH * KW * CI; j < Kp; ++j) col[m * K + j] = 0;
+    }
     if (h < 0 || h >= H) {
       for (int j = 0; j < KW * CI; ++j) dst[j] = 0;
       continue;
