@@ -557,9 +557,50 @@ __global__ __launch_bounds__(WR * WC * 64, 2) void gemm_kernel(const void* __res
 #undef LOAD_SLAB
   }
 
-  // epilogue: C/D fragment layout col = lane&15, row = 4*(lane>>4) + reg
+  // epilogue.  bf16 outputs go through a per-wave LDS transpose so each
+  // lane stores one contiguous 16 B octet (scalar 2 B stores measured the
+  // C-write path at ~2.3 TB/s vs ~4.4 for reads); the MFMA fragment layout
+  // (col = lane&15, row = 4*(lane>>4)+reg) cannot produce contiguous
+  // per-lane stores directly.  fp32/split-K keeps the scalar path (atomics).
   const int m_base = m0 + wr * 64;
   const int n_base = n0 + wc * 64;
+  if (EPI != EPI_F32 && (N & 7) == 0 && n_base + 64 <= N) {
+    // per-wave [16][68] fp32 scratch inside the (now idle) LDS staging
+    // buffers: As holds 4x(16x68x4B) for the 4-wave tiles; the 8-wave wide
+    // tile needs 34.8 KB and uses Bs (36.9 KB there)
+    constexpr int EPAD = 68;
+    float* ep = (THREADS == 512 ? (float*)&Bs[0][0] : (float*)&As[0][0]) + wid * 16 * EPAD;
+    const int orow = lane >> 2;          // 0..15 output row of the mi-slice
+    const int oct = lane & 3;            // which 16-col half-octet pair
+#pragma unroll
+    for (int mi = 0; mi < 4; ++mi) {
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) ep[(kg * 4 + r) * EPAD + ni * 16 + l15] = acc[mi][ni][r];
+      __builtin_amdgcn_s_waitcnt(0);  // lgkm: own-wave LDS writes visible
+      int m = m_base + mi * 16 + orow;
+      if (m < M) {
+#pragma unroll
+        for (int h = 0; h < 2; ++h) {
+          int c0 = oct * 16 + h * 8;
+          int n = n_base + c0;
+          alignas(16) short outp[8];
+          const float* src = ep + orow * EPAD + c0;
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            float val = src[j];
+            if (EPI == EPI_BIAS || EPI == EPI_BIAS_RELU) val += bias[n + j];
+            if (EPI == EPI_BIAS_RELU || EPI == EPI_RELU) val = fmaxf(val, 0.f);
+            outp[j] = (short)f32_to_bf16(val);
+          }
+          *(shortx8*)(Cb + (int64_t)m * N + n) = *(const shortx8*)outp;
+        }
+      }
+      __builtin_amdgcn_s_waitcnt(0);  // all lanes done reading before reuse
+    }
+    return;
+  }
 #pragma unroll
   for (int mi = 0; mi < 4; ++mi) {
 #pragma unroll
