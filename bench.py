@@ -187,9 +187,8 @@ def run_infer(args, model, device, on_gpu) -> int:
         x = x.to(torch.bfloat16)
 
         def step():
-            runner._static_in.copy_(x)
-            runner._graph.replay() if runner._graph else runner(x.cpu())
-        runner(torch.randn(bs, in_dim))  # capture
+            runner.replay_device(x)
+        step()  # capture
     else:
         def step():
             with torch.no_grad():

@@ -48,6 +48,17 @@ class GraphedForward:
         self._feat_shape = tuple(feat_shape)
 
     @torch.no_grad()
+    def replay_device(self, x_dev: torch.Tensor) -> torch.Tensor:
+        """Replay on a device-resident full batch (the serving hot loop: no
+        host copy, no ragged handling)."""
+        feat_shape = tuple(x_dev.shape[1:])
+        if self._graph is None or self._feat_shape != feat_shape:
+            self._capture(feat_shape)
+        self._static_in.copy_(x_dev)
+        self._graph.replay()
+        return self._static_out
+
+    @torch.no_grad()
     def __call__(self, x_cpu: torch.Tensor) -> torch.Tensor:
         feat_shape = tuple(x_cpu.shape[1:])
         n = x_cpu.shape[0]
