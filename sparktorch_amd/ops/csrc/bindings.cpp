@@ -63,11 +63,12 @@ hipError_t launch_bn_apply_nhwc(const bf16raw*, const bf16raw*, bf16raw*, const 
                                 const float*, const float*, const float*, int, int64_t, int,
                                 hipStream_t);
 hipError_t launch_bn_bwd_reduce_nhwc(const bf16raw*, const bf16raw*, const bf16raw*,
-                                     const float*, const float*, float*, float*, int64_t, int,
-                                     int, float*, int, hipStream_t);
+                                     const float*, const float*, const float*, const float*,
+                                     float*, float*, int64_t, int, int, float*, int,
+                                     hipStream_t);
 hipError_t launch_bn_bwd_dx_nhwc(const bf16raw*, const bf16raw*, const bf16raw*, const float*,
                                  const float*, const float*, const float*, const float*,
-                                 bf16raw*, int, int64_t, float, int, hipStream_t);
+                                 const float*, bf16raw*, int, int64_t, float, int, hipStream_t);
 }
 
 #define CHECK_HIP(err)                                                              \
@@ -711,8 +712,8 @@ at::Tensor bn_apply_nhwc(at::Tensor x, c10::optional<at::Tensor> res, at::Tensor
 }
 
 void bn_bwd_reduce_nhwc(at::Tensor dy, c10::optional<at::Tensor> yrelu, at::Tensor x,
-                        at::Tensor mean, at::Tensor invstd, at::Tensor dbeta,
-                        at::Tensor dgamma) {
+                        at::Tensor mean, at::Tensor invstd, at::Tensor gamma, at::Tensor beta,
+                        at::Tensor dbeta, at::Tensor dgamma) {
   check_gpu_contig(dy, at::kBFloat16, "dy");
   check_gpu_contig(x, at::kBFloat16, "x");
   check_gpu_contig(dbeta, at::kFloat, "dbeta");
@@ -734,14 +735,15 @@ void bn_bwd_reduce_nhwc(at::Tensor dy, c10::optional<at::Tensor> yrelu, at::Tens
   }
   CHECK_HIP(launch_bn_bwd_reduce_nhwc((const bf16raw*)dy.data_ptr(), yp,
                                       (const bf16raw*)x.data_ptr(), mean.data_ptr<float>(),
-                                      invstd.data_ptr<float>(), dbeta.data_ptr<float>(),
+                                      invstd.data_ptr<float>(), gamma.data_ptr<float>(),
+                                      beta.data_ptr<float>(), dbeta.data_ptr<float>(),
                                       dgamma.data_ptr<float>(), M, C, nsplit, sp, S,
                                       cur_stream()));
 }
 
 at::Tensor bn_bwd_dx_nhwc(at::Tensor dy, c10::optional<at::Tensor> yrelu, at::Tensor x,
-                          at::Tensor mean, at::Tensor invstd, at::Tensor gamma, at::Tensor dbeta,
-                          at::Tensor dgamma, bool train_stats) {
+                          at::Tensor mean, at::Tensor invstd, at::Tensor gamma, at::Tensor beta,
+                          at::Tensor dbeta, at::Tensor dgamma, bool train_stats) {
   check_gpu_contig(dy, at::kBFloat16, "dy");
   check_gpu_contig(x, at::kBFloat16, "x");
   const bf16raw* yp = nullptr;
@@ -752,9 +754,9 @@ at::Tensor bn_bwd_dx_nhwc(at::Tensor dy, c10::optional<at::Tensor> yrelu, at::Te
   CHECK_HIP(launch_bn_bwd_dx_nhwc((const bf16raw*)dy.data_ptr(), yp,
                                   (const bf16raw*)x.data_ptr(), mean.data_ptr<float>(),
                                   invstd.data_ptr<float>(), gamma.data_ptr<float>(),
-                                  dbeta.data_ptr<float>(), dgamma.data_ptr<float>(),
-                                  (bf16raw*)dx.data_ptr(), C, x.numel(), inv_count,
-                                  train_stats ? 1 : 0, cur_stream()));
+                                  beta.data_ptr<float>(), dbeta.data_ptr<float>(),
+                                  dgamma.data_ptr<float>(), (bf16raw*)dx.data_ptr(), C,
+                                  x.numel(), inv_count, train_stats ? 1 : 0, cur_stream()));
   return dx;
 }
 

@@ -364,11 +364,15 @@ __global__ void bn_bwd_reduce_nhwc_kernel(const bf16raw* __restrict__ dy,
   }
 }
 
-__global__ void bn_bwd_reduce_nhwc_vec_kernel(const bf16raw* __restrict__ dy,
-                                              const bf16raw* __restrict__ yrelu,
+// relu'd BN backward does NOT read the saved output: the mask is
+// recomputed as (gamma*xhat + beta) > 0 from the already-loaded x (2 VALU
+// ops instead of a third 2-byte stream).
+__global__ void bn_bwd_reduce_nhwc_vec_kernel(const bf16raw* __restrict__ dy, int relu,
                                               const bf16raw* __restrict__ x,
                                               const float* __restrict__ mean,
                                               const float* __restrict__ invstd,
+                                              const float* __restrict__ gamma,
+                                              const float* __restrict__ beta,
                                               float* __restrict__ pdb,
                                               float* __restrict__ pdg, int64_t M, int C) {
   // partials stored per block (plain stores; see atomic-contention note on
@@ -381,32 +385,26 @@ __global__ void bn_bwd_reduce_nhwc_vec_kernel(const bf16raw* __restrict__ dy,
   int64_t per = ceil_div_i64(M, gridDim.x);
   int64_t lo = (int64_t)blockIdx.x * per;
   int64_t hi = lo + per < M ? lo + per : M;
-  float mu[8], is[8];
+  float mu[8], is[8], ga[8], be[8];
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
     mu[j] = mean[c0 + j];
     is[j] = invstd[c0 + j];
+    ga[j] = gamma[c0 + j];
+    be[j] = beta[c0 + j];
   }
   float sdy[8] = {0, 0, 0, 0, 0, 0, 0, 0}, sdyx[8] = {0, 0, 0, 0, 0, 0, 0, 0};
   for (int64_t r = lo + rg; r < hi; r += rpg) {
     int64_t off = r * C + c0;
     const shortx8 g8 = *(const shortx8*)(dy + off);
     const shortx8 x8 = *(const shortx8*)(x + off);
-    if (yrelu != nullptr) {
-      const shortx8 y8 = *(const shortx8*)(yrelu + off);
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        float g = bf16_to_f32((bf16raw)y8[j]) > 0.f ? bf16_to_f32((bf16raw)g8[j]) : 0.f;
-        sdy[j] += g;
-        sdyx[j] += g * (bf16_to_f32((bf16raw)x8[j]) - mu[j]) * is[j];
-      }
-    } else {
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        float g = bf16_to_f32((bf16raw)g8[j]);
-        sdy[j] += g;
-        sdyx[j] += g * (bf16_to_f32((bf16raw)x8[j]) - mu[j]) * is[j];
-      }
+    for (int j = 0; j < 8; ++j) {
+      float xh = (bf16_to_f32((bf16raw)x8[j]) - mu[j]) * is[j];
+      float g = bf16_to_f32((bf16raw)g8[j]);
+      if (relu && fmaf(ga[j], xh, be[j]) <= 0.f) g = 0.f;
+      sdy[j] += g;
+      sdyx[j] += g * xh;
     }
   }
   __shared__ float l0[256][8], l1[256][8];
@@ -461,11 +459,13 @@ __global__ void bn_bwd_reduce_finalize_kernel(const float* __restrict__ pdb,
 
 extern "C" hipError_t launch_bn_bwd_reduce_nhwc(const bf16raw* dy, const bf16raw* yrelu,
                                                 const bf16raw* x, const float* mean,
-                                                const float* invstd, float* dbeta,
+                                                const float* invstd, const float* gamma,
+                                                const float* beta, float* dbeta,
                                                 float* dgamma, int64_t M, int C, int nsplit,
                                                 float* scratch, int S, hipStream_t stream) {
   if (scratch != nullptr) {
-    bn_bwd_reduce_nhwc_vec_kernel<<<S, 256, 0, stream>>>(dy, yrelu, x, mean, invstd, scratch,
+    bn_bwd_reduce_nhwc_vec_kernel<<<S, 256, 0, stream>>>(dy, yrelu != nullptr ? 1 : 0, x, mean,
+                                                         invstd, gamma, beta, scratch,
                                                          scratch + (int64_t)S * C, M, C);
     HIP_CHECK_LAUNCH();
     bn_bwd_reduce_finalize_kernel<<<C, 256, 0, stream>>>(scratch, scratch + (int64_t)S * C, S,
@@ -510,70 +510,55 @@ __global__ void bn_bwd_dx_nhwc_kernel(const bf16raw* __restrict__ dy,
 // a = gamma*invstd, b = gamma*invstd*sum_dy/N, d = gamma*invstd^2*sum_dyxhat/N
 // (algebraically dx = gamma*is*(g - sdy/N - xhat*sdyx/N)); channel octet is
 // loop-invariant as in bn_apply_nhwc_vec_kernel.
-__global__ void bn_bwd_dx_nhwc_vec_kernel(const bf16raw* __restrict__ dy,
-                                          const bf16raw* __restrict__ yrelu,
+__global__ void bn_bwd_dx_nhwc_vec_kernel(const bf16raw* __restrict__ dy, int relu,
                                           const bf16raw* __restrict__ x,
                                           const float* __restrict__ mean,
                                           const float* __restrict__ invstd,
                                           const float* __restrict__ gamma,
+                                          const float* __restrict__ beta,
                                           const float* __restrict__ dbeta,
                                           const float* __restrict__ dgamma,
                                           bf16raw* __restrict__ dx, int C, int64_t total8,
                                           float inv_count, int train_stats) {
   int64_t i0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int c0 = (int)((i0 << 3) % C);
-  float a[8], b[8], d[8], m[8];
+  float a[8], b[8], d[8], m[8], is8[8], ga[8], be[8];
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
     int c = c0 + j;
     float is = invstd[c];
     float gis = gamma[c] * is;
     a[j] = gis;
+    is8[j] = is;
+    ga[j] = gamma[c];
+    be[j] = beta != nullptr ? beta[c] : 0.f;
+    m[j] = mean[c];
     if (train_stats) {
       b[j] = gis * dbeta[c] * inv_count;
       d[j] = gis * is * dgamma[c] * inv_count;
-      m[j] = mean[c];
     } else {
       b[j] = 0.f;
       d[j] = 0.f;
-      m[j] = 0.f;
     }
   }
   for (int64_t i = i0; i < total8; i += (int64_t)gridDim.x * blockDim.x) {
     int64_t off = i << 3;
     const shortx8 g8 = *(const shortx8*)(dy + off);
     shortx8 out;
-    if (train_stats) {
+    if (train_stats || relu) {
       const shortx8 x8 = *(const shortx8*)(x + off);
-      if (yrelu != nullptr) {
-        const shortx8 y8 = *(const shortx8*)(yrelu + off);
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          float g = bf16_to_f32((bf16raw)y8[j]) > 0.f ? bf16_to_f32((bf16raw)g8[j]) : 0.f;
-          float o = a[j] * g - b[j] - (bf16_to_f32((bf16raw)x8[j]) - m[j]) * d[j];
-          out[j] = (short)f32_to_bf16(o);
-        }
-      } else {
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          float g = bf16_to_f32((bf16raw)g8[j]);
-          float o = a[j] * g - b[j] - (bf16_to_f32((bf16raw)x8[j]) - m[j]) * d[j];
-          out[j] = (short)f32_to_bf16(o);
-        }
+      for (int j = 0; j < 8; ++j) {
+        float dxm = bf16_to_f32((bf16raw)x8[j]) - m[j];
+        float g = bf16_to_f32((bf16raw)g8[j]);
+        if (relu && fmaf(ga[j], dxm * is8[j], be[j]) <= 0.f) g = 0.f;
+        float o = train_stats ? (a[j] * g - b[j] - dxm * d[j]) : a[j] * g;
+        out[j] = (short)f32_to_bf16(o);
       }
     } else {
-      if (yrelu != nullptr) {
-        const shortx8 y8 = *(const shortx8*)(yrelu + off);
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          float g = bf16_to_f32((bf16raw)y8[j]) > 0.f ? bf16_to_f32((bf16raw)g8[j]) : 0.f;
-          out[j] = (short)f32_to_bf16(a[j] * g);
-        }
-      } else {
-#pragma unroll
-        for (int j = 0; j < 8; ++j)
-          out[j] = (short)f32_to_bf16(a[j] * bf16_to_f32((bf16raw)g8[j]));
-      }
+      for (int j = 0; j < 8; ++j)
+        out[j] = (short)f32_to_bf16(a[j] * bf16_to_f32((bf16raw)g8[j]));
     }
     *(shortx8*)(dx + off) = out;
   }
@@ -582,16 +567,17 @@ __global__ void bn_bwd_dx_nhwc_vec_kernel(const bf16raw* __restrict__ dy,
 extern "C" hipError_t launch_bn_bwd_dx_nhwc(const bf16raw* dy, const bf16raw* yrelu,
                                             const bf16raw* x, const float* mean,
                                             const float* invstd, const float* gamma,
-                                            const float* dbeta, const float* dgamma,
-                                            bf16raw* dx, int C, int64_t total, float inv_count,
-                                            int train_stats, hipStream_t stream) {
+                                            const float* beta, const float* dbeta,
+                                            const float* dgamma, bf16raw* dx, int C,
+                                            int64_t total, float inv_count, int train_stats,
+                                            hipStream_t stream) {
   if ((C & 7) == 0 && C <= 2048 && (2048 % C) == 0) {
     int64_t total8 = total >> 3;
     int64_t g = ceil_div_i64(total8, 256);
     int grid = (int)(g > 8192 ? 8192 : (g < 1 ? 1 : g));
-    bn_bwd_dx_nhwc_vec_kernel<<<grid, 256, 0, stream>>>(dy, yrelu, x, mean, invstd, gamma,
-                                                        dbeta, dgamma, dx, C, total8, inv_count,
-                                                        train_stats);
+    bn_bwd_dx_nhwc_vec_kernel<<<grid, 256, 0, stream>>>(dy, yrelu != nullptr ? 1 : 0, x, mean,
+                                                        invstd, gamma, beta, dbeta, dgamma, dx,
+                                                        C, total8, inv_count, train_stats);
   } else {
     int64_t g = ceil_div_i64(total, 256);
     int grid = (int)(g > 8192 ? 8192 : (g < 1 ? 1 : g));

@@ -279,11 +279,12 @@ class _BatchNorm2dNHWCFn(torch.autograd.Function):
         else:
             dgamma_buf = torch.zeros_like(mean)
             dbeta_buf = torch.zeros_like(mean)
-        ext.bn_bwd_reduce_nhwc(dy, yrelu, x, mean, invstd, dbeta_buf, dgamma_buf)
+        beta = ctx.beta_ref
+        ext.bn_bwd_reduce_nhwc(dy, yrelu, x, mean, invstd, gamma, beta, dbeta_buf, dgamma_buf)
         dx = None
         if ctx.needs_input_grad[0]:
-            dx = ext.bn_bwd_dx_nhwc(dy, yrelu, x, mean, invstd, gamma, dbeta_buf, dgamma_buf,
-                                    ctx.train_stats)
+            dx = ext.bn_bwd_dx_nhwc(dy, yrelu, x, mean, invstd, gamma, beta, dbeta_buf,
+                                    dgamma_buf, ctx.train_stats)
         if direct:
             g_notify()
             b_notify()
