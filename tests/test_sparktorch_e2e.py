@@ -231,3 +231,34 @@ def test_training_improves_fit(data_df):
         trained_loss = nn.MSELoss()(trained(x), y).item()
         init_loss = nn.MSELoss()(net(x), y).item()
     assert trained_loss < init_loss
+
+
+def test_use_barrier_and_explicit_cpu_device(data_df):
+    """Reference test_barrier + test_cpu (test_sparktorch.py:166-180,238-253):
+    useBarrier hogwild mode plus an explicit device='cpu' sync fit."""
+    from sparktorch_amd.compat.local import free_port
+
+    obj = serialize_torch_obj(Net(), nn.MSELoss(), torch.optim.Adam, lr=0.01)
+    model = SparkTorch(
+        inputCol="features",
+        labelCol="label",
+        predictionCol="predicted",
+        torchObj=obj,
+        iters=3,
+        mode="hogwild",
+        useBarrier=True,
+        port=free_port(),
+        device="cpu",
+    ).fit(data_df)
+    assert model.transform(data_df).count() == 400
+
+    model2 = SparkTorch(
+        inputCol="features",
+        labelCol="label",
+        predictionCol="predicted",
+        torchObj=obj,
+        iters=3,
+        mode="synchronous",
+        device="cpu",
+    ).fit(data_df)
+    assert model2.transform(data_df).count() == 400
