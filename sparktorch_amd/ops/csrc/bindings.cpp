@@ -13,7 +13,8 @@ typedef uint16_t bf16raw;
 
 extern "C" {
 hipError_t launch_fused_adam(float*, const float*, float*, float*, int64_t, float, float, float,
-                             float, float, float, float, float, int, hipStream_t);
+                             float, float, float, float, float, int, const int*, hipStream_t);
+hipError_t launch_increment_i32(int*, hipStream_t);
 hipError_t launch_fused_sgd(float*, const float*, float*, int64_t, float, float, float, float,
                             float, int, int, int, hipStream_t);
 hipError_t launch_relu_bwd(const bf16raw*, const bf16raw*, bf16raw*, int64_t, hipStream_t);
@@ -88,13 +89,23 @@ static void check_gpu_contig(const at::Tensor& t, c10::ScalarType dt, const char
 
 void fused_adam(at::Tensor p, at::Tensor g, at::Tensor m, at::Tensor v, double lr, double b1,
                 double b2, double eps, double wd, double bc1, double bc2, double gscale,
-                bool adamw) {
+                bool adamw, c10::optional<at::Tensor> step_dev) {
   check_gpu_contig(p, at::kFloat, "p");
   check_gpu_contig(g, at::kFloat, "g");
+  const int* sp = nullptr;
+  if (step_dev.has_value()) {
+    check_gpu_contig(*step_dev, at::kInt, "step_dev");
+    sp = step_dev->data_ptr<int>();
+  }
   CHECK_HIP(launch_fused_adam(p.data_ptr<float>(), g.data_ptr<float>(), m.data_ptr<float>(),
                               v.data_ptr<float>(), p.numel(), (float)lr, (float)b1, (float)b2,
                               (float)eps, (float)wd, (float)bc1, (float)bc2, (float)gscale,
-                              adamw ? 1 : 0, cur_stream()));
+                              adamw ? 1 : 0, sp, cur_stream()));
+}
+
+void increment_i32(at::Tensor x) {
+  check_gpu_contig(x, at::kInt, "x");
+  CHECK_HIP(launch_increment_i32(x.data_ptr<int>(), cur_stream()));
 }
 
 void fused_sgd(at::Tensor p, at::Tensor g, at::Tensor buf, double lr, double momentum, double wd,
@@ -762,6 +773,7 @@ at::Tensor bn_bwd_dx_nhwc(at::Tensor dy, c10::optional<at::Tensor> yrelu, at::Te
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_adam", &fused_adam, "fused Adam on a flat bucket");
+  m.def("increment_i32", &increment_i32, "device step counter += 1 (graph-capturable)");
   m.def("fused_sgd", &fused_sgd, "fused SGD on a flat bucket");
   m.def("relu_bwd", &relu_bwd, "dz = dy * (y>0)");
   m.def("bias_grad", &bias_grad, "column-sum of dz");

@@ -25,10 +25,19 @@ static inline int ew_grid(int64_t work_items, int block) {
 // fused Adam
 // ---------------------------------------------------------------------------
 
+// step_dev != nullptr: read the (already incremented) step from device and
+// compute the bias corrections in-kernel — required for hipGraph capture,
+// where host-computed scalars would be frozen into the graph.
 __global__ void fused_adam_kernel(float* __restrict__ p, const float* __restrict__ g,
                                   float* __restrict__ m, float* __restrict__ v, int64_t n,
                                   float lr, float b1, float b2, float eps, float wd, float bc1,
-                                  float bc2, float gscale, int adamw) {
+                                  float bc2, float gscale, int adamw,
+                                  const int* __restrict__ step_dev) {
+  if (step_dev != nullptr) {
+    float t = (float)*step_dev;
+    bc1 = 1.f - __powf(b1, t);
+    bc2 = 1.f - __powf(b2, t);
+  }
   int64_t nvec = n >> 2;
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
@@ -82,13 +91,23 @@ __global__ void fused_adam_kernel(float* __restrict__ p, const float* __restrict
   }
 }
 
+__global__ void increment_i32_kernel(int* x) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) *x += 1;
+}
+
+extern "C" hipError_t launch_increment_i32(int* x, hipStream_t stream) {
+  increment_i32_kernel<<<1, 1, 0, stream>>>(x);
+  HIP_CHECK_LAUNCH();
+  return hipSuccess;
+}
+
 extern "C" hipError_t launch_fused_adam(float* p, const float* g, float* m, float* v, int64_t n,
                                         float lr, float b1, float b2, float eps, float wd,
                                         float bc1, float bc2, float gscale, int adamw,
-                                        hipStream_t stream) {
+                                        const int* step_dev, hipStream_t stream) {
   int block = 256;
   fused_adam_kernel<<<ew_grid(n / 4 + 1, block), block, 0, stream>>>(
-      p, g, m, v, n, lr, b1, b2, eps, wd, bc1, bc2, gscale, adamw);
+      p, g, m, v, n, lr, b1, b2, eps, wd, bc1, bc2, gscale, adamw, step_dev);
   HIP_CHECK_LAUNCH();
   return hipSuccess;
 }

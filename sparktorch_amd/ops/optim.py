@@ -58,9 +58,14 @@ class FusedAdam(_FusedOptimizerBase):
         self.adamw = adamw
         self.exp_avg: List[torch.Tensor] = []
         self.exp_avg_sq: List[torch.Tensor] = []
+        self.step_dev: Optional[torch.Tensor] = None  # device step counter
         for flat_p, _ in buckets.flat_pairs():
             self.exp_avg.append(torch.zeros_like(flat_p))
             self.exp_avg_sq.append(torch.zeros_like(flat_p))
+            if self.step_dev is None and flat_p.is_cuda:
+                # bias corrections are computed in-kernel from this counter,
+                # so a hipGraph-captured step stays correct across replays
+                self.step_dev = torch.zeros(1, dtype=torch.int32, device=flat_p.device)
 
     @torch.no_grad()
     def step(self, grad_scale: float = 1.0) -> None:
@@ -68,8 +73,12 @@ class FusedAdam(_FusedOptimizerBase):
         t = self.step_count
         bc1 = 1.0 - self.beta1**t
         bc2 = 1.0 - self.beta2**t
+        stepped_dev = False
         for i, (p, g) in enumerate(self.buckets.flat_pairs()):
             if p.is_cuda:
+                if not stepped_dev and self.step_dev is not None:
+                    ops.ext().increment_i32(self.step_dev)
+                    stepped_dev = True
                 ops.ext().fused_adam(
                     p,
                     g,
@@ -84,6 +93,7 @@ class FusedAdam(_FusedOptimizerBase):
                     bc2,
                     grad_scale,
                     self.adamw,
+                    self.step_dev,
                 )
             else:
                 m, v = self.exp_avg[i], self.exp_avg_sq[i]

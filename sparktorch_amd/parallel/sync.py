@@ -166,6 +166,21 @@ class SyncTrainer:
                 )
             static_x = x.clone()
             static_y = y.clone() if y is not None else None
+            # Warmup steps REALLY execute (capture itself records without
+            # executing) — snapshot the training state so the first replay
+            # starts exactly where eager would.
+            snap_params = [b.flat_param.clone() for b in self.buckets.buckets]
+            snap_opt = None
+            if self.optimizer is not None:
+                sd = getattr(self.optimizer, "step_dev", None)
+                snap_opt = (
+                    self.optimizer.step_count,
+                    [t.clone() for t in getattr(self.optimizer, "exp_avg", [])],
+                    [t.clone() for t in getattr(self.optimizer, "exp_avg_sq", [])],
+                    [t.clone() if t is not None else None
+                     for t in getattr(self.optimizer, "momentum_buf", [])],
+                    sd.clone() if sd is not None else None,
+                )
             # warmup on a side stream (required before capture)
             s = torch.cuda.Stream()
             s.wait_stream(torch.cuda.current_stream())
@@ -188,6 +203,22 @@ class SyncTrainer:
                     self.optimizer.step(grad_scale=1.0)
                 else:
                     self._torch_opt.step()
+            # restore pre-warmup state (the capture pass records, it does
+            # not execute — only the 2 warmup steps mutated anything)
+            for b, p0 in zip(self.buckets.buckets, snap_params):
+                b.flat_param.copy_(p0)
+            if snap_opt is not None:
+                self.optimizer.step_count = snap_opt[0]
+                for t, t0 in zip(getattr(self.optimizer, "exp_avg", []), snap_opt[1]):
+                    t.copy_(t0)
+                for t, t0 in zip(getattr(self.optimizer, "exp_avg_sq", []), snap_opt[2]):
+                    t.copy_(t0)
+                for t, t0 in zip(getattr(self.optimizer, "momentum_buf", []), snap_opt[3]):
+                    if t is not None and t0 is not None:
+                        t.copy_(t0)
+                sd = getattr(self.optimizer, "step_dev", None)
+                if sd is not None and snap_opt[4] is not None:
+                    sd.copy_(snap_opt[4])
             self._graph = g
             self._graph_inputs = (key, static_x, static_y, static_loss)
         _, static_x, static_y, static_loss = self._graph_inputs
