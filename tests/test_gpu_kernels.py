@@ -169,7 +169,8 @@ def test_fused_adam_matches_torch_gpu():
         opt.step()
         bc1 = 1 - 0.9**t
         bc2 = 1 - 0.999**t
-        ops.ext().fused_adam(p_hip, g, m, v, 0.01, 0.9, 0.999, 1e-8, 0.0, bc1, bc2, 1.0, False)
+        ops.ext().fused_adam(p_hip, g, m, v, 0.01, 0.9, 0.999, 1e-8, 0.0, bc1, bc2, 1.0, False,
+                             None)
     assert (p_hip - p_ref).abs().max().item() < 1e-5
 
 
