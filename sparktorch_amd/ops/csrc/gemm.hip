@@ -121,7 +121,6 @@ __device__ __forceinline__ void stage_tile(const void* __restrict__ src, bf16raw
     int gk = kt + k;
     if (gk < kmax) {
       const char* base = (const char*)src + (int64_t)gk * skol * (SRC_F32 ? 4 : 2);
-      int rrem = rmax - row0 - r0;
       int mem_rem = mem_rows - row0 - r0;
       if (mem_rem >= 16) {
         if (SRC_F32) {

@@ -211,9 +211,6 @@ __global__ void bn_stats_finalize2_kernel(const float* __restrict__ psum,
   }
 }
 
-static inline bool bn_vec_ok(int C) {
-  return C >= 8 && C <= 2048 && (C & 7) == 0 && (256 % (C >> 3)) == 0;
-}
 
 extern "C" hipError_t launch_bn_stats_nhwc(const bf16raw* x, float* sum, float* sumsq,
                                            float* mean, float* invstd, float* running_mean,
