@@ -288,3 +288,29 @@ def test_resnet18_fused_train_step_loss_decreases():
     torch.cuda.synchronize()
     assert all(l == l for l in losses), losses  # no NaN
     assert min(losses[4:]) < losses[0], losses
+
+
+def test_converted_generic_resnet_runs_native():
+    """convert_model_for_mi355x on a plain ResNet-18 swaps Conv2d/BatchNorm2d/
+    MaxPool2d for the NCHW-layout Hip modules (functional F.relu stays eager);
+    one fwd+bwd must run with native kernels and finite grads."""
+    from sparktorch_amd.models.resnet import ResNet18
+    from sparktorch_amd.ops.modules import (
+        HipBatchNorm2d,
+        HipConv2d,
+        HipMaxPool2d,
+        convert_model_for_mi355x,
+    )
+
+    torch.manual_seed(8)
+    model = convert_model_for_mi355x(ResNet18(num_classes=8)).to(DEV)
+    kinds = {type(m).__name__ for m in model.modules()}
+    assert {"HipConv2d", "HipBatchNorm2d", "HipMaxPool2d"} <= kinds
+
+    x = torch.randn(4, 3, 224, 224, device=DEV)
+    out = model(bf(x))
+    loss = F.cross_entropy(out.float(), torch.randint(0, 8, (4,), device=DEV))
+    loss.backward()
+    assert torch.isfinite(loss)
+    for p in model.parameters():
+        assert p.grad is None or torch.isfinite(p.grad).all()
