@@ -2,6 +2,12 @@
 push gradients over the binary-wire HTTP PS.  Mirrors the reference's
 mode='hogwild' path (README.md hogwild example)."""
 
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
 import numpy as np
 import torch
 import torch.nn as nn

@@ -2,6 +2,12 @@
 instantiated there (avoids driver memory for big nets).  Mirrors reference
 examples/lazy_load_cnn.py."""
 
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
 import numpy as np
 import torch
 import torch.nn as nn

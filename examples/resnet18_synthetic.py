@@ -9,6 +9,12 @@ works anywhere.  ``python bench.py --model resnet18`` is the measured,
 timed version of this workload.
 """
 
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
 import numpy as np
 import torch
 import torch.nn as nn

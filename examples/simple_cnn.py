@@ -2,6 +2,12 @@
 cnn_network.py).  On GPU the convs run on the hand-written implicit-GEMM MFMA
 path via the module converter."""
 
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
 import numpy as np
 import torch
 import torch.nn as nn

@@ -6,6 +6,12 @@ DataFrame/Pipeline on a cluster — the estimator code is identical).
 Run: python examples/simple_dnn.py [--device cuda:0]
 """
 
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
 import argparse
 import tempfile
 
