@@ -377,7 +377,7 @@ __device__ __forceinline__ void stage_swz_fallback(const bf16raw* __restrict__ s
 // register path otherwise (wgrad, where the reduction runs over the outer
 // stride of both operands).  B (small, L2-hot, possibly fp32 master
 // weights) always uses the register path.  K-loop is double-buffered.
-template <bool B_IS_F32, int EPI, bool SPLITK, int WR, int WC, bool AG>
+template <bool B_IS_F32, int EPI, bool SPLITK, int WR, int WC, bool AG, bool BG = false>
 __global__ __launch_bounds__(WR * WC * 64, 2) void gemm_kernel(const void* __restrict__ Ap,
                                                    const void* __restrict__ Bp,
                                                    float* __restrict__ Cf,
@@ -395,7 +395,7 @@ __global__ __launch_bounds__(WR * WC * 64, 2) void gemm_kernel(const void* __res
   constexpr int BROWS = BNt < 128 ? 128 : BNt;  // stage_tile writes 128 LDS rows
   constexpr int NBUF = AG ? 2 : 1;  // T14 register pipeline needs one buffer
   __shared__ bf16raw As[NBUF][AG ? (BMt * BKT) : (BMt * LP)];
-  __shared__ bf16raw Bs[NBUF][BROWS * LP];
+  __shared__ bf16raw Bs[NBUF][BG ? (BROWS * BKT) : (BROWS * LP)];
 
   // bijective XCD-aware swizzle of the flattened block id (guide §5.5 T1):
   // consecutive output tiles land on one XCD so shared operand rows stay in
@@ -435,9 +435,17 @@ __global__ __launch_bounds__(WR * WC * 64, 2) void gemm_kernel(const void* __res
     else                                                                                   \
       stage_swz_fallback<BMt, BKT>((const bf16raw*)Ap, As[bufi], m0, M, kt, k_end, sam);   \
     /* B staged TRANSPOSED: LDS row = n, col = k -> srow := sbn, skol := sbk */            \
-    _Pragma("unroll") for (int kh = 0; kh < BKT; kh += 32)                                 \
-        stage_tile<B_IS_F32, LP>(Bp, Bs[1 && (bufi)] + kh, n0, N, (kt) + kh, k_end, sbn,   \
-                                 sbk, ones_row);                                           \
+    if (BG) {                                                                              \
+      if (n0 + BROWS <= N && (kt) + BKT <= k_end)                                          \
+        stage_glds<BROWS, BKT>((const bf16raw*)Bp, Bs[1 && (bufi)], n0, sbn, kt);          \
+      else                                                                                 \
+        stage_swz_fallback<BROWS, BKT>((const bf16raw*)Bp, Bs[1 && (bufi)], n0, N, kt,     \
+                                       k_end, sbn);                                        \
+    } else {                                                                               \
+      _Pragma("unroll") for (int kh = 0; kh < BKT; kh += 32)                               \
+          stage_tile<B_IS_F32, LP>(Bp, Bs[1 && (bufi)] + kh, n0, N, (kt) + kh, k_end,      \
+                                   sbn, sbk, ones_row);                                    \
+    }                                                                                      \
   } while (0)
 
     int buf = 0;
@@ -459,7 +467,9 @@ __global__ __launch_bounds__(WR * WC * 64, 2) void gemm_kernel(const void* __res
 #pragma unroll
         for (int ni = 0; ni < 4; ++ni) {
           int rowb = wc * 64 + ni * 16 + l15;
-          b[ni] = *(const frag_t*)&Bs[buf][rowb * LP + sub * 32 + swz_col(rowb, kg * 8)];
+          int kq = kg + sub * 4;
+          b[ni] = BG ? *(const frag_t*)&Bs[buf][rowb * BKT + (((kq) ^ (rowb & (SLOTS - 1))) << 3)]
+                     : *(const frag_t*)&Bs[buf][rowb * LP + sub * 32 + swz_col(rowb, kg * 8)];
         }
 #pragma unroll
         for (int mi = 0; mi < 4; ++mi)
@@ -665,17 +675,27 @@ extern "C" hipError_t launch_gemm_bf16(const void* A, const void* B, int b_is_f3
   if (wide) block = dim3(512);
 
   const bool ag = ag0;
+  // B by DMA too when it is a bf16 k-contiguous operand (conv fwd with the
+  // pre-cast w2d) — the guide's step-3 ladder result (517->874 TF) stages
+  // BOTH operands by glds
+  const bool bg = (!b_is_f32) && (sbk == 1) && (sbn % 8 == 0) && ones_row < 0;
 
 #define DISPATCH(BF32, EPIC, SPK)                                                            \
   do {                                                                                       \
     if (wide)                                                                                \
       gemm_kernel<BF32, EPIC, SPK, 2, 4, false><<<grid, block, 0, stream>>>(                 \
           A, B, Cf, Cb, bias, M, N, K, sam, sak, sbk, sbn, kps, Db, ones_row);               \
+    else if (narrow && ag && bg)                                                             \
+      gemm_kernel<BF32, EPIC, SPK, 4, 1, true, true><<<grid, block, 0, stream>>>(            \
+          A, B, Cf, Cb, bias, M, N, K, sam, sak, sbk, sbn, kps, Db, ones_row);               \
     else if (narrow && ag)                                                                   \
       gemm_kernel<BF32, EPIC, SPK, 4, 1, true><<<grid, block, 0, stream>>>(                  \
           A, B, Cf, Cb, bias, M, N, K, sam, sak, sbk, sbn, kps, Db, ones_row);               \
     else if (narrow)                                                                         \
       gemm_kernel<BF32, EPIC, SPK, 4, 1, false><<<grid, block, 0, stream>>>(                 \
+          A, B, Cf, Cb, bias, M, N, K, sam, sak, sbk, sbn, kps, Db, ones_row);               \
+    else if (ag && bg)                                                                       \
+      gemm_kernel<BF32, EPIC, SPK, 2, 2, true, true><<<grid, block, 0, stream>>>(            \
           A, B, Cf, Cb, bias, M, N, K, sam, sak, sbk, sbn, kps, Db, ones_row);               \
     else if (ag)                                                                             \
       gemm_kernel<BF32, EPIC, SPK, 2, 2, true><<<grid, block, 0, stream>>>(                  \

@@ -191,7 +191,10 @@ class _Conv2dNHWCFn(torch.autograd.Function):
         w2d = w.permute(0, 2, 3, 1).reshape(CO, K)
         if col.shape[1] != K:  # zero-pad weights to the padded K
             w2d = F.pad(w2d, (0, col.shape[1] - K))
-        w2d = w2d.contiguous()
+        # pre-cast the (small) weight view so BOTH GEMM operands are bf16 and
+        # DMA-staged; fp32->bf16 happened at the LDS stage before, so the
+        # compute numerics are unchanged
+        w2d = w2d.to(torch.bfloat16).contiguous()
         y2d = ext.linear_fwd(col, w2d, b, relu)  # [B*HO*WO, CO] == NHWC
         ctx.save_for_backward(col, w2d, y2d)
         ctx.meta = (B, CI, H, W, CO, KH, KW, sh, sw, ph, pw, HO, WO, relu, b is not None)
