@@ -35,8 +35,13 @@ def _choose_splitk(batch: int, n_out: int, k_in: int) -> int:
 class _LinearFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x: torch.Tensor, w: torch.Tensor, b: Optional[torch.Tensor], relu: bool):
-        y = ops.ext().linear_fwd(x, w, b, relu)
+        # one small cast per step so the GEMM stages a bf16 B by DMA instead
+        # of re-reading + converting the fp32 master per M-tile (the cast
+        # point is where the stage converted anyway — numerics unchanged)
+        wb = ops.ext().cast_f32_bf16(w) if w.dtype == torch.float32 else w
+        y = ops.ext().linear_fwd(x, wb, b, relu)
         ctx.save_for_backward(x, w, y)
+        ctx.wb = wb
         ctx.relu = relu
         ctx.has_bias = b is not None
         ctx.bias_ref = b
@@ -50,7 +55,7 @@ class _LinearFn(torch.autograd.Function):
         if dy.dtype != torch.bfloat16:
             dy = dy.to(torch.bfloat16)
         dz = ext.relu_bwd(dy, y) if ctx.relu else dy
-        dx = ext.linear_dgrad(dz, w) if ctx.needs_input_grad[0] else None
+        dx = ext.linear_dgrad(dz, ctx.wb) if ctx.needs_input_grad[0] else None
 
         # Direct-grad path: when FlatBuckets installed this param, its .grad
         # is a pre-zeroed flat-bucket view — accumulate straight into it (no
