@@ -243,15 +243,6 @@ static int small_wgrad_slices(int64_t M) {
   return (int)(s < 256 ? 256 : (s > 2048 ? 2048 : s));
 }
 
-
-// wgrad A-operand trick: one cheap transpose of dz ([rows, CO] -> [CO, rows])
-// turns the reduction-major operand k-contiguous, so the GEMM takes the
-// async-DMA (glds) staging path instead of the register pipeline.  Pays when
-// CO fills at least one 128-row M-tile.
-static bool wgrad_transpose_path(int64_t rows, int64_t N) {
-  return N >= 128 && (N & 7) == 0 && rows % 8 == 0 && rows >= 4096;
-}
-
 at::Tensor linear_wgrad(at::Tensor dz, at::Tensor x, int64_t splitk) {
   check_gpu_contig(dz, at::kBFloat16, "dz");
   check_gpu_contig(x, at::kBFloat16, "x");
@@ -268,14 +259,6 @@ at::Tensor linear_wgrad(at::Tensor dz, at::Tensor x, int64_t splitk) {
   }
   at::Tensor dw = splitk > 1 ? at::zeros({N, K}, x.options().dtype(at::kFloat))
                              : at::empty({N, K}, x.options().dtype(at::kFloat));
-  if (wgrad_transpose_path(B, N)) {
-    auto dzt = dz.t().contiguous();
-    CHECK_HIP(launch_gemm_bf16(dzt.data_ptr(), x.data_ptr(), 0, dw.data_ptr<float>(), nullptr,
-                               nullptr, (int)N, (int)K, (int)B,
-                               /*sam*/ B, /*sak*/ 1, /*sbk*/ K, /*sbn*/ 1, 0, (int)splitk,
-                               nullptr, -1, cur_stream()));
-    return dw;
-  }
   CHECK_HIP(launch_gemm_bf16(dz.data_ptr(), x.data_ptr(), /*b_is_f32*/ 0, dw.data_ptr<float>(),
                              nullptr, nullptr, (int)N, (int)K, (int)B,
                              /*sam*/ 1, /*sak*/ N, /*sbk*/ K, /*sbn*/ 1, 0, (int)splitk,
@@ -301,13 +284,6 @@ void linear_wgrad_into(at::Tensor dz, at::Tensor x, at::Tensor dw, int64_t split
     return;
   }
   if (splitk < 1) splitk = 1;
-  if (wgrad_transpose_path(B, N)) {
-    auto dzt = dz.t().contiguous();
-    CHECK_HIP(launch_gemm_bf16(dzt.data_ptr(), x.data_ptr(), 0, dw.data_ptr<float>(), nullptr,
-                               nullptr, (int)N, (int)K, (int)B, B, 1, K, 1, 0, -(int)splitk,
-                               nullptr, -1, cur_stream()));
-    return;
-  }
   CHECK_HIP(launch_gemm_bf16(dz.data_ptr(), x.data_ptr(), 0, dw.data_ptr<float>(), nullptr,
                              nullptr, (int)N, (int)K, (int)B, 1, N, K, 1, 0, -(int)splitk,
                              nullptr, -1, cur_stream()));
@@ -324,13 +300,6 @@ void linear_wgrad_bias_into(at::Tensor dz, at::Tensor x, at::Tensor dw, at::Tens
   int64_t B = dz.size(0), N = dz.size(1), K = x.size(1);
   TORCH_CHECK(x.size(0) == B && dw.size(0) == N && dw.size(1) == K && db.numel() == N);
   if (splitk < 1) splitk = 1;
-  if (wgrad_transpose_path(B, N)) {
-    auto dzt = dz.t().contiguous();
-    CHECK_HIP(launch_gemm_bf16(dzt.data_ptr(), x.data_ptr(), 0, dw.data_ptr<float>(), nullptr,
-                               nullptr, (int)N, (int)(K + 1), (int)B, B, 1, K, 1, 0,
-                               -(int)splitk, db.data_ptr<float>(), (int)K, cur_stream()));
-    return;
-  }
   CHECK_HIP(launch_gemm_bf16(dz.data_ptr(), x.data_ptr(), 0, dw.data_ptr<float>(), nullptr,
                              nullptr, (int)N, (int)(K + 1), (int)B, 1, N, K, 1, 0, -(int)splitk,
                              db.data_ptr<float>(), (int)K, cur_stream()));
