@@ -52,3 +52,56 @@ def test_state_dict_roundtrip():
 def test_bad_magic_raises():
     with pytest.raises(ValueError):
         decode_tensors(b"\x00" * 16)
+
+
+def test_server_error_budget_and_recovery():
+    """The PS tolerates up to 10 malformed updates (HTTP 200 'tolerated'),
+    surfaces 500 after, and keeps serving parameters throughout (reference
+    server.py:141-144)."""
+    import urllib.request
+
+    import torch
+    import torch.nn as nn
+
+    from sparktorch_amd.compat.local import free_port
+    from sparktorch_amd.models.simple_net import Net
+    from sparktorch_amd.parallel.server import Server
+    from sparktorch_amd.parallel.wire import decode_state_dict
+    from sparktorch_amd.utils.serialize import serialize_torch_obj
+
+    obj = serialize_torch_obj(Net(), nn.MSELoss(), torch.optim.Adam, lr=0.01)
+    port = free_port()
+    srv = Server(obj, port=port, acquire_lock=True, early_stop_patience=-1, window_len=2)
+    srv.start_server()
+    try:
+        base = "http://127.0.0.1:%d" % port
+        # wait for the PS subprocess to bind
+        import time
+        for _ in range(100):
+            try:
+                with urllib.request.urlopen(base + "/", timeout=2):
+                    break
+            except Exception:
+                time.sleep(0.1)
+
+        def post_garbage():
+            req = urllib.request.Request(base + "/update", data=b"not-a-tensor-blob",
+                                         method="POST")
+            try:
+                with urllib.request.urlopen(req, timeout=10) as r:
+                    return r.status, r.read()
+            except urllib.error.HTTPError as e:
+                return e.code, e.read()
+
+        for i in range(10):
+            code, body = post_garbage()
+            assert code == 200 and body == b"tolerated", (i, code, body)
+        code, _ = post_garbage()
+        assert code == 500
+
+        # parameters still served after the budget blew
+        with urllib.request.urlopen(base + "/parameters", timeout=10) as r:
+            sd = decode_state_dict(r.read())
+        assert "fc1.weight" in sd or len(sd) > 0
+    finally:
+        srv.stop_server()
