@@ -160,13 +160,17 @@ def test_add_relu_fwd_bwd():
     assert torch.allclose(bb.grad.float(), b2.grad, atol=1e-2, rtol=1e-2)
 
 
-@pytest.mark.parametrize("CI,CO,K,s,p", [(64, 64, 3, 1, 1), (64, 128, 3, 2, 1), (3, 16, 7, 2, 3),
-                                         (128, 256, 1, 2, 0)])
-def test_conv2d_nhwc_fwd_bwd_vs_torch(CI, CO, K, s, p):
+@pytest.mark.parametrize("CI,CO,K,s,p,H,W", [
+    (64, 64, 3, 1, 1, 14, 14), (64, 128, 3, 2, 1, 14, 14), (3, 16, 7, 2, 3, 14, 14),
+    (128, 256, 1, 2, 0, 14, 14),
+    (64, 64, 3, 2, 1, 13, 15),   # odd spatial dims, strided
+    (8, 24, 5, 1, 2, 9, 11),     # small channels, odd dims, big pad
+])
+def test_conv2d_nhwc_fwd_bwd_vs_torch(CI, CO, K, s, p, H, W):
     from sparktorch_amd.ops.functional import hip_conv2d_nhwc
 
     torch.manual_seed(10)
-    B, H, W = 4, 14, 14
+    B = 4
     x = torch.randn(B, H, W, CI, device=DEV)
     w = torch.randn(CO, CI, K, K, device=DEV) * (1.0 / (CI * K * K) ** 0.5)
 
