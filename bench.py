@@ -61,7 +61,7 @@ def main() -> int:
 
     from sparktorch_amd.parallel.sync import SyncTrainer
 
-    torch.manual_seed(1234 + rank)
+    torch.manual_seed(1234)  # identical initial weights on every rank
     if args.model == "mnist_cnn":
         if on_gpu:
             from sparktorch_amd.ops.modules import MnistCNNFused
@@ -106,6 +106,7 @@ def main() -> int:
 
     in_dim = 3 * 224 * 224 if args.model == "resnet18" else 784
     n_classes = 1000 if args.model == "resnet18" else 10
+    torch.manual_seed(1234 + rank)  # rank-local data shard
     x = torch.randn(args.batch, in_dim, device=device)
     if on_gpu:
         x = x.to(torch.bfloat16)
