@@ -574,6 +574,9 @@ __global__ __launch_bounds__(WR * WC * 64, 2) void gemm_kernel(const void* __res
   const int m_base = m0 + wr * 64;
   const int n_base = n0 + wc * 64;
   if (EPI != EPI_F32 && (N & 7) == 0 && n_base + 64 <= N) {
+    // all waves must be past their final fragment reads before the scratch
+    // overwrites the staging images (the T14 loop has no trailing barrier)
+    __syncthreads();
     // per-wave [16][68] fp32 scratch inside the (now idle) LDS staging
     // buffers: As holds 4x(16x68x4B) for the 4-wave tiles; the 8-wave wide
     // tile needs 34.8 KB and uses Bs (36.9 KB there)
