@@ -34,7 +34,9 @@ def main() -> int:
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=30)
     ap.add_argument("--warmup", type=int, default=10)
-    ap.add_argument("--batch", type=int, default=131072, help="per-GPU batch (weak scaling)")
+    ap.add_argument("--batch", type=int, default=0,
+                    help="per-GPU batch (weak scaling); 0 = per-model default "
+                         "(mlp 2097152, cnn 131072, resnet18 2048 — measured sweet spots)")
     ap.add_argument("--device", type=str, default=None)
     ap.add_argument("--hipgraph", action="store_true", help="capture the train step in a HIP graph")
     ap.add_argument("--model", type=str, default="mnist_mlp",
@@ -63,6 +65,8 @@ def main() -> int:
 
     torch.manual_seed(1234)  # identical initial weights on every rank
     if args.model == "mnist_cnn":
+        if args.batch == 0:
+            args.batch = 131072
         if on_gpu:
             from sparktorch_amd.ops.modules import MnistCNNFused
 
@@ -72,8 +76,8 @@ def main() -> int:
 
             model = MnistCNN()
     elif args.model == "resnet18":
-        if args.batch == 131072:  # default was sized for the MLP
-            args.batch = 256
+        if args.batch == 0:
+            args.batch = 2048
         if on_gpu:
             from sparktorch_amd.ops.modules import ResNet18Fused
 
@@ -83,10 +87,14 @@ def main() -> int:
 
             model = ResNet18()
     elif on_gpu:
+        if args.batch == 0:
+            args.batch = 2097152
         from sparktorch_amd.ops.modules import MnistMLPFused
 
         model = MnistMLPFused()
     else:
+        if args.batch == 0:
+            args.batch = 8192  # CPU-sized
         from sparktorch_amd.models.mnist import MnistMLP
 
         model = MnistMLP()
