@@ -52,10 +52,19 @@ def pick_device(device: str, rank: int) -> str:
     return "cpu"
 
 
-def select_backend(device: str, backend: Optional[str] = None) -> str:
+def select_backend(device: str, backend: Optional[str] = None, world_size: int = 1) -> str:
     if backend:
         return backend
-    return "nccl" if device.startswith("cuda") else "gloo"
+    if device.startswith("cuda"):
+        # RCCL wants exactly one rank per GPU ("Duplicate GPU detected"
+        # otherwise).  More ranks than visible GPUs happens on shared-GPU
+        # dev boxes / oversubscribed partitions: collectives fall back to
+        # gloo (which accepts CUDA tensors) while compute stays on the GPU.
+        n_gpu = torch.cuda.device_count() if torch.cuda.is_available() else 0
+        if world_size <= n_gpu:
+            return "nccl"
+        return "gloo"
+    return "gloo"
 
 
 def init_process_group_from_barrier(
@@ -86,7 +95,7 @@ def init_process_group_from_barrier(
     os.environ["WORLD_SIZE"] = str(world_size)
 
     dist.init_process_group(
-        select_backend(device, backend),
+        select_backend(device, backend, world_size),
         rank=rank,
         world_size=world_size,
         timeout=datetime.timedelta(seconds=timeout_s),
