@@ -43,8 +43,13 @@ def main() -> int:
                     choices=["mnist_mlp", "mnist_cnn", "resnet18"],
                     help="mnist_mlp = BASELINE flagship; mnist_cnn = examples/simple_cnn config; "
                          "resnet18 = BASELINE config 4 (synthetic 3x224x224)")
-    ap.add_argument("--mode", type=str, default="train", choices=["train", "infer"],
-                    help="infer = saved-pipeline batch inference (HIP-graph forward)")
+    ap.add_argument("--mode", type=str, default="train",
+                    choices=["train", "infer", "time_to_loss"],
+                    help="infer = saved-pipeline batch inference (HIP-graph forward); "
+                         "time_to_loss = seconds of training until --target-loss on a "
+                         "fixed synthetic batch (the BASELINE.json companion metric)")
+    ap.add_argument("--target-loss", type=float, default=1.0)
+    ap.add_argument("--max-steps", type=int, default=2000)
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -101,6 +106,8 @@ def main() -> int:
 
     if args.mode == "infer":
         return run_infer(args, model, device, on_gpu)
+    if args.mode == "time_to_loss":
+        return run_time_to_loss(args, model, device, on_gpu, world, rank)
 
     opt = torch.optim.Adam(model.parameters(), lr=1e-3)
     trainer = SyncTrainer(
@@ -175,6 +182,58 @@ def main() -> int:
         }
         print(json.dumps(out), flush=True)
 
+    if world > 1:
+        dist.destroy_process_group()
+    return 0
+
+
+def run_time_to_loss(args, model, device, on_gpu, world, rank) -> int:
+    """Train on one fixed synthetic batch until loss < target; report wall
+    seconds (strong-scaling companion to the throughput metric)."""
+    from sparktorch_amd.parallel.sync import SyncTrainer
+
+    opt = torch.optim.Adam(model.parameters(), lr=1e-3)
+    trainer = SyncTrainer(model, nn.CrossEntropyLoss(), opt, device=device, world_size=world)
+    in_dim = 3 * 224 * 224 if args.model == "resnet18" else 784
+    n_classes = 1000 if args.model == "resnet18" else 10
+    torch.manual_seed(1234 + rank)
+    x = torch.randn(args.batch, in_dim, device=device)
+    if on_gpu:
+        x = x.to(torch.bfloat16)
+    y = torch.randint(0, n_classes, (args.batch,), device=device)
+
+    trainer.train_step(x, y)  # warmup/compile outside the clock
+    if on_gpu:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    steps = 0
+    loss = float("inf")
+    while steps < args.max_steps:
+        loss = trainer.train_step(x, y)
+        steps += 1
+        if loss < args.target_loss:
+            break
+    if on_gpu:
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+    if rank == 0:
+        print(json.dumps({
+            "metric": "time_to_loss_s",
+            "value": elapsed,
+            "unit": "s",
+            "n_gpus": world if world > 1 else 1,
+            "steps": steps,
+            "warmup": 1,
+            "ms_per_step": elapsed / max(1, steps) * 1000.0,
+            "higher_is_better": False,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": "bf16" if on_gpu else "fp32",
+            "data": "synthetic",
+            "config": {"model": args.model, "global_batch": args.batch,
+                       "target_loss": args.target_loss, "reached": loss < args.target_loss,
+                       "final_loss": loss, "seq_len": None, "parallelism": "dp1"},
+        }), flush=True)
     if world > 1:
         dist.destroy_process_group()
     return 0
