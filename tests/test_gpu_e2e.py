@@ -117,3 +117,27 @@ def test_rccl_world1_init_allreduce_and_train():
         assert l1 == l1 and l1 <= l0 * 1.5
     finally:
         dist.destroy_process_group()
+
+
+def test_estimator_resnet_fit_transform_on_gpu():
+    """Full-stack GPU fit of a (tiny-classes) ResNet-18 through the
+    estimator: dill serialization of the 11M-param net, converter swap to
+    the native NHWC-capable modules, sync training, batched transform."""
+    import numpy as np
+
+    from sparktorch_amd.models.resnet import ResNet18
+
+    rng = np.random.RandomState(7)
+    feats = rng.standard_normal((24, 3 * 224 * 224)).astype(np.float64) * 0.1
+    labels = list(rng.randint(0, 4, 24).astype(np.float64))
+    df = LocalDataFrame.from_arrays(feats, labels, num_partitions=1)
+
+    obj = serialize_torch_obj(ResNet18(num_classes=4), nn.CrossEntropyLoss(),
+                              torch.optim.Adam, lr=1e-3)
+    model = SparkTorch(
+        inputCol="features", labelCol="label", predictionCol="predicted",
+        torchObj=obj, iters=2, miniBatch=8, device="cuda:0", mode="synchronous",
+    ).fit(df)
+    out = model.transform(df).collect()
+    assert len(out) == 24
+    assert all(r["predicted"] in (0.0, 1.0, 2.0, 3.0) for r in out)
