@@ -124,6 +124,52 @@ __global__ void im2col_nhwc_rowcopy_kernel(const bf16raw* __restrict__ x,
   }
 }
 
+// LDS-row-staged variant for small CI (the 3-channel stem): one thread owns
+// one full output row — scalar gathers land in an LDS scratch row, the
+// global write is Kp/8 contiguous shortx8 stores, and adjacent threads
+// (adjacent wo) overlap their stride-2 source windows in L1.
+__global__ void im2col_nhwc_rowstage_kernel(const bf16raw* __restrict__ x,
+                                            bf16raw* __restrict__ col, int B, int CI, int H,
+                                            int W, int KH, int KW, int HO, int WO, int sh,
+                                            int sw, int ph, int pw, int Kp) {
+  extern __shared__ bf16raw rowbuf[];  // [256][Kp]
+  bf16raw* my = rowbuf + threadIdx.x * Kp;
+  const int KWCI = KW * CI;
+  const int K = KH * KWCI;
+  int64_t total_m = (int64_t)B * HO * WO;
+  for (int64_t m = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; m < total_m;
+       m += (int64_t)gridDim.x * blockDim.x) {
+    int wo = (int)(m % WO);
+    int64_t t = m / WO;
+    int ho = (int)(t % HO);
+    int b = (int)(t / HO);
+    int w0 = wo * sw - pw;
+    for (int kh = 0; kh < KH; ++kh) {
+      int h = ho * sh - ph + kh;
+      bf16raw* dst = my + kh * KWCI;
+      if (h < 0 || h >= H) {
+        for (int j = 0; j < KWCI; ++j) dst[j] = 0;
+        continue;
+      }
+      const bf16raw* src = x + (((int64_t)b * H + h) * W + w0) * CI;
+      if (w0 >= 0 && w0 + KW <= W) {
+        for (int j = 0; j < KWCI; ++j) dst[j] = src[j];
+      } else {
+        for (int kw = 0; kw < KW; ++kw) {
+          int w = w0 + kw;
+          for (int ci = 0; ci < CI; ++ci)
+            dst[kw * CI + ci] = (w >= 0 && w < W) ? src[kw * CI + ci] : (bf16raw)0;
+        }
+      }
+    }
+    for (int j = K; j < Kp; ++j) my[j] = 0;
+    // own-thread LDS readback: vector store the assembled row
+    shortx8* gout = (shortx8*)(col + m * Kp);
+    const shortx8* lin = (const shortx8*)my;
+    for (int v = 0; v < (Kp >> 3); ++v) gout[v] = lin[v];
+  }
+}
+
 extern "C" hipError_t launch_im2col_nhwc(const bf16raw* x, bf16raw* col, int B, int CI, int H,
                                          int W, int KH, int KW, int HO, int WO, int sh, int sw,
                                          int ph, int pw, int Kp, hipStream_t stream) {
@@ -131,6 +177,11 @@ extern "C" hipError_t launch_im2col_nhwc(const bf16raw* x, bf16raw* col, int B, 
     int64_t total = (int64_t)B * HO * WO * KH * KW * (CI >> 3);
     im2col_nhwc_vec_kernel<<<cgrid8(total, 256), 256, 0, stream>>>(x, col, B, CI, H, W, KH, KW,
                                                                    HO, WO, sh, sw, ph, pw, Kp);
+  } else if (Kp <= 160) {  // 256 threads x Kp x 2B LDS <= 80 KB -> 2 blocks/CU
+    int64_t total_m = (int64_t)B * HO * WO;
+    size_t lds = (size_t)256 * Kp * 2;
+    im2col_nhwc_rowstage_kernel<<<cgrid8(total_m, 256), 256, lds, stream>>>(
+        x, col, B, CI, H, W, KH, KW, HO, WO, sh, sw, ph, pw, Kp);
   } else {
     int64_t total = (int64_t)B * HO * WO * KH;
     im2col_nhwc_rowcopy_kernel<<<cgrid8(total, 256), 256, 0, stream>>>(x, col, B, CI, H, W, KH,
