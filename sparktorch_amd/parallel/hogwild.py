@@ -102,8 +102,10 @@ def handle_model(
     y_val = feats.y_val.to(device) if feats.y_val is not None else (x_val if x_val is not None else None)
 
     pinned: Optional[dict] = None
+    side_stream = None
     if device.startswith("cuda"):
         pinned = {k: torch.empty_like(v, device="cpu").pin_memory() for k, v in model.state_dict().items()}
+        side_stream = torch.cuda.Stream()  # hipMemcpyAsync H2D off the compute stream
 
     n = x_train.shape[0]
     for i in range(iters):
@@ -111,7 +113,10 @@ def handle_model(
         if pinned is not None:
             for k, v in sd.items():
                 pinned[k].copy_(v)
-            sd = {k: v.to(device, non_blocking=True) for k, v in pinned.items()}
+            # async H2D on a side stream; the compute stream waits on it
+            with torch.cuda.stream(side_stream):
+                sd = {k: v.to(device, non_blocking=True) for k, v in pinned.items()}
+            torch.cuda.current_stream().wait_stream(side_stream)
         model.load_state_dict(sd)
 
         if 0 < mini_batch < n:
