@@ -74,3 +74,36 @@ def test_lazy_no_params():
     s = serialize_torch_obj_lazy(Net, nn.MSELoss, torch.optim.Adam, optimizer_params={"lr": 0.1})
     loaded = load_torch_model(s, from_json=True)
     assert isinstance(loaded.model, Net)
+
+
+def test_fuzz_serialize_roundtrip_random_models():
+    """Random small architectures through the full envelope: serialize ->
+    load -> state_dict equality -> pipeline stopwords codec round-trip."""
+    import random
+
+    import torch
+    import torch.nn as nn
+
+    from sparktorch_amd.utils.codec import obj_to_stopwords, stopwords_to_obj
+    from sparktorch_amd.utils.serialize import load_torch_model, serialize_torch_obj
+
+    for seed in range(5):
+        rng = random.Random(seed)
+        torch.manual_seed(seed)
+        dims = [rng.randint(2, 17) for _ in range(rng.randint(2, 4))]
+        layers = []
+        for a, b in zip(dims, dims[1:]):
+            layers.append(nn.Linear(a, b))
+            if rng.random() < 0.5:
+                layers.append(nn.ReLU())
+        net = nn.Sequential(*layers)
+        crit = rng.choice([nn.MSELoss(), nn.CrossEntropyLoss()])
+        obj = serialize_torch_obj(net, crit, torch.optim.Adam, lr=10 ** -rng.randint(2, 4))
+        loaded = load_torch_model(obj, from_json=True)
+        sd0, sd1 = net.state_dict(), loaded.model.state_dict()
+        assert sd0.keys() == sd1.keys()
+        for k in sd0:
+            assert torch.equal(sd0[k], sd1[k]), (seed, k)
+        # checkpoint carrier codec round-trip of the serialized object
+        words = obj_to_stopwords(obj)
+        assert stopwords_to_obj(words) == obj
