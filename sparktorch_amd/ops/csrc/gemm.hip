@@ -1,25 +1,42 @@
-// Hand-written MFMA GEMM for gfx950 (CDNA4) — the hot op behind Linear
-// forward/backward (replaces the reference's eager model(x)/loss.backward()
-// CPU path, reference distributed.py:150,176).
+// Hand-written MFMA GEMM for gfx950 (CDNA4) — the hot op behind every
+// Linear/conv forward and backward (replaces the reference's eager
+// model(x)/loss.backward() CPU path, reference distributed.py:150,176).
 //
-// C[M,N] = A[M,K] x B[K,N]; A is bf16, B is bf16 or fp32 (weights are fp32
-// master copies in the flat buckets and are cast to bf16 during the LDS
-// stage); fp32 MFMA accumulation (v_mfma_f32_16x16x32_bf16).
+// C[M,N] = A[M,K] x B[K,N]; bf16 operands (fp32 master weights are cast at
+// the stage or pre-cast by the caller), fp32 MFMA accumulation
+// (v_mfma_f32_16x16x32_bf16, accumulators in AGPRs).
 //
-// Geometry: 128x128 block tile, BK=32, 256 threads = 4 waves in 2x2, each
-// wave owns a 64x64 sub-tile = 4x4 fragments of 16x16x32 MFMA; accumulators
-// live in AGPRs (4 fp32 per lane per fragment).  LDS holds A as [BM][BK] and
-// B TRANSPOSED as [BN][BK] (+8 bf16 pad per row, preserving the 16-byte
-// alignment of every ds_read_b128 fragment read — guide §2/§6 G4/G17), so
-// both fragment loads are contiguous 16-byte LDS reads.
+// Tile variants (wave always owns a 64x64 sub-tile = 4x4 fragments):
+//   <2,2>  128x128, BK=64 — the default
+//   <4,1>  256x64,  BK=32 — narrow-N shapes (64-ch convs, small heads)
+//   <2,4>  128x256, BK=64, 512 threads — big-N register-staged wgrad
 //
-// One GEMM serves all three Linear gradients via strides:
-//   fwd   Y = X  W^T : A=X(row),   B=W^T (sbk=1,   sbn=K)
-//   dX    = dZ W     : A=dZ(row),  B=W   (sbk=Kin, sbn=1)
+// Staging (the performance-critical part; every choice here is measured —
+// see profiles/bench_resnet18_optimization_log.md):
+//   * k-contiguous operands (AG/BG): async global_load_lds DMA into a
+//     pad-free LDS image whose 16B k-slots are XOR-swizzled by row (lane-
+//     linear for the DMA, bank-spread for ds_read_b128 fragment reads);
+//     boundary slabs take a vectorized swizzled fallback.
+//   * transposing operands (wgrad — the reduction runs over the outer
+//     stride of both matrices): T14 register pipeline — hold the slab in
+//     registers, write LDS after the barrier, issue the next slab's loads
+//     so they span the MFMA section.  The padded LDS image carries a
+//     granule-XOR swizzle (a 16-row staging stride hits one bank group
+//     whatever the row pad).
+//   * double-buffered K-loop in the DMA path, single buffer + two barriers
+//     in the register path.
+//
+// One kernel serves all three Linear/conv gradients via strides:
+//   fwd   Y = X  W^T : A=X(row),   B=W^T (sbk=1,   sbn=K)    [A glds, B glds if bf16]
+//   dX    = dZ W     : A=dZ(row),  B=W   (sbk=Kin, sbn=1)    [A glds]
 //   dW    = dZ^T X   : A=dZ^T (sam=1, sak=N), B=X (sbk=Kin, sbn=1), split-K
-//           over the batch dim with fp32 atomicAdd combine (fills 256 CUs
-//           even for small weight shapes).
-// Epilogues fuse bias add + ReLU into the C-write (EPI codes below).
+//           over the batch dim with fp32 atomicAdd combine   [T14 registers]
+//
+// Epilogues: bias+ReLU fused into the C-write; bf16 outputs go through a
+// per-wave LDS transpose so every lane stores one contiguous 16 B octet;
+// the fused dW+db trick treats bias as a virtual all-ones B column
+// (ones_row).  The XCD-aware block swizzle keeps an output tile's operand
+// rows inside one XCD's L2.
 
 #include "common.h"
 
