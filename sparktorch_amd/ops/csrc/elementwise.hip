@@ -5,7 +5,11 @@
 //    (reference distributed.py:198, server.py:139): ONE grid-stride float4
 //    kernel per flat bucket, 1/world grad averaging folded in.
 //  * relu_bwd — the dY*(Y>0) mask for the fused linear backward.
-//  * bias_grad — column sum of dZ (bf16 -> fp32).
+//  * bias_grad — column sum of dZ (bf16 -> fp32); small-N variant with
+//    8-column thread groups + scratch partials (no atomic chains).
+//  * small_wgrad — batched outer product for tiny dW over huge reductions
+//    (a 16x32 dW in a 128-row MFMA tile is ~98% dead).
+//  * increment_i32 — device step counter for hipGraph-capturable Adam.
 //  * cast_f64_f32 — Spark DenseVector rows arrive float64; pack/cast on
 //    device (reference does np.stack + .float() on CPU, util.py:87-99).
 //

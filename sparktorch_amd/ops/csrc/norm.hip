@@ -7,7 +7,14 @@
 // activations, fp32 statistics/params, LDS wave reductions, grids split
 // over (channel x batch-slice) so small-C layers still cover all 256 CUs.
 //
-// All NCHW contiguous.
+// Two layout families share the file: the original NCHW kernels (used by
+// the generic model converter) and the NHWC (channels-last) kernels the
+// fused ResNet path runs on.  NHWC reductions treat x as [M, C] with
+// coalesced 8-channel shortx8 groups per thread; BN reductions write block
+// partials to scratch and tree-finalize (same-address atomicAdd was
+// measured at ~200 ns/round — tools/bn_bench.hip); apply/dx cache
+// per-thread scale/shift (6.6 TB/s measured) and recompute the fused-ReLU
+// mask from x instead of reading the saved output.
 
 #include "common.h"
 
