@@ -46,3 +46,27 @@ def test_local_barrier_allgather_two_partitions():
     assert len(out) == 2
     for _idx, gathered in out:
         assert gathered == ["0", "1"]
+
+
+def test_sync_engine_empty_partition_raises_clearly():
+    """A partition with no rows must fail with the actionable repartition
+    message, not a shape error deep in torch (reference leaves this as an
+    obscure crash)."""
+    import numpy as np
+    import pytest
+    import torch
+    import torch.nn as nn
+
+    from sparktorch_amd.compat.local import LocalDataFrame
+    from sparktorch_amd.models.simple_net import Net
+    from sparktorch_amd.parallel.sync import train_distributed
+    from sparktorch_amd.utils.data import handle_data
+    from sparktorch_amd.utils.serialize import serialize_torch_obj
+
+    rng = np.random.RandomState(0)
+    # 3 partitions but only 2 rows -> one partition is empty
+    df = LocalDataFrame.from_arrays(rng.rand(2, 10), [0.0, 1.0], num_partitions=3)
+    obj = serialize_torch_obj(Net(), nn.MSELoss(), torch.optim.Adam, lr=0.01)
+    rdd = df.rdd.mapPartitions(handle_data("features", "label"))
+    with pytest.raises(Exception, match="repartition|empty"):
+        train_distributed(rdd, obj, iters=1, device="cpu", backend="gloo")
