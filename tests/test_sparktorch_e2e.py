@@ -262,3 +262,24 @@ def test_use_barrier_and_explicit_cpu_device(data_df):
         device="cpu",
     ).fit(data_df)
     assert model2.transform(data_df).count() == 400
+
+
+def test_partition_shuffles_hogwild_and_sync(data_df):
+    """partitionShuffles > 1 re-randomizes partitions between rounds
+    (reference hogwild.py:161-177); the sync engine honors it too (the
+    reference hardcoded 1 there — torch_distributed.py:309, knowing fix)."""
+    from sparktorch_amd.compat.local import free_port
+
+    obj = serialize_torch_obj(Net(), nn.MSELoss(), torch.optim.Adam, lr=0.01)
+    m1 = SparkTorch(
+        inputCol="features", labelCol="label", predictionCol="predicted",
+        torchObj=obj, iters=2, partitionShuffles=2, mode="hogwild",
+        port=free_port(), device="cpu",
+    ).fit(data_df)
+    assert m1.transform(data_df).count() == 400
+
+    m2 = SparkTorch(
+        inputCol="features", labelCol="label", predictionCol="predicted",
+        torchObj=obj, iters=2, partitionShuffles=2, mode="synchronous", device="cpu",
+    ).fit(data_df)
+    assert m2.transform(data_df).count() == 400
