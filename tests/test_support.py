@@ -146,3 +146,24 @@ def test_step_metrics_and_trace_range():
     with trace_range("cpu-noop"):  # no-op without a GPU
         x = 1 + 1
     assert x == 2
+
+
+def test_converter_recurses_nested_modules():
+    import torch.nn as nn
+
+    from sparktorch_amd.ops.modules import HipConv2d, HipLinear, convert_model_for_mi355x
+
+    model = nn.Sequential(
+        nn.Sequential(nn.Linear(8, 8), nn.ReLU(), nn.Sequential(nn.Linear(8, 4))),
+        nn.Conv2d(3, 8, 3),
+    )
+    out = convert_model_for_mi355x(model)
+    kinds = [type(m).__name__ for m in out.modules()]
+    assert kinds.count("HipLinear") == 2
+    assert kinds.count("HipConv2d") == 1
+    assert "Linear" not in kinds and "Conv2d" not in kinds
+    # adopted parameters are the SAME objects (buckets/optimizers unaffected)
+    import torch
+
+    x = torch.randn(2, 8)
+    assert out[0](x).shape == (2, 4)
