@@ -26,6 +26,7 @@ import torch
 from sparktorch_amd.parallel.sync import compute_loss
 from sparktorch_amd.parallel.wire import decode_state_dict, encode_tensors
 from sparktorch_amd.utils.data import handle_features
+from sparktorch_amd.utils.trace import StepMetrics, trace_range
 from sparktorch_amd.utils.serialize import load_torch_model
 
 
@@ -108,7 +109,9 @@ def handle_model(
         side_stream = torch.cuda.Stream()  # hipMemcpyAsync H2D off the compute stream
 
     n = x_train.shape[0]
+    metrics = StepMetrics()
     for i in range(iters):
+        metrics.start()
         sd = get_state_dict(master_url)
         if pinned is not None:
             for k, v in sd.items():
@@ -134,7 +137,9 @@ def handle_model(
             (p.grad if p.grad is not None else torch.zeros_like(p)).detach()
             for p in model.parameters()
         ]
-        put_deltas_to_server(master_url, grads)
+        with trace_range("hogwild_push"):
+            put_deltas_to_server(master_url, grads)
+        metrics.stop(float(loss))
 
         loss_for_es = loss
         if x_val is not None:
@@ -150,6 +155,9 @@ def handle_model(
 
         if verbose:
             print("hogwild iter %d loss %.6f" % (i, float(loss)), flush=True)
+
+    if verbose:
+        print("hogwild metrics %r" % metrics, flush=True)
 
     return iter([])
 
