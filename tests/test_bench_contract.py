@@ -1,0 +1,51 @@
+"""Guard the bench.py driver contract: one JSON line on stdout with the
+agreed keys, runnable with no flags on CPU, all three modes intact."""
+
+import json
+import os
+import subprocess
+import sys
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+REQUIRED = [
+    "metric", "value", "unit", "n_gpus", "steps", "warmup", "ms_per_step",
+    "higher_is_better", "scaling", "vs_baseline", "dtype", "data", "config",
+]
+
+
+def _run(*args):
+    out = subprocess.run(
+        [sys.executable, os.path.join(REPO, "bench.py"), *args],
+        capture_output=True, text=True, timeout=600, cwd=REPO,
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = out.stdout.strip().splitlines()[-1]
+    return json.loads(line)
+
+
+def test_train_contract_cpu():
+    d = _run("--steps", "2", "--warmup", "1", "--batch", "256")
+    for k in REQUIRED:
+        assert k in d, k
+    assert d["metric"] == "samples_per_sec"
+    assert d["higher_is_better"] is True
+    assert d["scaling"] == "weak"
+    assert d["config"]["global_batch"] == 256
+    assert d["value"] > 0 and d["ms_per_step"] > 0
+    assert d["config"]["model"].startswith("mnist_mlp")
+
+
+def test_infer_contract_cpu():
+    d = _run("--mode", "infer", "--steps", "2", "--warmup", "1", "--batch", "128")
+    assert d["metric"] == "inference_samples_per_sec"
+    assert d["value"] > 0
+
+
+def test_time_to_loss_contract_cpu():
+    d = _run("--mode", "time_to_loss", "--steps", "2", "--warmup", "1",
+             "--batch", "256", "--target-loss", "5.0", "--max-steps", "3")
+    assert d["metric"] == "time_to_loss_s"
+    assert d["higher_is_better"] is False
+    assert d["scaling"] == "strong"
+    assert "reached" in d["config"]
