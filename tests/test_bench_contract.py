@@ -49,3 +49,17 @@ def test_time_to_loss_contract_cpu():
     assert d["higher_is_better"] is False
     assert d["scaling"] == "strong"
     assert "reached" in d["config"]
+
+
+def test_graft_entry_contract():
+    """__graft_entry__ must expose build() and smoke() (driver contract);
+    build() must be idempotent-cheap when the extension is current."""
+    sys.path.insert(0, REPO)
+    try:
+        import __graft_entry__ as g
+    finally:
+        sys.path.pop(0)
+    assert callable(g.build) and callable(g.smoke)
+    g.build()  # no-op rebuild when sources unchanged; must not raise on CPU
+    from sparktorch_amd import ops
+    assert ops.available()
