@@ -107,3 +107,32 @@ def test_fuzz_serialize_roundtrip_random_models():
         # checkpoint carrier codec round-trip of the serialized object
         words = obj_to_stopwords(obj)
         assert stopwords_to_obj(words) == obj
+
+
+def test_nested_pipeline_unwrap():
+    """Carrier stages inside a nested pipeline are unwrapped recursively
+    (reference pipeline_util.py:66-67)."""
+    import torch
+    import torch.nn as nn
+
+    from sparktorch_amd import (
+        LocalPipeline,
+        PysparkPipelineWrapper,
+        SparkTorchModel,
+        create_spark_torch_model,
+    )
+    from sparktorch_amd.pipeline_util import _CarrierStage
+    from sparktorch_amd.utils.codec import obj_to_stopwords
+
+    net = nn.Linear(4, 2)
+    m = create_spark_torch_model(net, inputCol="f", predictionCol="p")
+    carrier = _CarrierStage(obj_to_stopwords(m))
+    inner = LocalPipeline(stages=[carrier])
+    outer = LocalPipeline(stages=[inner, _CarrierStage(obj_to_stopwords(m))])
+
+    unwrapped = PysparkPipelineWrapper.unwrap(outer)
+    assert isinstance(unwrapped.stages[0].stages[0], SparkTorchModel)
+    assert isinstance(unwrapped.stages[1], SparkTorchModel)
+    sd0 = net.state_dict()
+    sd1 = unwrapped.stages[1].getPytorchModel().state_dict()
+    assert all(torch.equal(sd0[k], sd1[k]) for k in sd0)
