@@ -69,9 +69,15 @@ class LocalRDD:
         return len(self._partitions)
 
     def repartition(self, n: int) -> "LocalRDD":
+        import random as _random
+
         flat: list = []
         for i, part in enumerate(self._partitions):
             flat.extend(_apply_chain(i, part, self._chain))
+        # Spark's repartition is a randomizing shuffle — re-randomize row
+        # placement so partitionShuffles>1 actually re-mixes data
+        # (reference hogwild.py:175-177 relies on this)
+        _random.Random().shuffle(flat)
         parts: List[list] = [[] for _ in range(n)]
         for i, row in enumerate(flat):
             parts[i % n].append(row)

@@ -139,7 +139,7 @@ def handle_model(
         ]
         with trace_range("hogwild_push"):
             put_deltas_to_server(master_url, grads)
-        metrics.stop(float(loss))
+        metrics.stop(float(loss.detach()))
 
         loss_for_es = loss
         if x_val is not None:
@@ -193,7 +193,11 @@ def train(
 
             rdd.mapPartitions(worker).foreach(lambda _x: None)
             if shuffle_round + 1 < partition_shuffles:
-                rdd = rdd.repartition(rdd.getNumPartitions())
+                # a barrier-wrapped rdd (pyspark RDDBarrier) exposes only
+                # mapPartitions*; shuffle the underlying RDD and re-wrap
+                base = getattr(rdd, "rdd", rdd)
+                shuffled = base.repartition(base.getNumPartitions())
+                rdd = shuffled.barrier() if base is not rdd else shuffled
 
         return get_state_dict(master_url)
     finally:

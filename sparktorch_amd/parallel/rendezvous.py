@@ -52,17 +52,37 @@ def pick_device(device: str, rank: int) -> str:
     return "cpu"
 
 
-def select_backend(device: str, backend: Optional[str] = None, world_size: int = 1) -> str:
+def select_backend(
+    device: str,
+    backend: Optional[str] = None,
+    world_size: int = 1,
+    local_ranks: Optional[int] = None,
+) -> str:
+    """Pick the collective backend.
+
+    RCCL wants exactly one rank per GPU per host ("Duplicate GPU detected"
+    otherwise), so the comparison is ranks-on-THIS-host vs visible GPUs —
+    not global world size, which would wrongly demote any multi-node run
+    (e.g. 2x8 ranks on 8-GPU hosts) to gloo.  ``local_ranks`` comes from
+    LOCAL_WORLD_SIZE or from counting barrier-task addresses sharing this
+    host; absent both, assume the whole world is local (single-node).
+    """
     if backend:
         return backend
     if device.startswith("cuda"):
-        # RCCL wants exactly one rank per GPU ("Duplicate GPU detected"
-        # otherwise).  More ranks than visible GPUs happens on shared-GPU
-        # dev boxes / oversubscribed partitions: collectives fall back to
-        # gloo (which accepts CUDA tensors) while compute stays on the GPU.
         n_gpu = torch.cuda.device_count() if torch.cuda.is_available() else 0
-        if world_size <= n_gpu:
+        if local_ranks is None:
+            env = os.environ.get("LOCAL_WORLD_SIZE")
+            local_ranks = int(env) if env else world_size
+        if local_ranks <= n_gpu:
             return "nccl"
+        import warnings
+
+        warnings.warn(
+            "falling back to gloo collectives: %d ranks share this host's %d "
+            "visible GPU(s); give each rank its own GPU for RCCL over xGMI"
+            % (local_ranks, n_gpu)
+        )
         return "gloo"
     return "gloo"
 
@@ -89,13 +109,18 @@ def init_process_group_from_barrier(
     if master_host in ("", "0.0.0.0", "localhost"):
         master_host = "127.0.0.1"
 
+    # ranks co-located on this host = tasks whose executor address shares
+    # our host (drives the nccl-vs-gloo choice in select_backend)
+    hosts = [i.address.split(":")[0] for i in infos]
+    local_ranks = hosts.count(hosts[rank]) if rank < len(hosts) else world_size
+
     os.environ["MASTER_ADDR"] = master_host
     os.environ["MASTER_PORT"] = master_port
     os.environ["RANK"] = str(rank)
     os.environ["WORLD_SIZE"] = str(world_size)
 
     dist.init_process_group(
-        select_backend(device, backend, world_size),
+        select_backend(device, backend, world_size, local_ranks=local_ranks),
         rank=rank,
         world_size=world_size,
         timeout=datetime.timedelta(seconds=timeout_s),

@@ -396,7 +396,13 @@ def train_distributed(
                 compile_mode=compile_mode,
             )
 
-        states = rdd.mapPartitionsWithIndex(worker).collect()
+        # barrier scheduling is mandatory for sync mode: every rank must be
+        # alive simultaneously for the allGather rendezvous (reference builds
+        # PipelinedRDD(isFromBarrier=True), distributed.py:53-63).  Applied
+        # per round so repartition below still sees the plain RDD (pyspark's
+        # RDDBarrier exposes only mapPartitions*).
+        brdd = rdd.barrier() if hasattr(rdd, "barrier") else rdd
+        states = brdd.mapPartitionsWithIndex(worker).collect()
         state = states[0]
 
         if shuffle_round + 1 < partition_shuffles:

@@ -37,7 +37,7 @@ class PysparkReaderWriter:
     format (reference pipeline_util.py:81-131)."""
 
     def write(self):
-        if HAS_PYSPARK:  # pragma: no cover - needs JVM
+        if HAS_PYSPARK:
             from pyspark.ml.util import JavaMLWriter
 
             return JavaMLWriter(self._to_carrier())
@@ -48,7 +48,7 @@ class PysparkReaderWriter:
 
     @classmethod
     def read(cls):
-        if HAS_PYSPARK:  # pragma: no cover - needs JVM
+        if HAS_PYSPARK:
             from pyspark.ml.feature import StopWordsRemover
             from pyspark.ml.util import JavaMLReader
 
@@ -60,10 +60,16 @@ class PysparkReaderWriter:
         loaded = cls.read().load(path)
         return PysparkPipelineWrapper.unwrap(loaded)
 
-    def _to_carrier(self):  # pragma: no cover - needs JVM
+    def _to_carrier(self):
         from pyspark.ml.feature import StopWordsRemover
 
         return StopWordsRemover(stopWords=obj_to_stopwords(self))
+
+    # pyspark's Pipeline persistence converts python stages to their JVM
+    # counterpart via _to_java (reference pipeline_util.py:113-131); aliasing
+    # the carrier builder makes sparktorch stages saveable inside a plain
+    # pyspark Pipeline/PipelineModel.
+    _to_java = _to_carrier
 
 
 class _LocalStageWriter:
@@ -115,9 +121,16 @@ class PysparkPipelineWrapper:
             pipeline.stages = stages
             return pipeline
 
-        if HAS_PYSPARK:  # pragma: no cover - needs JVM
+        if HAS_PYSPARK:
             from pyspark.ml import Pipeline, PipelineModel
             from pyspark.ml.feature import StopWordsRemover
+
+            # a bare carrier loaded via PysparkReaderWriter.load (not nested
+            # in a pipeline) decodes directly
+            if isinstance(pipeline, StopWordsRemover) and is_sparktorch_stopwords(
+                pipeline.getStopWords()
+            ):
+                return stopwords_to_obj(pipeline.getStopWords())
 
             if isinstance(pipeline, (Pipeline, PipelineModel)):
                 stages = (

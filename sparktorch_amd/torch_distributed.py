@@ -147,22 +147,24 @@ class SparkTorchModel(
             return dataset.withColumn(out_col, preds)
 
         # pyspark path: batched mapPartitions, then rebuild the DataFrame with
-        # the appended prediction column.
-        if not HAS_PYSPARK:  # pragma: no cover
+        # the appended prediction column.  Exercised against real pyspark or
+        # the vendored double (vendor/pyspark).
+        if not HAS_PYSPARK:
             raise RuntimeError("unsupported dataset type: %r" % type(dataset))
-        from pyspark.ml.linalg import Vectors, VectorUDT  # pragma: no cover
-        from pyspark.sql import functions as F  # noqa: F401  # pragma: no cover
-        from pyspark.sql.types import DoubleType, StructField  # pragma: no cover
+        from pyspark.ml.linalg import Vectors, VectorUDT
+        from pyspark.sql.types import DoubleType, StructField
 
-        mod_str = self.getOrDefault(self.modStr)
         bs = int(self.getOrDefault(self.batchSize))
         dev = self._resolve_device()
+        # broadcast the deserialized net once (reference torch_distributed.py:104)
+        # instead of re-deserializing the dill payload in every partition task
+        bc_model = dataset.sparkSession.sparkContext.broadcast(self.getPytorchModel())
 
-        def map_parts(partition):  # pragma: no cover - needs JVM
+        def map_parts(partition):
             import numpy as _np
             import torch as _torch
 
-            model = b64_to_obj(mod_str).to(dev)
+            model = bc_model.value.to(dev)
             model.eval()
             rows = list(partition)
             if not rows:
@@ -182,10 +184,10 @@ class SparkTorchModel(
                         else:
                             yield row + (float(pred[j].reshape(-1)[0]),)
 
-        schema = dataset.schema.add(  # pragma: no cover - needs JVM
+        schema = dataset.schema.add(
             StructField(out_col, VectorUDT() if use_vector else DoubleType())
         )
-        return dataset.rdd.mapPartitions(map_parts).toDF(schema)  # pragma: no cover
+        return dataset.rdd.mapPartitions(map_parts).toDF(schema)
 
 
 class SparkTorch(
@@ -363,8 +365,12 @@ class SparkTorch(
         if mode == "synchronous":
             from sparktorch_amd.parallel.sync import train_distributed
 
+            # sync mode always runs under barrier scheduling (reference
+            # distributed.py:53-63 builds PipelinedRDD(isFromBarrier=True));
+            # train_distributed applies rdd.barrier() per shuffle round so
+            # repartition still happens on the plain RDD.
             state = train_distributed(
-                rdd if _is_local_df(dataset) else _as_barrier(rdd),
+                rdd,
                 torch_obj,
                 iters=iters,
                 partition_shuffles=shuffles,
@@ -380,9 +386,11 @@ class SparkTorch(
             from sparktorch_amd.parallel.server import Server, determine_master
 
             port = self.getOrDefault(self.port)
+            # partition count BEFORE any barrier wrap: pyspark's RDDBarrier
+            # exposes only mapPartitions* (no getNumPartitions)
+            n_parts = rdd.getNumPartitions()
             if use_barrier:
                 rdd = rdd.barrier() if hasattr(rdd, "barrier") else rdd
-            n_parts = rdd.getNumPartitions()
             master_url = _driver_url(dataset, port)
             server = Server(
                 torch_obj,
@@ -421,14 +429,10 @@ class SparkTorch(
         )
 
 
-def _as_barrier(rdd):  # pragma: no cover - pyspark path
-    return rdd
-
-
 def _driver_url(dataset, port: int) -> str:
     if _is_local_df(dataset):
         return "127.0.0.1:%d" % port
-    if HAS_PYSPARK:  # pragma: no cover - needs JVM
+    if HAS_PYSPARK:
         try:
             from pyspark.sql import SparkSession
 
