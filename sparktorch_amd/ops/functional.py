@@ -40,8 +40,10 @@ class _LinearFn(torch.autograd.Function):
         # point is where the stage converted anyway — numerics unchanged)
         wb = ops.ext().cast_f32_bf16(w) if w.dtype == torch.float32 else w
         y = ops.ext().linear_fwd(x, wb, b, relu)
-        ctx.save_for_backward(x, w, y)
-        ctx.wb = wb
+        # wb rides save_for_backward so autograd's saved-tensor versioning
+        # covers it (in-place mutation of w between fwd and bwd is detected);
+        # only the Parameter reference for _bucket_notify stays an attribute.
+        ctx.save_for_backward(x, w, y, wb)
         ctx.relu = relu
         ctx.has_bias = b is not None
         ctx.bias_ref = b
@@ -49,13 +51,13 @@ class _LinearFn(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, dy: torch.Tensor):
-        x, w, y = ctx.saved_tensors
+        x, w, y, wb = ctx.saved_tensors
         ext = ops.ext()
         dy = dy.contiguous()
         if dy.dtype != torch.bfloat16:
             dy = dy.to(torch.bfloat16)
         dz = ext.relu_bwd(dy, y) if ctx.relu else dy
-        dx = ext.linear_dgrad(dz, ctx.wb) if ctx.needs_input_grad[0] else None
+        dx = ext.linear_dgrad(dz, wb) if ctx.needs_input_grad[0] else None
 
         # Direct-grad path: when FlatBuckets installed this param, its .grad
         # is a pre-zeroed flat-bucket view — accumulate straight into it (no

@@ -165,15 +165,24 @@ class HipBatchNorm2d(nn.Module):
 
 
 class HipGlobalAvgPool(nn.Module):
-    """adaptive_avg_pool2d(x, 1).flatten(1) as one reduction kernel."""
+    """Global average pool as one reduction kernel.
 
-    def __init__(self, layout: str = "nchw"):
+    ``keepdim=False`` (default) fuses the usual
+    ``adaptive_avg_pool2d(x, 1).flatten(1)`` idiom and returns [B, C];
+    ``keepdim=True`` preserves AdaptiveAvgPool2d(1)'s [B, C, 1, 1] shape for
+    models that consume the 4D output (SE blocks, later convs, views)."""
+
+    def __init__(self, layout: str = "nchw", keepdim: bool = False):
         super().__init__()
         self.layout = layout
+        self.keepdim = keepdim
 
     def forward(self, x):
         fn = hip_global_avg_pool_nhwc if self.layout == "nhwc" else hip_global_avg_pool
-        return fn(x)
+        y = fn(x)
+        if self.keepdim:
+            return y.view(y.shape[0], y.shape[1], 1, 1)
+        return y
 
 
 class HipDropout(nn.Module):
@@ -337,8 +346,10 @@ def convert_model_for_mi355x(model: nn.Module) -> nn.Module:
             except (ValueError, AttributeError):
                 pass
         elif cname == "AdaptiveAvgPool2d" and child.output_size in (1, (1, 1)):
-            # NB: produces [B,C] (flattened) — matches the F.adaptive_avg_pool2d(x,1).flatten(1) idiom
-            setattr(model, name, HipGlobalAvgPool())
+            # keepdim: the generic converter must preserve AdaptiveAvgPool2d's
+            # [B,C,1,1] output rank — fused models opt into the flattened
+            # variant explicitly
+            setattr(model, name, HipGlobalAvgPool(keepdim=True))
         elif cname == "Dropout2d":
             setattr(model, name, HipDropout(p=child.p, channel_wise=True))
         elif cname == "Dropout":

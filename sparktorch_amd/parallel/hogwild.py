@@ -114,6 +114,10 @@ def handle_model(
         metrics.start()
         sd = get_state_dict(master_url)
         if pinned is not None:
+            # the previous iteration's async H2D must have drained before the
+            # pinned buffers are rewritten — explicit, not implied by some
+            # later blocking call on the compute stream
+            side_stream.synchronize()
             for k, v in sd.items():
                 pinned[k].copy_(v)
             # async H2D on a side stream; the compute stream waits on it

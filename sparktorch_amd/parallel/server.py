@@ -55,6 +55,11 @@ class _PSState:
         self.optimizer = loaded.optimizer
         self.model.share_memory()
         self.lock: Optional[RWLock] = RWLock() if acquire_lock else None
+        # aux state (error budget / loss window / stop flag) is mutated from
+        # ThreadingHTTPServer handler threads regardless of acquireLock —
+        # guard it with its own mutex so a read-modify-reset of the loss
+        # window can't drop or double-count entries under concurrency
+        self.aux_lock = threading.Lock()
         self.error_count = 0
         self.loss_window = []
         self.window_len = max(1, window_len)
@@ -108,22 +113,25 @@ def _make_handler(state: _PSState):
                             state.lock.release()
                     self._send(200, b"ok", "text/plain")
                 except Exception as e:  # error budget, reference server.py:141-144
-                    state.error_count += 1
-                    if state.error_count > 10:
+                    with state.aux_lock:
+                        state.error_count += 1
+                        over = state.error_count > 10
+                    if over:
                         self._send(500, str(e).encode(), "text/plain")
                     else:
                         self._send(200, b"tolerated", "text/plain")
             elif self.path == "/losses":
                 d = json.loads(data.decode("utf-8"))
-                stop = state.should_stop
-                if state.es is not None and not stop:
-                    state.loss_window.append(float(d["loss"]))
-                    if len(state.loss_window) >= state.window_len:
-                        avg = sum(state.loss_window) / len(state.loss_window)
-                        state.loss_window = []
-                        if state.es.step(avg):
-                            state.should_stop = True
-                            stop = True
+                with state.aux_lock:
+                    stop = state.should_stop
+                    if state.es is not None and not stop:
+                        state.loss_window.append(float(d["loss"]))
+                        if len(state.loss_window) >= state.window_len:
+                            avg = sum(state.loss_window) / len(state.loss_window)
+                            state.loss_window = []
+                            if state.es.step(avg):
+                                state.should_stop = True
+                                stop = True
                 self._send(200, json.dumps({"stop": bool(stop)}).encode(), "application/json")
             else:
                 self._send(404, b"not found", "text/plain")

@@ -17,8 +17,10 @@ Re-design of the reference's sync engine (distributed.py:66-261) for MI355X:
 
 Reference behaviors preserved: long-label retry for classification criteria
 (distributed.py:152-157), autoencoder mode when no labels (distributed.py:136),
-minibatch sampling (distributed.py:145-148), early-stop consensus via two tiny
-all-reduces (distributed.py:184-196), identical final states on every rank.
+minibatch sampling (distributed.py:145-148), early-stop consensus — one loss
+all-reduce per iteration; every rank derives the same stop decision from the
+identical reduced value (vs two collectives at distributed.py:184-196) —
+identical final states on every rank.
 """
 
 from __future__ import annotations
@@ -170,6 +172,9 @@ class SyncTrainer:
             # executing) — snapshot the training state so the first replay
             # starts exactly where eager would.
             snap_params = [b.flat_param.clone() for b in self.buckets.buckets]
+            # module buffers mutate during the 2 real warmup steps too
+            # (BatchNorm running stats / num_batches_tracked)
+            snap_buffers = {k: v.clone() for k, v in self.model.named_buffers()}
             snap_opt = None
             if self.optimizer is not None:
                 sd = getattr(self.optimizer, "step_dev", None)
@@ -207,6 +212,8 @@ class SyncTrainer:
             # not execute — only the 2 warmup steps mutated anything)
             for b, p0 in zip(self.buckets.buckets, snap_params):
                 b.flat_param.copy_(p0)
+            for k, v in self.model.named_buffers():
+                v.copy_(snap_buffers[k])
             if snap_opt is not None:
                 self.optimizer.step_count = snap_opt[0]
                 for t, t0 in zip(getattr(self.optimizer, "exp_avg", []), snap_opt[1]):
@@ -301,7 +308,6 @@ def handle_model(
         n = x_train.shape[0]
         # collectives must live on the comm device (RCCL wants GPU tensors)
         comm_dev = dev if dev.startswith("cuda") else "cpu"
-        should_stop_t = torch.zeros(1, device=comm_dev)
         metrics = StepMetrics()
 
         for i in range(iters):
@@ -321,12 +327,15 @@ def handle_model(
                 loss_for_es = (
                     trainer.validation_loss(x_val, y_val) if x_val is not None else loss
                 )
+                # ONE collective per early-stop iteration (vs the reference's
+                # two, distributed.py:188,194): all-reduce guarantees every
+                # rank receives the identical summed loss, and EarlyStopping
+                # is deterministic in its input history, so every rank reaches
+                # the same stop decision locally — the consensus flag needs no
+                # second collective.
                 lt = torch.tensor([loss_for_es], device=comm_dev)
                 dist.all_reduce(lt)
-                avg = float(lt) / world_size
-                should_stop_t[0] = 1.0 if es.step(avg) else 0.0
-                dist.all_reduce(should_stop_t)
-                if float(should_stop_t) > 0:
+                if es.step(float(lt) / world_size):
                     break
 
             if verbose:

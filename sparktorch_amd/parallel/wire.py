@@ -5,7 +5,8 @@ Replaces the reference's dill-serialized state_dict/gradient HTTP payloads
 ``[magic u32][count u32]`` then per tensor ``[dtype u8][ndim u8][pad u16]
 [shape u64 x ndim][nbytes u64][raw bytes]``.  State dicts prepend a JSON key
 table.  Raw bytes move via uint8 views, so bf16/fp16 round-trip without numpy
-dtype support; decode is zero-copy into torch via frombuffer.
+dtype support; decode makes exactly one copy per tensor out of the request
+buffer (frombuffer + copy) so returned tensors own writable memory.
 """
 
 from __future__ import annotations
