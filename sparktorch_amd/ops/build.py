@@ -54,15 +54,26 @@ def _run(cmd, verbose):
     return r
 
 
-def build_extension(verbose: bool = False, force: bool = False) -> str:
-    out = _ext_path()
+def build_extension(
+    verbose: bool = False,
+    force: bool = False,
+    out: str = None,
+    build_dir: str = None,
+) -> str:
+    """Build the extension.  Default: in-tree, mtime-gated.  ``out`` /
+    ``build_dir`` redirect the artifact and objects elsewhere (the from-clean
+    build check uses a temp dir so it always compiles every TU without
+    touching the committed in-tree binary)."""
+    in_tree = out is None
+    out = out or _ext_path()
+    build_dir = build_dir or BUILD_DIR
     srcs = [os.path.join(CSRC, s) for s in HIP_SOURCES + CPP_SOURCES] + [
         os.path.join(CSRC, "common.h")
     ]
-    if not force and not _needs_build(out, srcs):
+    if in_tree and not force and not _needs_build(out, srcs):
         return out
 
-    os.makedirs(BUILD_DIR, exist_ok=True)
+    os.makedirs(build_dir, exist_ok=True)
     torch_inc = cpp_extension.include_paths()
     torch_lib = cpp_extension.library_paths()[0]
     py_inc = sysconfig.get_paths()["include"]
@@ -71,7 +82,7 @@ def build_extension(verbose: bool = False, force: bool = False) -> str:
     objs = []
     hipcc = os.path.join(ROCM, "bin", "hipcc")
     for s in HIP_SOURCES:
-        obj = os.path.join(BUILD_DIR, s.replace(".hip", ".o"))
+        obj = os.path.join(build_dir, s.replace(".hip", ".o"))
         _run(
             [
                 hipcc,
@@ -90,7 +101,7 @@ def build_extension(verbose: bool = False, force: bool = False) -> str:
         objs.append(obj)
 
     for s in CPP_SOURCES:
-        obj = os.path.join(BUILD_DIR, s.replace(".cpp", ".o"))
+        obj = os.path.join(build_dir, s.replace(".cpp", ".o"))
         cmd = (
             ["g++", "-O2", "-std=c++17", "-fPIC", "-c", os.path.join(CSRC, s), "-o", obj]
             + ["-I" + p for p in torch_inc + [py_inc, os.path.join(ROCM, "include")]]
