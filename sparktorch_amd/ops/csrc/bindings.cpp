@@ -23,6 +23,7 @@ hipError_t launch_bias_grad(const bf16raw*, float*, int, int, float*, int,
 hipError_t launch_small_wgrad(const bf16raw*, const bf16raw*, float*, float*, int, int64_t, int,
                               int, hipStream_t);
 hipError_t launch_cast_f64_f32(const double*, float*, int64_t, hipStream_t);
+hipError_t launch_cast_f64_bf16(const double*, bf16raw*, int64_t, hipStream_t);
 hipError_t launch_cast_f32_bf16(const float*, bf16raw*, int64_t, hipStream_t);
 hipError_t launch_ce_fused(const bf16raw*, const int64_t*, float*, bf16raw*, int, int,
                            hipStream_t);
@@ -160,6 +161,14 @@ at::Tensor cast_f64_f32(at::Tensor src) {
   auto dst = at::empty(src.sizes(), src.options().dtype(at::kFloat));
   CHECK_HIP(launch_cast_f64_f32(src.data_ptr<double>(), dst.data_ptr<float>(), src.numel(),
                                 cur_stream()));
+  return dst;
+}
+
+at::Tensor cast_f64_bf16(at::Tensor src) {
+  check_gpu_contig(src, at::kDouble, "src");
+  auto dst = at::empty(src.sizes(), src.options().dtype(at::kBFloat16));
+  CHECK_HIP(launch_cast_f64_bf16(src.data_ptr<double>(), (bf16raw*)dst.data_ptr(), src.numel(),
+                                 cur_stream()));
   return dst;
 }
 
@@ -778,6 +787,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("relu_bwd", &relu_bwd, "dz = dy * (y>0)");
   m.def("bias_grad", &bias_grad, "column-sum of dz");
   m.def("cast_f64_f32", &cast_f64_f32);
+  m.def("cast_f64_bf16", &cast_f64_bf16, "one-pass device Vector pack cast");
   m.def("cast_f32_bf16", &cast_f32_bf16);
   m.def("ce_fused", &ce_fused, "cross-entropy fwd+bwd fused");
   m.def("mse_fused", &mse_fused, "mse fwd+bwd fused");

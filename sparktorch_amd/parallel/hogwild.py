@@ -25,7 +25,7 @@ import torch
 
 from sparktorch_amd.parallel.sync import compute_loss
 from sparktorch_amd.parallel.wire import decode_state_dict, encode_tensors
-from sparktorch_amd.utils.data import handle_features
+from sparktorch_amd.utils.data import handle_features, handle_features_device
 from sparktorch_amd.utils.trace import StepMetrics, trace_range
 from sparktorch_amd.utils.serialize import load_torch_model
 
@@ -89,7 +89,15 @@ def handle_model(
     if device.startswith("cuda") and not torch.cuda.is_available():
         raise RuntimeError("device=%r requested but no GPU is visible" % device)
 
-    feats = handle_features(data, validation_pct)
+    if device.startswith("cuda"):
+        # pinned staging + device-side cast; fp32 activations (the hogwild
+        # model runs eager fp32 — its per-iteration cost is the PS pull/push)
+        import torch as _torch
+
+        feats = handle_features_device(data, validation_pct, device=device,
+                                       dtype=_torch.float32)
+    else:
+        feats = handle_features(data, validation_pct)
     if feats.x_train is None:
         return iter([])
 
@@ -97,10 +105,16 @@ def handle_model(
     model = loaded.model.to(device)
     criterion = loaded.criterion
 
-    x_train = feats.x_train.to(device)
-    y_train = feats.y_train.to(device) if feats.y_train is not None else x_train
-    x_val = feats.x_val.to(device) if feats.x_val is not None else None
-    y_val = feats.y_val.to(device) if feats.y_val is not None else (x_val if x_val is not None else None)
+    x_train = feats.x_train if feats.x_train.is_cuda else feats.x_train.to(device)
+    y_train = (
+        (feats.y_train if feats.y_train.is_cuda else feats.y_train.to(device))
+        if feats.y_train is not None else x_train
+    )
+    x_val = (feats.x_val if feats.x_val.is_cuda else feats.x_val.to(device)) if feats.x_val is not None else None
+    y_val = (
+        (feats.y_val if feats.y_val.is_cuda else feats.y_val.to(device))
+        if feats.y_val is not None else (x_val if x_val is not None else None)
+    )
 
     pinned: Optional[dict] = None
     side_stream = None

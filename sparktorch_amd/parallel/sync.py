@@ -39,7 +39,7 @@ from sparktorch_amd.parallel.rendezvous import (
     init_process_group_from_barrier,
     pick_device,
 )
-from sparktorch_amd.utils.data import handle_features
+from sparktorch_amd.utils.data import handle_features, handle_features_device
 from sparktorch_amd.utils.early_stopper import EarlyStopping
 from sparktorch_amd.utils.trace import StepMetrics, trace_range
 from sparktorch_amd.utils.serialize import (
@@ -276,24 +276,31 @@ def handle_model(
 
     try:
         loaded = load_torch_model(torch_obj_str, device=dev)
-        data = handle_features(iterator, validation_pct)
+        if dev.startswith("cuda"):
+            # device ingest: pinned fp64 staging + hipMemcpyAsync + on-device
+            # cast to bf16 (utils.data.handle_features_device)
+            data = handle_features_device(iterator, validation_pct, device=dev)
+        else:
+            data = handle_features(iterator, validation_pct)
         if data.x_train is None:
             raise RuntimeError(
                 "rank %d received an empty partition; repartition so every "
                 "barrier task has data" % rank
             )
 
-        x_train = data.x_train.to(dev)
+        x_train = data.x_train.to(dev) if not data.x_train.is_cuda else data.x_train
         # autoencoder mode: no labels -> y = x (reference distributed.py:136)
-        y_train = data.y_train.to(dev) if data.y_train is not None else x_train
-        if dev.startswith("cuda"):
-            # bf16 activations once, outside the hot loop
-            ae = y_train is x_train
-            x_train = x_train.to(torch.bfloat16)
-            if ae:
-                y_train = x_train
-        x_val = data.x_val.to(dev) if data.x_val is not None else None
-        y_val = data.y_val.to(dev) if data.y_val is not None else (x_val if x_val is not None else None)
+        y_train = (
+            (data.y_train.to(dev) if not data.y_train.is_cuda else data.y_train)
+            if data.y_train is not None
+            else x_train
+        )
+        x_val = data.x_val.to(dev) if data.x_val is not None and not data.x_val.is_cuda else data.x_val
+        y_val = (
+            (data.y_val.to(dev) if not data.y_val.is_cuda else data.y_val)
+            if data.y_val is not None
+            else (x_val if x_val is not None else None)
+        )
 
         trainer = SyncTrainer(
             loaded.model,

@@ -71,6 +71,78 @@ def pack_partition(data: List[DataObj]):
     return x, y
 
 
+def handle_features_device(
+    data: List[DataObj],
+    validation_pct: float = 0.0,
+    device: str = "cuda:0",
+    dtype=None,
+) -> DataObj:
+    """GPU ingest: partition rows -> device tensors with the cast ON DEVICE.
+
+    Replaces the reference's CPU pack (np.stack of fp64 rows + ``.float()``
+    cast + blocking ``.to(device)``, reference util.py:87-99) with the
+    MI355X path promised in SURVEY.md §2.2:
+
+      np.stack (C-level row gather, fp64 kept as-is, no CPU cast pass)
+      -> pinned staging tensor -> hipMemcpyAsync H2D
+      -> one ``cast_f64_bf16`` kernel (8 TB/s HBM vs a CPU cast pass)
+      -> optional on-device validation split by index_select.
+
+    Rows that already arrive fp32 skip the cast kernel.  Returns a DataObj of
+    DEVICE tensors: x in ``dtype`` (default bf16), y fp32 (downstream losses
+    cast labels as needed).
+    """
+    import torch as _torch
+
+    dtype = dtype or _torch.bfloat16
+    data = list(data)
+    if len(data) == 0:
+        return DataObj(None, None, None, None)
+
+    from sparktorch_amd import ops as _ops
+
+    ext = _ops.ext()
+
+    rows = [np.asarray(d.x_train).reshape(-1) for d in data]
+    x_host = np.stack(rows)  # C gather; stays fp64 if rows are fp64
+    x_pin = _torch.from_numpy(x_host).pin_memory()
+    x_dev = x_pin.to(device, non_blocking=True)  # hipMemcpyAsync from pinned
+    if x_dev.dtype == _torch.float64:
+        x = ext.cast_f64_bf16(x_dev) if dtype == _torch.bfloat16 else ext.cast_f64_f32(x_dev)
+    elif x_dev.dtype != dtype:
+        x = ext.cast_f32_bf16(x_dev) if dtype == _torch.bfloat16 else x_dev.to(dtype)
+    else:
+        x = x_dev
+
+    y = None
+    if data[0].y_train is not None:
+        y_rows = [
+            [d.y_train] if np.isscalar(d.y_train)
+            else np.asarray(d.y_train, dtype=np.float32).reshape(-1)
+            for d in data
+        ]
+        y_pin = _torch.from_numpy(np.asarray(y_rows, dtype=np.float32)).pin_memory()
+        y = y_pin.to(device, non_blocking=True)
+
+    n = x.shape[0]
+    x_val = y_val = None
+    if validation_pct and validation_pct > 0.0 and n > 1:
+        n_val = int(n * validation_pct)
+        if n_val > 0:
+            val_idx = np.random.choice(n, n_val, replace=False)
+            mask = np.ones(n, dtype=bool)
+            mask[val_idx] = False
+            tr = _torch.from_numpy(np.nonzero(mask)[0]).to(device, non_blocking=True)
+            va = _torch.from_numpy(np.nonzero(~mask)[0]).to(device, non_blocking=True)
+            x_val = x.index_select(0, va)
+            x = x.index_select(0, tr)
+            if y is not None:
+                y_val = y.index_select(0, va)
+                y = y.index_select(0, tr)
+    _torch.cuda.synchronize()  # pinned buffers die with this frame
+    return DataObj(x_train=x, y_train=y, x_val=x_val, y_val=y_val)
+
+
 def handle_features(data: List[DataObj], validation_pct: float = 0.0) -> DataObj:
     """Partition of row-DataObj -> one DataObj of stacked float32 torch tensors.
 
