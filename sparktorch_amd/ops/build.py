@@ -20,7 +20,7 @@ OPS_DIR = os.path.dirname(os.path.abspath(__file__))
 CSRC = os.path.join(OPS_DIR, "csrc")
 BUILD_DIR = os.path.join(OPS_DIR, "build")
 
-HIP_SOURCES = ["elementwise.hip", "loss.hip", "gemm.hip", "conv.hip", "norm.hip", "conv_nhwc.hip"]
+HIP_SOURCES = ["elementwise.hip", "loss.hip", "gemm.hip", "conv.hip", "norm.hip", "conv_nhwc.hip", "conv_implicit.hip"]
 CPP_SOURCES = ["bindings.cpp"]
 
 ROCM = os.environ.get("ROCM_PATH", "/opt/rocm")

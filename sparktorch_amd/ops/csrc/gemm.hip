@@ -52,6 +52,10 @@
 #define EPI_BIAS 2       // bf16 store, + bias[n]
 #define EPI_BIAS_RELU 3  // bf16 store, + bias[n], relu
 #define EPI_RELU 4       // bf16 store, relu (no bias)
+#define EPI_F32_SLAB 5   // fp32 per-slice slab store (split-K without atomics;
+                         // a separate reduce kernel combines the slabs —
+                         // guide §5 "splitk-seam": slab reducer beats an
+                         // fp32-atomicAdd accumulator)
 
 typedef shortx8 frag_t;  // 8 bf16 (4 VGPRs)
 
@@ -590,7 +594,7 @@ __global__ __launch_bounds__(WR * WC * 64, 2) void gemm_kernel(const void* __res
   // per-lane stores directly.  fp32/split-K keeps the scalar path (atomics).
   const int m_base = m0 + wr * 64;
   const int n_base = n0 + wc * 64;
-  if (EPI != EPI_F32 && (N & 7) == 0 && n_base + 64 <= N) {
+  if (EPI != EPI_F32 && EPI != EPI_F32_SLAB && (N & 7) == 0 && n_base + 64 <= N) {
     // all waves must be past their final fragment reads before the scratch
     // overwrites the staging images (the T14 loop has no trailing barrier)
     __syncthreads();
@@ -630,6 +634,28 @@ __global__ __launch_bounds__(WR * WC * 64, 2) void gemm_kernel(const void* __res
     }
     return;
   }
+  if (EPI == EPI_F32_SLAB) {
+    // split-K slab: this slice's full [BMtxBNt] fp32 tile goes to its own
+    // workspace slab with PLAIN coalesced stores (consecutive lanes hit
+    // consecutive n) — no fabric-serialized per-dword atomics.  The ones
+    // column rides along at its n; the reduce kernel routes it to db.
+    float* ws = Cf + (int64_t)blockIdx.z * M * N;
+#pragma unroll
+    for (int mi = 0; mi < 4; ++mi) {
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        int n = n_base + ni * 16 + l15;
+        if (n >= N) continue;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int m = m_base + mi * 16 + kg * 4 + r;
+          if (m >= M) continue;
+          ws[(int64_t)m * N + n] = acc[mi][ni][r];
+        }
+      }
+    }
+    return;
+  }
 #pragma unroll
   for (int mi = 0; mi < 4; ++mi) {
 #pragma unroll
@@ -658,6 +684,131 @@ __global__ __launch_bounds__(WR * WC * 64, 2) void gemm_kernel(const void* __res
       }
     }
   }
+}
+
+// Combine the split-K slabs: dw[m,n] += sum_z ws[z,m,n]; the virtual ones
+// column (bias) routes to db.  Memory-bound: zs*M*N fp32 reads, float4 loads
+// where the row layout allows.
+// TRANSPOSE=false: dw is [M][N-ones] row-major (+ db for the ones column).
+// TRANSPOSE=true: the GEMM computed dW^T (slabs are [M=Kcol][N=CO]); store
+// dw[n][m] — the transposed-wgrad route for CO<=64 layers where the direct
+// orientation wastes half a 128-row tile (M utilization).  The scattered
+// 4 B writes are MN elements of a tiny matrix; the coalesced zs*MN slab
+// reads dominate.
+template <bool TRANSPOSE>
+__global__ void wgrad_reduce_kernel(const float* __restrict__ ws, float* __restrict__ dw,
+                                    float* __restrict__ db, int64_t MN, int N, int zs,
+                                    int ones_row) {
+  int64_t i0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 4;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x * 4;
+  const int nw = N - (ones_row >= 0 ? 1 : 0);
+  const int64_t Mrows = MN / N;
+  for (int64_t i = i0; i < MN; i += stride) {
+    if (i + 4 <= MN && (i % N) + 4 <= (int64_t)N) {
+      floatx4 s = {0.f, 0.f, 0.f, 0.f};
+      for (int z = 0; z < zs; ++z) {
+        floatx4 v = *(const floatx4*)(ws + (int64_t)z * MN + i);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) s[j] += v[j];
+      }
+      int n = (int)(i % N);
+      int64_t m = i / N;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        if (TRANSPOSE) {
+          dw[(int64_t)(n + j) * Mrows + m] += s[j];
+        } else if (ones_row >= 0 && n + j == ones_row) {
+          if (db) db[m] += s[j];
+        } else {
+          dw[m * nw + n + j] += s[j];
+        }
+      }
+    } else {
+      for (int64_t k = i; k < i + 4 && k < MN; ++k) {
+        float s = 0.f;
+        for (int z = 0; z < zs; ++z) s += ws[(int64_t)z * MN + k];
+        int n = (int)(k % N);
+        int64_t m = k / N;
+        if (TRANSPOSE) {
+          dw[(int64_t)n * Mrows + m] += s;
+        } else if (ones_row >= 0 && n == ones_row) {
+          if (db) db[m] += s;
+        } else {
+          dw[m * nw + n] += s;
+        }
+      }
+    }
+  }
+}
+
+// Host helper: number of K slices the slab path will use for (K, splitk) —
+// callers size the workspace as zs * M * N floats.
+extern "C" int wgrad_slab_slices(int K, int splitk) {
+  if (splitk < 0) splitk = -splitk;
+  if (splitk < 1) splitk = 1;
+  if (splitk == 1) return 1;
+  int kps = (int)ceil_div_i64(ceil_div_i64(K, splitk), 64) * 64;
+  return (int)ceil_div_i64(K, kps);
+}
+
+// Split-K wgrad without atomics: per-slice slab stores + one reduce launch.
+// Same operand contract as launch_gemm_bf16's dW path (A = dz^T k-strided,
+// B = x k-strided, optional virtual ones column for the fused bias grad);
+// ws must hold wgrad_slab_slices(K, splitk) * M * N floats.
+// transpose != 0: the caller swapped operands to compute dW^T (M = Kcol,
+// N = CO); the reduce writes dw back in [CO][Kcol] orientation.  Use for
+// CO <= 64 layers (stem/l1-class) where direct orientation leaves half the
+// 128-row tile dead.  ones_row must be -1 with transpose.
+extern "C" hipError_t launch_wgrad_slab(const void* A, const void* B, int b_is_f32, float* dw,
+                                        float* db, int M, int N, int K, int64_t sam, int64_t sak,
+                                        int64_t sbk, int64_t sbn, int splitk, int ones_row,
+                                        float* ws, int transpose, hipStream_t stream) {
+  if (splitk < 0) splitk = -splitk;
+  if (splitk < 1) splitk = 1;
+  int kps = K;
+  int zs = 1;
+  if (splitk > 1) {
+    kps = (int)ceil_div_i64(ceil_div_i64(K, splitk), 64) * 64;
+    zs = (int)ceil_div_i64(K, kps);
+  }
+  const bool narrow = (N <= 64 && (transpose ? M > 64 : M > 256));
+  const bool wide = !narrow && N >= 256 && M > 64;
+  const int bm = narrow ? 256 : BM, bn = narrow ? 64 : (wide ? 256 : BN);
+  dim3 grid((unsigned)ceil_div_i64(M, bm), (unsigned)ceil_div_i64(N, bn), (unsigned)zs);
+  dim3 block(wide ? 512 : 256);
+
+#define DISPATCH_SLAB(BF32)                                                                   \
+  do {                                                                                        \
+    if (wide)                                                                                 \
+      gemm_kernel<BF32, EPI_F32_SLAB, true, 2, 4, false><<<grid, block, 0, stream>>>(         \
+          A, B, ws, nullptr, nullptr, M, N, K, sam, sak, sbk, sbn, kps, nullptr, ones_row); \
+    else if (narrow)                                                                          \
+      gemm_kernel<BF32, EPI_F32_SLAB, true, 4, 1, false><<<grid, block, 0, stream>>>(         \
+          A, B, ws, nullptr, nullptr, M, N, K, sam, sak, sbk, sbn, kps, nullptr, ones_row); \
+    else                                                                                      \
+      gemm_kernel<BF32, EPI_F32_SLAB, true, 2, 2, false><<<grid, block, 0, stream>>>(         \
+          A, B, ws, nullptr, nullptr, M, N, K, sam, sak, sbk, sbn, kps, nullptr, ones_row); \
+  } while (0)
+
+  if (b_is_f32)
+    DISPATCH_SLAB(true);
+  else
+    DISPATCH_SLAB(false);
+#undef DISPATCH_SLAB
+  HIP_CHECK_LAUNCH();
+
+  int64_t MN = (int64_t)M * N;
+  int rblock = 256;
+  int64_t rg = ceil_div_i64(MN, (int64_t)rblock * 4);
+  if (rg > 2048) rg = 2048;
+  if (transpose)
+    wgrad_reduce_kernel<true><<<dim3((unsigned)rg), dim3(rblock), 0, stream>>>(ws, dw, db, MN, N,
+                                                                               zs, ones_row);
+  else
+    wgrad_reduce_kernel<false><<<dim3((unsigned)rg), dim3(rblock), 0, stream>>>(ws, dw, db, MN, N,
+                                                                                zs, ones_row);
+  HIP_CHECK_LAUNCH();
+  return hipSuccess;
 }
 
 extern "C" hipError_t launch_gemm_bf16(const void* A, const void* B, int b_is_f32, float* Cf,

@@ -104,8 +104,16 @@ def handle_features_device(
     ext = _ops.ext()
 
     rows = [np.asarray(d.x_train).reshape(-1) for d in data]
-    x_host = np.stack(rows)  # C gather; stays fp64 if rows are fp64
-    x_pin = _torch.from_numpy(x_host).pin_memory()
+    # stack straight INTO the pinned staging tensor: one C-level gather pass,
+    # no separate pin_memory() copy of the whole matrix afterwards
+    np_dtype = np.asarray(rows[0]).dtype
+    t_dtype = {np.dtype(np.float64): _torch.float64,
+               np.dtype(np.float32): _torch.float32}.get(np_dtype)
+    if t_dtype is None:
+        rows = [r.astype(np.float32) for r in rows]
+        t_dtype = _torch.float32
+    x_pin = _torch.empty((len(rows), rows[0].size), dtype=t_dtype, pin_memory=True)
+    np.stack(rows, out=x_pin.numpy())
     x_dev = x_pin.to(device, non_blocking=True)  # hipMemcpyAsync from pinned
     if x_dev.dtype == _torch.float64:
         x = ext.cast_f64_bf16(x_dev) if dtype == _torch.bfloat16 else ext.cast_f64_f32(x_dev)
