@@ -22,13 +22,13 @@ hipError_t launch_bias_grad(const bf16raw*, float*, int, int, float*, int,
                             hipStream_t);
 hipError_t launch_small_wgrad(const bf16raw*, const bf16raw*, float*, float*, int, int64_t, int,
                               int, hipStream_t);
-hipError_t launch_pad_nhwc(const bf16raw*, bf16raw*, int, int, int, int, hipStream_t);
-hipError_t launch_flip_w2d(const bf16raw*, bf16raw*, int, int, hipStream_t);
-hipError_t launch_conv3x3s1_fwd(const bf16raw*, const bf16raw*, const float*, bf16raw*, int, int,
-                                int, int, int, int, hipStream_t);
-hipError_t launch_conv3x3s1_wgrad(const bf16raw*, const bf16raw*, float*, int, int, int, int,
-                                  int, int, float*, hipStream_t);
-int conv3x3s1_wgrad_slices(int, int);
+hipError_t launch_pad_nhwc(const bf16raw*, bf16raw*, int, int, int, int, int, hipStream_t);
+hipError_t launch_flip_w2d(const bf16raw*, bf16raw*, int, int, int, hipStream_t);
+hipError_t launch_conv_implicit_fwd(const bf16raw*, const bf16raw*, const float*, bf16raw*, int,
+                                    int, int, int, int, int, int, int, int, hipStream_t);
+hipError_t launch_conv_implicit_wgrad(const bf16raw*, const bf16raw*, float*, int, int, int, int,
+                                      int, int, int, int, float*, hipStream_t);
+int conv_implicit_wgrad_slices(int, int);
 hipError_t launch_cast_f64_f32(const double*, float*, int64_t, hipStream_t);
 hipError_t launch_cast_f64_bf16(const double*, bf16raw*, int64_t, hipStream_t);
 hipError_t launch_cast_f32_bf16(const float*, bf16raw*, int64_t, hipStream_t);
@@ -165,65 +165,70 @@ at::Tensor bias_grad(at::Tensor dz) {
 
 // ---- implicit-GEMM 3x3 s1 conv (conv_implicit.hip) ----
 
-at::Tensor pad_nhwc(at::Tensor x) {
+at::Tensor pad_nhwc(at::Tensor x, int64_t P) {
   check_gpu_contig(x, at::kBFloat16, "x");
   TORCH_CHECK(x.dim() == 4 && x.size(3) % 8 == 0, "pad_nhwc wants [B,H,W,C], C%8==0");
   int B = (int)x.size(0), H = (int)x.size(1), W = (int)x.size(2), C = (int)x.size(3);
-  auto xP = at::empty({B, H + 2, W + 2, C}, x.options());
+  auto xP = at::empty({B, H + 2 * P, W + 2 * P, C}, x.options());
   CHECK_HIP(launch_pad_nhwc((const bf16raw*)x.data_ptr(), (bf16raw*)xP.data_ptr(), B, H, W, C,
-                            cur_stream()));
+                            (int)P, cur_stream()));
   return xP;
 }
 
-at::Tensor flip_w2d(at::Tensor w2d, int64_t CI) {
+at::Tensor flip_w2d(at::Tensor w2d, int64_t CI, int64_t KHW) {
   check_gpu_contig(w2d, at::kBFloat16, "w2d");
-  TORCH_CHECK(w2d.dim() == 2 && w2d.size(1) == 9 * CI, "w2d must be [CO, 9*CI]");
+  TORCH_CHECK(w2d.dim() == 2 && w2d.size(1) == KHW * CI, "w2d must be [CO, KH*KW*CI]");
   int CO = (int)w2d.size(0);
-  auto wf = at::empty({CI, 9 * CO}, w2d.options());
+  auto wf = at::empty({CI, KHW * CO}, w2d.options());
   CHECK_HIP(launch_flip_w2d((const bf16raw*)w2d.data_ptr(), (bf16raw*)wf.data_ptr(), CO, (int)CI,
-                            cur_stream()));
+                            (int)KHW, cur_stream()));
   return wf;
 }
 
-at::Tensor conv3x3s1_fwd(at::Tensor xP, at::Tensor wmat, c10::optional<at::Tensor> bias,
-                         bool relu) {
+at::Tensor conv_implicit_fwd(at::Tensor xP, at::Tensor wmat, c10::optional<at::Tensor> bias,
+                             int64_t KH, int64_t KW, bool relu) {
   check_gpu_contig(xP, at::kBFloat16, "xP");
   check_gpu_contig(wmat, at::kBFloat16, "wmat");
-  int B = (int)xP.size(0), Hp = (int)xP.size(1), Wp = (int)xP.size(2), CR = (int)xP.size(3);
-  int H = Hp - 2, W = Wp - 2;
+  int B = (int)xP.size(0), Hp = (int)xP.size(1), Wp = (int)xP.size(2), CI = (int)xP.size(3);
+  int OH = Hp - (int)KH + 1, OW = Wp - (int)KW + 1;
   int N = (int)wmat.size(0);
-  TORCH_CHECK(wmat.size(1) == 9 * CR, "wmat k-dim must be 9*C of xP");
-  TORCH_CHECK(CR % 64 == 0 && N % 64 == 0, "implicit conv wants C%64==0");
-  auto y = at::empty({(int64_t)B * H * W, N}, xP.options());
+  int Kpad = (int)wmat.size(1);
+  TORCH_CHECK(Kpad % 64 == 0 && Kpad >= KH * KW * CI, "wmat k-dim must be K padded to x64");
+  TORCH_CHECK(CI % 16 == 0 && N % 8 == 0, "implicit conv wants CI%16==0, N%8==0");
+  auto y = at::empty({(int64_t)B * OH * OW, N}, xP.options());
   const float* bp = nullptr;
   if (bias.has_value()) {
     check_gpu_contig(*bias, at::kFloat, "bias");
     bp = bias->data_ptr<float>();
   }
-  CHECK_HIP(launch_conv3x3s1_fwd((const bf16raw*)xP.data_ptr(), (const bf16raw*)wmat.data_ptr(),
-                                 bp, (bf16raw*)y.data_ptr(), B, H, W, CR, N, relu ? 1 : 0,
-                                 cur_stream()));
+  CHECK_HIP(launch_conv_implicit_fwd((const bf16raw*)xP.data_ptr(),
+                                     (const bf16raw*)wmat.data_ptr(), bp, (bf16raw*)y.data_ptr(),
+                                     B, Hp, Wp, CI, (int)KH, (int)KW, N, Kpad, relu ? 1 : 0,
+                                     cur_stream()));
   return y;
 }
 
-at::Tensor conv3x3s1_wgrad(at::Tensor dz, at::Tensor xP, int64_t splitk, bool slab) {
+at::Tensor conv_implicit_wgrad(at::Tensor dz, at::Tensor xP, int64_t KH, int64_t KW,
+                               int64_t splitk, bool slab) {
   check_gpu_contig(dz, at::kBFloat16, "dz");
   check_gpu_contig(xP, at::kBFloat16, "xP");
   int B = (int)xP.size(0), Hp = (int)xP.size(1), Wp = (int)xP.size(2), CI = (int)xP.size(3);
-  int H = Hp - 2, W = Wp - 2;
+  int OH = Hp - (int)KH + 1, OW = Wp - (int)KW + 1;
   int CO = (int)dz.size(1);
-  TORCH_CHECK(dz.size(0) == (int64_t)B * H * W, "dz rows must be B*H*W");
-  auto dw = at::zeros({CO, 9 * (int64_t)CI}, xP.options().dtype(at::kFloat));
+  TORCH_CHECK(dz.size(0) == (int64_t)B * OH * OW, "dz rows must be B*OH*OW");
+  int64_t N = KH * KW * CI;
+  auto dw = at::zeros({CO, N}, xP.options().dtype(at::kFloat));
   float* wsp = nullptr;
   at::Tensor ws;
   if (slab) {
-    int zs = conv3x3s1_wgrad_slices(B * H * W, (int)splitk);
-    ws = at::empty({zs, CO, 9 * (int64_t)CI}, dw.options());
+    int zs = conv_implicit_wgrad_slices(B * OH * OW, (int)splitk);
+    ws = at::empty({zs, CO, N}, dw.options());
     wsp = ws.data_ptr<float>();
   }
-  CHECK_HIP(launch_conv3x3s1_wgrad((const bf16raw*)dz.data_ptr(), (const bf16raw*)xP.data_ptr(),
-                                   dw.data_ptr<float>(), CO, B, H, W, CI, (int)splitk, wsp,
-                                   cur_stream()));
+  CHECK_HIP(launch_conv_implicit_wgrad((const bf16raw*)dz.data_ptr(),
+                                       (const bf16raw*)xP.data_ptr(), dw.data_ptr<float>(), CO,
+                                       B, Hp, Wp, CI, (int)KH, (int)KW, (int)splitk, wsp,
+                                       cur_stream()));
   return dw;
 }
 
@@ -857,10 +862,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_sgd", &fused_sgd, "fused SGD on a flat bucket");
   m.def("relu_bwd", &relu_bwd, "dz = dy * (y>0)");
   m.def("bias_grad", &bias_grad, "column-sum of dz");
-  m.def("pad_nhwc", &pad_nhwc, "zero-pad ring for implicit 3x3 s1 conv");
-  m.def("flip_w2d", &flip_w2d, "dgrad weight transform W[co,(kh,kw),ci] -> W'[ci,(2-kh,2-kw),co]");
-  m.def("conv3x3s1_fwd", &conv3x3s1_fwd, "implicit-GEMM 3x3 s1 conv forward / dgrad");
-  m.def("conv3x3s1_wgrad", &conv3x3s1_wgrad, "implicit-GEMM 3x3 s1 conv weight grad");
+  m.def("pad_nhwc", &pad_nhwc, "zero-pad ring (width P) for implicit conv");
+  m.def("flip_w2d", &flip_w2d, "dgrad weight transform W[co,k,ci] -> W'[ci,rev(k),co]");
+  m.def("conv_implicit_fwd", &conv_implicit_fwd, "implicit-GEMM s1 conv forward / dgrad");
+  m.def("conv_implicit_wgrad", &conv_implicit_wgrad, "implicit-GEMM s1 conv weight grad");
   m.def("cast_f64_f32", &cast_f64_f32);
   m.def("cast_f64_bf16", &cast_f64_bf16, "one-pass device Vector pack cast");
   m.def("cast_f32_bf16", &cast_f32_bf16);

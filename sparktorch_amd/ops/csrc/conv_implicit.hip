@@ -1,27 +1,28 @@
-// Implicit-GEMM 3x3 stride-1 NHWC conv for gfx950 — no col matrix.
+// Implicit-GEMM stride-1 NHWC conv for gfx950 — no col matrix.
 //
-// The explicit im2col path (conv_nhwc.hip) materializes col [B*H*W, 9*CI]
-// (9x the activation bytes), then streams it through the GEMM and back
-// through col2im for dgrad: for ResNet-18 at B=256 that is ~3.6 ms/step of
-// pure data movement plus 9x-duplicated wgrad re-reads.  Here the GEMM's
-// A-operand staging computes patch addresses directly into a ZERO-PADDED
-// NHWC tensor xP [B][H+2][W+2][CI]:
+// The explicit im2col path (conv_nhwc.hip) materializes col [M, KH*KW*CI]
+// (KH*KW times the activation bytes), streams it through the GEMM, and
+// scatters back through col2im for dgrad.  Here the GEMM's A-operand
+// staging computes patch addresses directly into a ZERO-RING-PADDED NHWC
+// tensor xP [B][H+2P][W+2P][CI]:
 //
-//   * one k-chunk (BKT <= CI, CI % BKT == 0 for the resnet channel counts)
-//     lies inside a single (kh,kw) ci-run, so every staged LDS row is one
-//     contiguous 128 B (or 64 B) global read — async global_load_lds with
-//     the same row-XOR slot swizzle as the explicit-GEMM path;
-//   * padding costs zero branches: the pad ring is materialized once per
-//     tensor (pad_nhwc), so no patch address is ever out of bounds;
-//   * dgrad IS this kernel: dx = conv3x3s1(pad(dz), flip(W)) with
-//     Wflip[ci, (kh,kw), co] = W[co, (2-kh,2-kw), ci] (flip_w2d) — no dcol,
-//     no col2im scatter;
-//   * wgrad keeps the T14 register pipeline of gemm.hip but stages its B
-//     tile from x patches — the 9x-duplicated reads become L2-served
-//     re-reads of the 1x tensor.
+//   * every 8-element (16 B) k-granule lies inside one (kh,kw) ci-run
+//     (CI % 8 == 0), so clean k-chunks stage by async global_load_lds with
+//     the usual row-XOR slot swizzle; the K tail (K % BKT != 0) takes a
+//     zero-filling register fallback into the same swizzled image;
+//   * the pad ring is materialized once per tensor (pad_nhwc) — no patch
+//     address is ever out of bounds, no branches in the hot stage;
+//   * dgrad IS the forward kernel: dx = conv_s1(pad(dz, KH-1-p), flipW)
+//     with flipW[ci, (kh,kw), co] = W[co, (KH-1-kh, KW-1-kw), ci]
+//     (flip_w2d) — no dcol, no col2im scatter;
+//   * wgrad keeps gemm.hip's T14 register pipeline but stages its B tile
+//     from x patches (CI % 16 == 0) — the duplicated col stream becomes
+//     L2-served re-reads of the 1x tensor; split-K combines by fp32
+//     atomics or per-slice slabs + reduce (measured per shape).
 //
-// Supported: KH=KW=3, stride 1, pad 1, CI % 64 == 0 (CI % 32 for the
-// narrow tile), any CO % 64 == 0.  Other convs keep the explicit path.
+// Eligibility (enforced by the python dispatch): stride 1, CI % 16 == 0,
+// N (output channels) >= 32; wmat column-padded to a multiple of 64.
+// Everything else keeps the explicit im2col path.
 
 #include "common.h"
 
@@ -33,14 +34,25 @@ __device__ __forceinline__ int ic_swz(int row, int col) {
   return col ^ (((row >> 4) & 3) << 3);
 }
 
+// Patch geometry: maps (output row, k) -> padded-input address.
+struct ConvGeom {
+  int OW;    // output width (stride 1: OW = Wp - KW + 1)
+  int OHW;   // OH * OW
+  int Hp, Wp;  // padded input spatial
+  int CI;
+  int KW;
+  int KWCI;  // KW * CI
+  int Kreal; // KH * KW * CI
+};
+
 // ---------------------------------------------------------------------------
 // helpers: pad ring + weight flip
 // ---------------------------------------------------------------------------
 
-// xP[b, h+1, w+1, :] = x[b, h, w, :]; ring = 0.  CI % 8 == 0.
+// xP[b, h+P, w+P, :] = x[b, h, w, :]; ring = 0.  CI % 8 == 0.
 __global__ void pad_nhwc_kernel(const bf16raw* __restrict__ x, bf16raw* __restrict__ xP, int B,
-                                int H, int W, int CI) {
-  const int Hp = H + 2, Wp = W + 2;
+                                int H, int W, int CI, int P) {
+  const int Hp = H + 2 * P, Wp = W + 2 * P;
   int64_t total = (int64_t)B * Hp * Wp * (CI >> 3);
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total; i += stride) {
@@ -51,8 +63,8 @@ __global__ void pad_nhwc_kernel(const bf16raw* __restrict__ x, bf16raw* __restri
     int h = (int)(t % Hp);
     int b = (int)(t / Hp);
     shortx8* dst = (shortx8*)(xP + (((int64_t)b * Hp + h) * Wp + w) * CI + (c8 << 3));
-    if (h >= 1 && h <= H && w >= 1 && w <= W) {
-      *dst = *(const shortx8*)(x + (((int64_t)b * H + (h - 1)) * W + (w - 1)) * CI + (c8 << 3));
+    if (h >= P && h < H + P && w >= P && w < W + P) {
+      *dst = *(const shortx8*)(x + (((int64_t)b * H + (h - P)) * W + (w - P)) * CI + (c8 << 3));
     } else {
       shortx8 z = {0, 0, 0, 0, 0, 0, 0, 0};
       *dst = z;
@@ -61,50 +73,50 @@ __global__ void pad_nhwc_kernel(const bf16raw* __restrict__ x, bf16raw* __restri
 }
 
 extern "C" hipError_t launch_pad_nhwc(const bf16raw* x, bf16raw* xP, int B, int H, int W, int CI,
-                                      hipStream_t stream) {
-  int64_t total = (int64_t)B * (H + 2) * (W + 2) * (CI >> 3);
+                                      int P, hipStream_t stream) {
+  int64_t total = (int64_t)B * (H + 2 * P) * (W + 2 * P) * (CI >> 3);
   int64_t g = ceil_div_i64(total, 256);
   if (g > 8192) g = 8192;
-  pad_nhwc_kernel<<<dim3((unsigned)g), dim3(256), 0, stream>>>(x, xP, B, H, W, CI);
+  pad_nhwc_kernel<<<dim3((unsigned)g), dim3(256), 0, stream>>>(x, xP, B, H, W, CI, P);
   HIP_CHECK_LAUNCH();
   return hipSuccess;
 }
 
-// Wflip[ci][(kh*3+kw)*CO + co] = W[co][((2-kh)*3+(2-kw))*CI + ci]
+// Wflip[ci][(kh*KW+kw)*CO + co] = W[co][((KH-1-kh)*KW + (KW-1-kw))*CI + ci]
 __global__ void flip_w2d_kernel(const bf16raw* __restrict__ w2d, bf16raw* __restrict__ wf,
-                                int CO, int CI) {
-  int64_t total = (int64_t)CO * 9 * CI;
+                                int CO, int CI, int KHW) {
+  int64_t total = (int64_t)CO * KHW * CI;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total; i += stride) {
     int co = (int)(i % CO);  // co fastest -> coalesced writes
     int64_t t = i / CO;
-    int kk = (int)(t % 9);
-    int ci = (int)(t / 9);
-    wf[((int64_t)ci * 9 + kk) * CO + co] = w2d[(int64_t)co * 9 * CI + (8 - kk) * CI + ci];
+    int kk = (int)(t % KHW);
+    int ci = (int)(t / KHW);
+    wf[((int64_t)ci * KHW + kk) * CO + co] =
+        w2d[(int64_t)co * KHW * CI + (int64_t)(KHW - 1 - kk) * CI + ci];
   }
 }
 
-extern "C" hipError_t launch_flip_w2d(const bf16raw* w2d, bf16raw* wf, int CO, int CI,
+extern "C" hipError_t launch_flip_w2d(const bf16raw* w2d, bf16raw* wf, int CO, int CI, int KHW,
                                       hipStream_t stream) {
-  int64_t g = ceil_div_i64((int64_t)CO * 9 * CI, 256);
+  int64_t g = ceil_div_i64((int64_t)CO * KHW * CI, 256);
   if (g > 4096) g = 4096;
-  flip_w2d_kernel<<<dim3((unsigned)g), dim3(256), 0, stream>>>(w2d, wf, CO, CI);
+  flip_w2d_kernel<<<dim3((unsigned)g), dim3(256), 0, stream>>>(w2d, wf, CO, CI, KHW);
   HIP_CHECK_LAUNCH();
   return hipSuccess;
 }
 
 // ---------------------------------------------------------------------------
 // forward / dgrad kernel: Y[M, N] = patches(xP) @ Wmat^T
-//   M = B*H*W (stride-1 pad-1: OH=H, OW=W), K = 9*CRUN, N = output channels.
-//   Wmat [N][K] k-contiguous (w2d for fwd; flip_w2d output for dgrad).
+//   M = B*OH*OW, K = KH*KW*CI (padded to Kpad % BKT == 0 in wmat), N = CO.
 // ---------------------------------------------------------------------------
 
-// A: one LDS row = BKT elements of one patch row — a contiguous slice of a
-// (kh,kw) ci-run; per-row global base from (row -> b,oh,ow; kt -> kh,kw,ci0).
+// clean chunk (kt + BKT <= Kreal): one glds per (row, slot); slot decode of
+// (kh, kw, ci) — each 16 B granule is inside one ci-run.
 template <int ROWS, int BKT>
 __device__ __forceinline__ void stage_patch_glds(const bf16raw* __restrict__ xP,
                                                  bf16raw* __restrict__ lds, int row0, int M,
-                                                 int HW, int W, int CI, int kt) {
+                                                 const ConvGeom g, int kt) {
   constexpr int SLOTS = BKT / 8;
   constexpr int CH_ROWS = 64 / SLOTS;
   constexpr int NCHUNK = ROWS / CH_ROWS;
@@ -112,34 +124,84 @@ __device__ __forceinline__ void stage_patch_glds(const bf16raw* __restrict__ xP,
   const int wid = threadIdx.x >> 6;
   const int r_in = lane / SLOTS;
   const int slot = lane % SLOTS;
-  const int Wp = W + 2;
-  const int kh = kt / (3 * CI);
-  const int kw = (kt / CI) % 3;
-  const int ci0 = kt % CI;
 #pragma unroll
   for (int c = wid; c < NCHUNK; c += 4) {
     int lrow = c * CH_ROWS + r_in;
     int row = row0 + lrow;
     if (row >= M) row = M - 1;  // clamp: dead rows read valid garbage, never stored
-    int b = row / HW;
-    int rem = row - b * HW;
-    int oh = rem / W;
-    int ow = rem - oh * W;
+    int b = row / g.OHW;
+    int rem = row - b * g.OHW;
+    int oh = rem / g.OW;
+    int ow = rem - oh * g.OW;
     int sslot = slot ^ (lrow & (SLOTS - 1));
-    const bf16raw* g =
-        xP + (((int64_t)b * (HW / W + 2) + oh + kh) * Wp + ow + kw) * CI + ci0 + sslot * 8;
-    __builtin_amdgcn_global_load_lds((const __attribute__((address_space(1))) unsigned int*)g,
-                                     (__attribute__((address_space(3))) unsigned int*)(lds +
-                                                                                       c * CH_ROWS * BKT),
-                                     16, 0, 0);
+    int k0 = kt + sslot * 8;
+    int kh = k0 / g.KWCI;
+    int r2 = k0 - kh * g.KWCI;
+    int kw = r2 / g.CI;
+    int ci = r2 - kw * g.CI;
+    const bf16raw* gp =
+        xP + (((int64_t)b * g.Hp + oh + kh) * g.Wp + ow + kw) * g.CI + ci;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)gp,
+        (__attribute__((address_space(3))) unsigned int*)(lds + c * CH_ROWS * BKT), 16, 0, 0);
   }
 }
 
-// B: Wmat rows (k-contiguous) — standard glds with the same swizzled image.
+// K-tail chunk: per-granule register loads with zero fill past Kreal, same
+// swizzled LDS image.
+template <int ROWS, int BKT>
+__device__ __forceinline__ void stage_patch_tail(const bf16raw* __restrict__ xP,
+                                                 bf16raw* __restrict__ lds, int row0, int M,
+                                                 const ConvGeom g, int kt) {
+  constexpr int SLOTS = BKT / 8;
+  constexpr int GRANULES = ROWS * SLOTS;
+  constexpr int PER_T = GRANULES / 256;
+  const int t = threadIdx.x;
+#pragma unroll
+  for (int i = 0; i < PER_T; ++i) {
+    int idx = t * PER_T + i;
+    int lrow = idx / SLOTS;
+    int slot = idx % SLOTS;
+    int row = row0 + lrow;
+    if (row >= M) row = M - 1;
+    int b = row / g.OHW;
+    int rem = row - b * g.OHW;
+    int oh = rem / g.OW;
+    int ow = rem - oh * g.OW;
+    int k0 = kt + slot * 8;
+    bf16raw* dst = lds + lrow * BKT + ((slot ^ (lrow & (SLOTS - 1))) << 3);
+    if (k0 + 8 <= g.Kreal) {
+      int kh = k0 / g.KWCI;
+      int r2 = k0 - kh * g.KWCI;
+      int kw = r2 / g.CI;
+      int ci = r2 - kw * g.CI;
+      *(shortx8*)dst = *(const shortx8*)(xP + (((int64_t)b * g.Hp + oh + kh) * g.Wp + ow + kw) *
+                                                  g.CI + ci);
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        int k = k0 + j;
+        bf16raw v = 0;
+        if (k < g.Kreal) {
+          int kh = k / g.KWCI;
+          int r2 = k - kh * g.KWCI;
+          int kw = r2 / g.CI;
+          int ci = r2 - kw * g.CI;
+          v = xP[(((int64_t)b * g.Hp + oh + kh) * g.Wp + ow + kw) * g.CI + ci];
+        }
+        dst[j] = v;
+      }
+    }
+  }
+}
+
+// B: Wmat rows (k-contiguous, column-padded to Kpad) — standard glds with
+// the same swizzled image; rows clamp to N-1 (finite garbage, masked at
+// the C-write) so N need not be a tile multiple.
 template <int ROWS, int BKT>
 __device__ __forceinline__ void stage_w_glds(const bf16raw* __restrict__ src,
-                                             bf16raw* __restrict__ lds, int row0, int64_t srow,
-                                             int kt) {
+                                             bf16raw* __restrict__ lds, int row0, int N,
+                                             int64_t srow, int kt) {
   constexpr int SLOTS = BKT / 8;
   constexpr int CH_ROWS = 64 / SLOTS;
   constexpr int NCHUNK = ROWS / CH_ROWS;
@@ -150,19 +212,20 @@ __device__ __forceinline__ void stage_w_glds(const bf16raw* __restrict__ src,
 #pragma unroll
   for (int c = wid; c < NCHUNK; c += 4) {
     int row = c * CH_ROWS + r_in;
+    int grow = row0 + row;
+    if (grow >= N) grow = N - 1;
     int sslot = slot ^ (row & (SLOTS - 1));
-    const bf16raw* g = src + (int64_t)(row0 + row) * srow + kt + sslot * 8;
-    __builtin_amdgcn_global_load_lds((const __attribute__((address_space(1))) unsigned int*)g,
-                                     (__attribute__((address_space(3))) unsigned int*)(lds +
-                                                                                       c * CH_ROWS * BKT),
-                                     16, 0, 0);
+    const bf16raw* g = src + (int64_t)grow * srow + kt + sslot * 8;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)g,
+        (__attribute__((address_space(3))) unsigned int*)(lds + c * CH_ROWS * BKT), 16, 0, 0);
   }
 }
 
 template <int WR, int WC, bool RELU, bool BIAS>
-__global__ __launch_bounds__(WR * WC * 64, 2) void conv3x3s1_fwd_kernel(
+__global__ __launch_bounds__(WR * WC * 64, 2) void conv_implicit_fwd_kernel(
     const bf16raw* __restrict__ xP, const bf16raw* __restrict__ wmat, bf16raw* __restrict__ y,
-    const float* __restrict__ bias, int M, int N, int K, int HW, int W, int CI) {
+    const float* __restrict__ bias, int M, int N, int Kpad, ConvGeom g) {
   constexpr int BMt = WR * 64;
   constexpr int BNt = WC * 64;
   constexpr int BKT = (WC == 1) ? 32 : 64;
@@ -189,18 +252,21 @@ __global__ __launch_bounds__(WR * WC * 64, 2) void conv3x3s1_fwd_kernel(
 
   floatx4 acc[4][4] = {};
 
-#define IC_STAGE(bufi, kt)                                            \
-  do {                                                                \
-    stage_patch_glds<BMt, BKT>(xP, As[bufi], m0, M, HW, W, CI, kt);   \
-    stage_w_glds<BNt, BKT>(wmat, Bs[bufi], n0, K, kt);              \
+#define IC_STAGE(bufi, kt)                                                 \
+  do {                                                                     \
+    if ((kt) + BKT <= g.Kreal)                                             \
+      stage_patch_glds<BMt, BKT>(xP, As[bufi], m0, M, g, kt);              \
+    else                                                                   \
+      stage_patch_tail<BMt, BKT>(xP, As[bufi], m0, M, g, kt);              \
+    stage_w_glds<BNt, BKT>(wmat, Bs[bufi], n0, N, Kpad, kt);               \
   } while (0)
 
   int buf = 0;
   IC_STAGE(0, 0);
   __syncthreads();
 
-  for (int kt = 0; kt < K; kt += BKT) {
-    if (kt + BKT < K) IC_STAGE(buf ^ 1, kt + BKT);
+  for (int kt = 0; kt < Kpad; kt += BKT) {
+    if (kt + BKT < Kpad) IC_STAGE(buf ^ 1, kt + BKT);
 
 #pragma unroll
     for (int sub = 0; sub < SUBS; ++sub) {
@@ -231,7 +297,7 @@ __global__ __launch_bounds__(WR * WC * 64, 2) void conv3x3s1_fwd_kernel(
 #undef IC_STAGE
 
   // bf16 epilogue via per-wave LDS transpose (one contiguous 16 B store per
-  // lane) — same layout trick as gemm.hip's bf16 epilogue.
+  // lane) — same layout trick as gemm.hip's bf16 epilogue.  N % 8 == 0.
   const int m_base = m0 + wr * 64;
   const int n_base = n0 + wc * 64;
   __syncthreads();
@@ -270,12 +336,21 @@ __global__ __launch_bounds__(WR * WC * 64, 2) void conv3x3s1_fwd_kernel(
   }
 }
 
-extern "C" hipError_t launch_conv3x3s1_fwd(const bf16raw* xP, const bf16raw* wmat,
-                                           const float* bias, bf16raw* y, int B, int H, int W,
-                                           int CRUN, int N, int relu, hipStream_t stream) {
-  const int M = B * H * W;
-  const int K = 9 * CRUN;
-  const int HW = H * W;
+extern "C" hipError_t launch_conv_implicit_fwd(const bf16raw* xP, const bf16raw* wmat,
+                                               const float* bias, bf16raw* y, int B, int Hp,
+                                               int Wp, int CI, int KH, int KW, int N, int Kpad,
+                                               int relu, hipStream_t stream) {
+  const int OH = Hp - KH + 1, OW = Wp - KW + 1;
+  const int M = B * OH * OW;
+  ConvGeom g;
+  g.OW = OW;
+  g.OHW = OH * OW;
+  g.Hp = Hp;
+  g.Wp = Wp;
+  g.CI = CI;
+  g.KW = KW;
+  g.KWCI = KW * CI;
+  g.Kreal = KH * KW * CI;
   const bool narrow = (N <= 64);
   const int bm = narrow ? 256 : 128, bn = narrow ? 64 : 128;
   dim3 grid((unsigned)ceil_div_i64(M, bm), (unsigned)ceil_div_i64(N, bn), 1);
@@ -284,11 +359,11 @@ extern "C" hipError_t launch_conv3x3s1_fwd(const bf16raw* xP, const bf16raw* wma
 #define IC_DISPATCH(RELUV, BIASV)                                                           \
   do {                                                                                      \
     if (narrow)                                                                             \
-      conv3x3s1_fwd_kernel<4, 1, RELUV, BIASV><<<grid, block, 0, stream>>>(                 \
-          xP, wmat, y, bias, M, N, K, HW, W, CRUN);                                         \
+      conv_implicit_fwd_kernel<4, 1, RELUV, BIASV><<<grid, block, 0, stream>>>(             \
+          xP, wmat, y, bias, M, N, Kpad, g);                                                \
     else                                                                                    \
-      conv3x3s1_fwd_kernel<2, 2, RELUV, BIASV><<<grid, block, 0, stream>>>(                 \
-          xP, wmat, y, bias, M, N, K, HW, W, CRUN);                                         \
+      conv_implicit_fwd_kernel<2, 2, RELUV, BIASV><<<grid, block, 0, stream>>>(             \
+          xP, wmat, y, bias, M, N, Kpad, g);                                                \
   } while (0)
 
   if (relu && bias) IC_DISPATCH(true, true);
@@ -301,14 +376,11 @@ extern "C" hipError_t launch_conv3x3s1_fwd(const bf16raw* xP, const bf16raw* wma
 }
 
 // ---------------------------------------------------------------------------
-// wgrad: dW[CO, 9*CI] = dz^T @ patches(xP), split-K over the batch rows.
-// T14 register pipeline (both operands "transposing"): A = dz^T staged as in
-// gemm.hip; B rows are patch slices read straight from xP — the col matrix's
-// 9x-duplicated stream becomes L2-served re-reads of the padded tensor.
+// wgrad: dW[CO, KH*KW*CI] = dz^T @ patches(xP), split-K over batch rows.
+// T14 register pipeline; B rows are patch slices read straight from xP.
 // ---------------------------------------------------------------------------
 
 // A slab: [128 m-rows][32 k] of dz^T == dz rows kt..kt+31, cols m0..m0+127.
-// (identical geometry to gemm.hip stage_load srow==1; 16 regs per thread)
 __device__ __forceinline__ void wg_stage_a(const bf16raw* __restrict__ dz, int m0, int CO,
                                            int kt, int kmax, bf16raw* __restrict__ regs, int t) {
   int k = t >> 3;
@@ -332,26 +404,25 @@ __device__ __forceinline__ void wg_stage_a(const bf16raw* __restrict__ dz, int m
   }
 }
 
-// B slab: [128 n-cols][32 k] — n = (kh,kw,ci) patch coordinate, k = batch
-// row r.  Thread covers 16 consecutive n at fixed r; its 16-n span lies
-// inside one ci-run (runs are CI-aligned, CI % 64 == 0, n0 16-aligned).
+// B slab: [128 n-cols][32 k] — n = (kh,kw,ci), k = batch row.  A thread's
+// 16-n span stays inside one ci-run (CI % 16 == 0, n0 16-aligned).
 __device__ __forceinline__ void wg_stage_b(const bf16raw* __restrict__ xP, int n0, int Nmax,
-                                           int kt, int kmax, int HW, int W, int CI,
+                                           int kt, int kmax, const ConvGeom g,
                                            bf16raw* __restrict__ regs, int t) {
   int k = t >> 3;
   int r0 = (t & 7) * 16;
   int gk = kt + k;  // batch row
-  if (gk < kmax && n0 + r0 < Nmax) {
-    int b = gk / HW;
-    int rem = gk - b * HW;
-    int oh = rem / W;
-    int ow = rem - oh * W;
-    int n = n0 + r0;
-    int kh = n / (3 * CI);
-    int kw = (n / CI) % 3;
-    int ci = n % CI;
-    const bf16raw* sp =
-        xP + (((int64_t)b * (HW / W + 2) + oh + kh) * (W + 2) + ow + kw) * CI + ci;
+  int n = n0 + r0;
+  if (gk < kmax && n < Nmax) {
+    int b = gk / g.OHW;
+    int rem = gk - b * g.OHW;
+    int oh = rem / g.OW;
+    int ow = rem - oh * g.OW;
+    int kh = n / g.KWCI;
+    int r2 = n - kh * g.KWCI;
+    int kw = r2 / g.CI;
+    int ci = r2 - kw * g.CI;
+    const bf16raw* sp = xP + (((int64_t)b * g.Hp + oh + kh) * g.Wp + ow + kw) * g.CI + ci;
     *(shortx8*)regs = *(const shortx8*)sp;
     *(shortx8*)(regs + 8) = *(const shortx8*)(sp + 8);
   } else {
@@ -370,9 +441,9 @@ __device__ __forceinline__ void wg_stage_write(bf16raw* __restrict__ lds,
 }
 
 template <bool SLAB>
-__global__ __launch_bounds__(256, 2) void conv3x3s1_wgrad_kernel(
+__global__ __launch_bounds__(256, 2) void conv_implicit_wgrad_kernel(
     const bf16raw* __restrict__ dz, const bf16raw* __restrict__ xP, float* __restrict__ out,
-    int CO, int N, int K, int HW, int W, int CI, int k_per_split) {
+    int CO, int N, int K, ConvGeom g, int k_per_split) {
   constexpr int BKT = 64;
   constexpr int LP = BKT + IC_LDS_PAD;
   __shared__ bf16raw As[128 * LP];
@@ -406,8 +477,8 @@ __global__ __launch_bounds__(256, 2) void conv3x3s1_wgrad_kernel(
   do {                                                                 \
     wg_stage_a(dz, m0, CO, (kt), k_end, rA, t);                        \
     wg_stage_a(dz, m0, CO, (kt) + 32, k_end, rA + 16, t);              \
-    wg_stage_b(xP, n0, N, (kt), k_end, HW, W, CI, rB, t);              \
-    wg_stage_b(xP, n0, N, (kt) + 32, k_end, HW, W, CI, rB + 16, t);    \
+    wg_stage_b(xP, n0, N, (kt), k_end, g, rB, t);                      \
+    wg_stage_b(xP, n0, N, (kt) + 32, k_end, g, rB + 16, t);            \
   } while (0)
 
   if (k_begin < k_end) WG_LOAD(k_begin);
@@ -466,8 +537,7 @@ __global__ __launch_bounds__(256, 2) void conv3x3s1_wgrad_kernel(
   }
 }
 
-// Combine slabs: dw[m,n] += sum_z ws[z,m,n] (no ones column here — conv
-// weights carry no fused bias; BN provides the shift).
+// Combine slabs: dw[m,n] += sum_z ws[z,m,n].
 __global__ void ic_wgrad_reduce_kernel(const float* __restrict__ ws, float* __restrict__ dw,
                                        int64_t MN, int zs) {
   int64_t i0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 4;
@@ -494,12 +564,22 @@ __global__ void ic_wgrad_reduce_kernel(const float* __restrict__ ws, float* __re
   }
 }
 
-extern "C" hipError_t launch_conv3x3s1_wgrad(const bf16raw* dz, const bf16raw* xP, float* dw,
-                                             int CO, int B, int H, int W, int CI, int splitk,
-                                             float* ws, hipStream_t stream) {
-  const int N = 9 * CI;
-  const int K = B * H * W;
-  const int HW = H * W;
+extern "C" hipError_t launch_conv_implicit_wgrad(const bf16raw* dz, const bf16raw* xP, float* dw,
+                                                 int CO, int B, int Hp, int Wp, int CI, int KH,
+                                                 int KW, int splitk, float* ws,
+                                                 hipStream_t stream) {
+  const int OH = Hp - KH + 1, OW = Wp - KW + 1;
+  const int N = KH * KW * CI;
+  const int K = B * OH * OW;
+  ConvGeom g;
+  g.OW = OW;
+  g.OHW = OH * OW;
+  g.Hp = Hp;
+  g.Wp = Wp;
+  g.CI = CI;
+  g.KW = KW;
+  g.KWCI = KW * CI;
+  g.Kreal = N;
   if (splitk < 1) splitk = 1;
   int kps = K, zs = 1;
   if (splitk > 1) {
@@ -508,22 +588,22 @@ extern "C" hipError_t launch_conv3x3s1_wgrad(const bf16raw* dz, const bf16raw* x
   }
   dim3 grid((unsigned)ceil_div_i64(CO, 128), (unsigned)ceil_div_i64(N, 128), (unsigned)zs);
   if (ws != nullptr) {
-    conv3x3s1_wgrad_kernel<true><<<grid, dim3(256), 0, stream>>>(dz, xP, ws, CO, N, K, HW, W, CI,
-                                                                 kps);
+    conv_implicit_wgrad_kernel<true><<<grid, dim3(256), 0, stream>>>(dz, xP, ws, CO, N, K, g,
+                                                                     kps);
     HIP_CHECK_LAUNCH();
     int64_t MN = (int64_t)CO * N;
     int64_t rg = ceil_div_i64(MN, 1024);
     if (rg > 2048) rg = 2048;
     ic_wgrad_reduce_kernel<<<dim3((unsigned)rg), dim3(256), 0, stream>>>(ws, dw, MN, zs);
   } else {
-    conv3x3s1_wgrad_kernel<false><<<grid, dim3(256), 0, stream>>>(dz, xP, dw, CO, N, K, HW, W,
-                                                                  CI, kps);
+    conv_implicit_wgrad_kernel<false><<<grid, dim3(256), 0, stream>>>(dz, xP, dw, CO, N, K, g,
+                                                                      kps);
   }
   HIP_CHECK_LAUNCH();
   return hipSuccess;
 }
 
-extern "C" int conv3x3s1_wgrad_slices(int K, int splitk) {
+extern "C" int conv_implicit_wgrad_slices(int K, int splitk) {
   if (splitk < 1) splitk = 1;
   if (splitk == 1) return 1;
   int kps = (int)ceil_div_i64(ceil_div_i64(K, splitk), 64) * 64;

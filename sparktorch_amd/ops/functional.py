@@ -239,54 +239,79 @@ class _Conv2dNHWCFn(torch.autograd.Function):
         return dx, dw, db, None, None, None
 
 
-class _Conv3x3s1NHWCFn(torch.autograd.Function):
-    """Implicit-GEMM 3x3 stride-1 pad-1 conv (conv_implicit.hip): no col
-    matrix at all.  Forward stages x-patches straight from a zero-padded
-    NHWC tensor; wgrad reads the same padded tensor (L2-served re-reads
-    instead of a 9x-duplicated col stream); dgrad runs the SAME forward
-    kernel on pad(dz) with flipped/transposed weights — no col2im scatter.
-    Also an 8x activation-memory saving: xP (1.07x of x) is saved for
-    backward instead of the 9x col matrix."""
+def _pad64(t2d):
+    """Zero-pad a [rows, K] bf16 matrix's K to a multiple of 64 (the implicit
+    kernel's LDS chunk) so the weight glds never reads past a row."""
+    K = t2d.shape[1]
+    Kp = (K + 63) // 64 * 64
+    return t2d if Kp == K else F.pad(t2d, (0, Kp - K))
+
+
+class _ConvImplicitNHWCFn(torch.autograd.Function):
+    """Implicit-GEMM stride-1 conv (conv_implicit.hip): no col matrix at
+    all.  Forward stages x-patches straight from a zero-ring-padded NHWC
+    tensor; wgrad reads the same padded tensor (L2-served re-reads instead
+    of a KH*KW-duplicated col stream); dgrad runs the SAME forward kernel
+    on pad(dz, KH-1-p) with flipped/transposed weights — no dcol, no
+    col2im scatter.  Also a KH*KW-fold activation-memory saving: xP is
+    saved for backward instead of the col matrix.
+    Eligible: stride 1, ph == pw <= KH-1, KH == KW, CI % 16 == 0, CO % 8 == 0."""
 
     @staticmethod
-    def forward(ctx, x, w, b, relu):
+    def forward(ctx, x, w, b, padding, relu):
         ext = ops.ext()
         B, H, W, CI = x.shape
-        CO = w.shape[0]
-        w2d = w.permute(0, 2, 3, 1).reshape(CO, 9 * CI)
-        w2d = w2d.to(torch.bfloat16).contiguous()
-        xP = ext.pad_nhwc(x)
-        y2d = ext.conv3x3s1_fwd(xP, w2d, b, relu)
+        CO, _, KH, KW = w.shape
+        p = padding[0]
+        K = KH * KW * CI
+        w2d = w.permute(0, 2, 3, 1).reshape(CO, K).to(torch.bfloat16).contiguous()
+        xP = ext.pad_nhwc(x, p) if p > 0 else x
+        y2d = ext.conv_implicit_fwd(xP, _pad64(w2d).contiguous(), b, KH, KW, relu)
+        OH, OW = H + 2 * p - KH + 1, W + 2 * p - KW + 1
         ctx.save_for_backward(xP, w2d, y2d)
-        ctx.meta = (B, CI, H, W, CO, relu, b is not None)
+        ctx.meta = (B, CI, H, W, CO, KH, KW, p, OH, OW, relu, b is not None)
         ctx.b_ref = b
-        return y2d.view(B, H, W, CO)
+        return y2d.view(B, OH, OW, CO)
 
     @staticmethod
     def backward(ctx, dy):
         ext = ops.ext()
         xP, w2d, y2d = ctx.saved_tensors
-        B, CI, H, W, CO, relu, has_bias = ctx.meta
-        dy2d = dy.reshape(B * H * W, CO).contiguous()
+        B, CI, H, W, CO, KH, KW, p, OH, OW, relu, has_bias = ctx.meta
+        dy2d = dy.reshape(B * OH * OW, CO).contiguous()
         if dy2d.dtype != torch.bfloat16:
             dy2d = dy2d.to(torch.bfloat16)
         dz = ext.relu_bwd(dy2d, y2d) if relu else dy2d
 
         dw = db = None
         if ctx.needs_input_grad[1]:
-            sk = _choose_splitk(dz.shape[0], CO, 9 * CI)
+            sk = _choose_splitk(dz.shape[0], CO, KH * KW * CI)
             slab = CO >= 512  # measured: slab combine beats atomics on the l4 shape
-            dwp = ext.conv3x3s1_wgrad(dz, xP, sk, slab)
-            dw = dwp.reshape(CO, 3, 3, CI).permute(0, 3, 1, 2)
+            dwp = ext.conv_implicit_wgrad(dz, xP, KH, KW, sk, slab)
+            dw = dwp.reshape(CO, KH, KW, CI).permute(0, 3, 1, 2)
         if has_bias and ctx.needs_input_grad[2]:
             db = ext.bias_grad(dz)
 
         dx = None
         if ctx.needs_input_grad[0]:
-            wf = ext.flip_w2d(w2d, CI)
-            dzP = ext.pad_nhwc(dz.view(B, H, W, CO))
-            dx = ext.conv3x3s1_fwd(dzP, wf, None, False).view(B, H, W, CI)
-        return dx, dw, db, None
+            wf = ext.flip_w2d(w2d, CI, KH * KW)
+            ring = KH - 1 - p
+            dzv = dz.view(B, OH, OW, CO)
+            dzP = ext.pad_nhwc(dzv, ring) if ring > 0 else dzv
+            dx = ext.conv_implicit_fwd(dzP, _pad64(wf).contiguous(), None, KH, KW, False)
+            dx = dx.view(B, H, W, CI)
+        return dx, dw, db, None, None
+
+
+def _implicit_eligible(CI, CO, KH, KW, stride, padding):
+    return (
+        tuple(stride) == (1, 1)
+        and KH == KW
+        and padding[0] == padding[1]
+        and padding[0] <= KH - 1
+        and CI % 16 == 0
+        and CO % 8 == 0
+    )
 
 
 def hip_conv2d_nhwc(x, weight, bias=None, stride=(1, 1), padding=(0, 0), relu=False):
@@ -295,14 +320,8 @@ def hip_conv2d_nhwc(x, weight, bias=None, stride=(1, 1), padding=(0, 0), relu=Fa
         if x.dtype != torch.bfloat16:
             x = x.to(torch.bfloat16)
         CO, CI_w, KH, KW = weight.shape
-        if (
-            (KH, KW) == (3, 3)
-            and tuple(stride) == (1, 1)
-            and tuple(padding) == (1, 1)
-            and x.shape[3] % 64 == 0
-            and CO % 64 == 0
-        ):
-            return _Conv3x3s1NHWCFn.apply(x.contiguous(), weight, bias, relu)
+        if _implicit_eligible(x.shape[3], CO, KH, KW, stride, padding):
+            return _ConvImplicitNHWCFn.apply(x.contiguous(), weight, bias, tuple(padding), relu)
         return _Conv2dNHWCFn.apply(x.contiguous(), weight, bias, stride, padding, relu)
     y = F.conv2d(x.permute(0, 3, 1, 2), weight, bias, stride=stride, padding=padding)
     y = F.relu(y) if relu else y

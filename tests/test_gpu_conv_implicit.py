@@ -1,7 +1,9 @@
-"""Numerics for the implicit-GEMM 3x3 s1 conv vs plain fp32 torch.
+"""Numerics for the implicit-GEMM stride-1 conv vs plain fp32 torch.
 
 Every op compares against an fp32 torch reference computed on the
-bf16-quantized inputs the kernels actually see.
+bf16-quantized inputs the kernels actually see.  Shapes cover the ResNet
+3x3 p1 layers, the MNIST-CNN 3x3 p0 CI=16 layer (K % 64 != 0 exercises the
+tail stage), a 5x5 case, and partial M/N tiles.
 """
 
 import pytest
@@ -14,7 +16,7 @@ if not torch.cuda.is_available():  # collected but skipped off-GPU
     pytest.skip("needs MI355X", allow_module_level=True)
 
 from sparktorch_amd import ops
-from sparktorch_amd.ops.functional import _Conv3x3s1NHWCFn, hip_conv2d_nhwc
+from sparktorch_amd.ops.functional import _ConvImplicitNHWCFn, hip_conv2d_nhwc
 
 DEV = "cuda:0"
 
@@ -27,50 +29,51 @@ def _nhwc(x_nchw):
     return x_nchw.permute(0, 2, 3, 1).contiguous()
 
 
+# B, CI, CO, H, W, KH, pad
 SHAPES = [
-    # B, CI, CO, H, W  (l1/l2-class + a misaligned-M case)
-    (2, 64, 64, 56, 56),
-    (2, 128, 128, 28, 28),
-    (1, 64, 128, 14, 14),
-    (3, 64, 64, 7, 7),  # M = 147: partial last M tile exercises the clamp
+    (2, 64, 64, 56, 56, 3, 1),    # resnet l1
+    (2, 128, 128, 28, 28, 3, 1),  # resnet l2
+    (1, 64, 128, 14, 14, 3, 1),   # partial M tile
+    (3, 64, 64, 7, 7, 3, 1),      # M = 147: clamp path
+    (4, 16, 32, 24, 24, 3, 0),    # MNIST conv2 (K = 144: tail stage, p0)
+    (2, 16, 32, 20, 20, 5, 2),    # 5x5 p2 (K = 400: tail)
+    (2, 64, 64, 16, 16, 1, 0),    # 1x1 as implicit GEMM
 ]
 
 
-def test_pad_nhwc():
+def test_pad_nhwc_rings():
     x = torch.randn(2, 5, 6, 64, device=DEV).to(torch.bfloat16).contiguous()
-    xP = ops.ext().pad_nhwc(x)
-    assert xP.shape == (2, 7, 8, 64)
-    assert torch.equal(xP[:, 1:-1, 1:-1, :], x)
-    assert xP[:, 0].abs().sum().item() == 0
-    assert xP[:, :, 0].abs().sum().item() == 0
-    assert xP[:, -1].abs().sum().item() == 0
-    assert xP[:, :, -1].abs().sum().item() == 0
+    for P in (1, 2):
+        xP = ops.ext().pad_nhwc(x, P)
+        assert xP.shape == (2, 5 + 2 * P, 6 + 2 * P, 64)
+        assert torch.equal(xP[:, P:-P, P:-P, :], x)
+        assert xP[:, :P].abs().sum().item() == 0
+        assert xP[:, :, -P:].abs().sum().item() == 0
 
 
 def test_flip_w2d():
-    CO, CI = 64, 64
-    w = torch.randn(CO, CI, 3, 3, device=DEV)
-    w2d = w.permute(0, 2, 3, 1).reshape(CO, 9 * CI).to(torch.bfloat16).contiguous()
-    wf = ops.ext().flip_w2d(w2d, CI)
-    assert wf.shape == (CI, 9 * CO)
-    # W'[ci, kh, kw, co] == W[co, 2-kh, 2-kw, ci]
-    wf4 = wf.view(CI, 3, 3, CO)
-    w4 = w2d.view(CO, 3, 3, CI)
-    for kh in (0, 1, 2):
-        for kw in (0, 1, 2):
+    CO, CI, KH = 64, 64, 3
+    w = torch.randn(CO, CI, KH, KH, device=DEV)
+    w2d = w.permute(0, 2, 3, 1).reshape(CO, KH * KH * CI).to(torch.bfloat16).contiguous()
+    wf = ops.ext().flip_w2d(w2d, CI, KH * KH)
+    wf4 = wf.view(CI, KH, KH, CO)
+    w4 = w2d.view(CO, KH, KH, CI)
+    for kh in range(KH):
+        for kw in range(KH):
             assert torch.equal(
-                wf4[:, kh, kw, :], w4[:, 2 - kh, 2 - kw, :].t().contiguous()
+                wf4[:, kh, kw, :], w4[:, KH - 1 - kh, KH - 1 - kw, :].t().contiguous()
             )
 
 
-@pytest.mark.parametrize("B,CI,CO,H,W", SHAPES)
-def test_implicit_fwd_matches_torch(B, CI, CO, H, W):
+@pytest.mark.parametrize("B,CI,CO,H,W,KH,pad", SHAPES)
+def test_implicit_fwd_matches_torch(B, CI, CO, H, W, KH, pad):
     torch.manual_seed(0)
     x = torch.randn(B, CI, H, W, device=DEV)
-    w = torch.randn(CO, CI, 3, 3, device=DEV) * 0.1
-    y = hip_conv2d_nhwc(_nhwc(bf(x)), w, None, stride=(1, 1), padding=(1, 1), relu=False)
-    ref = F.conv2d(bf(x).float(), bf(w).float(), None, stride=1, padding=1)
+    w = torch.randn(CO, CI, KH, KH, device=DEV) * 0.1
+    y = hip_conv2d_nhwc(_nhwc(bf(x)), w, None, stride=(1, 1), padding=(pad, pad), relu=False)
+    ref = F.conv2d(bf(x).float(), bf(w).float(), None, stride=1, padding=pad)
     ref = _nhwc(ref)
+    assert y.shape == ref.shape
     err = (y.float() - ref).abs().max().item()
     scale = ref.abs().max().item() + 1e-6
     assert err / scale < 0.03, (err, scale)
@@ -88,21 +91,21 @@ def test_implicit_fwd_relu_bias():
     assert err < 0.05, err
 
 
-@pytest.mark.parametrize("B,CI,CO,H,W", SHAPES[:3])
-def test_implicit_backward_matches_torch(B, CI, CO, H, W):
+@pytest.mark.parametrize("B,CI,CO,H,W,KH,pad", SHAPES[:3] + SHAPES[4:6])
+def test_implicit_backward_matches_torch(B, CI, CO, H, W, KH, pad):
     torch.manual_seed(2)
     x32 = torch.randn(B, CI, H, W, device=DEV)
-    w32 = torch.randn(CO, CI, 3, 3, device=DEV) * 0.1
+    w32 = torch.randn(CO, CI, KH, KH, device=DEV) * 0.1
 
     xh = _nhwc(bf(x32)).requires_grad_(True)
     wh = w32.clone().requires_grad_(True)
-    y = _Conv3x3s1NHWCFn.apply(xh, wh, None, False)
+    y = _ConvImplicitNHWCFn.apply(xh, wh, None, (pad, pad), False)
     gy = torch.randn_like(y.float()) * 0.1
     y.backward(bf(gy))
 
     xr = bf(x32).float().requires_grad_(True)
     wr = bf(w32).float().requires_grad_(True)
-    yr = F.conv2d(xr, wr, None, stride=1, padding=1)
+    yr = F.conv2d(xr, wr, None, stride=1, padding=pad)
     # gy is NHWC (same layout as y); the torch reference is NCHW
     yr.backward(bf(gy).float().permute(0, 3, 1, 2))
 
@@ -121,9 +124,9 @@ def test_implicit_wgrad_slab_matches_atomic():
     B, CI, CO, H, W = 2, 64, 128, 28, 28
     x = _nhwc(torch.randn(B, CI, H, W, device=DEV).to(torch.bfloat16))
     dz = torch.randn(B * H * W, CO, device=DEV).to(torch.bfloat16).contiguous()
-    xP = ops.ext().pad_nhwc(x)
-    a = ops.ext().conv3x3s1_wgrad(dz, xP, 8, False)
-    b = ops.ext().conv3x3s1_wgrad(dz, xP, 8, True)
+    xP = ops.ext().pad_nhwc(x, 1)
+    a = ops.ext().conv_implicit_wgrad(dz, xP, 3, 3, 8, False)
+    b = ops.ext().conv_implicit_wgrad(dz, xP, 3, 3, 8, True)
     assert torch.allclose(a, b, atol=1e-3, rtol=1e-4)
 
 
@@ -157,5 +160,22 @@ def test_resnet_block_uses_implicit_and_trains():
     tr = SyncTrainer(model, nn.CrossEntropyLoss(), opt, device=DEV, world_size=1)
     x = torch.randn(256, 64 * 8 * 8, device=DEV).to(torch.bfloat16)
     yl = torch.randint(0, 10, (256,), device=DEV)
+    losses = [tr.train_step(x, yl) for _ in range(8)]
+    assert losses[-1] < losses[0]
+
+
+def test_mnist_cnn_fused_trains():
+    """The fused MNIST CNN (conv2 now on the implicit path) still trains."""
+    import torch.nn as nn
+
+    from sparktorch_amd.ops.modules import MnistCNNFused
+    from sparktorch_amd.parallel.sync import SyncTrainer
+
+    torch.manual_seed(5)
+    model = MnistCNNFused()
+    opt = torch.optim.Adam(model.parameters(), lr=1e-3)
+    tr = SyncTrainer(model, nn.CrossEntropyLoss(), opt, device=DEV, world_size=1)
+    x = torch.randn(4096, 784, device=DEV).to(torch.bfloat16)
+    yl = torch.randint(0, 10, (4096,), device=DEV)
     losses = [tr.train_step(x, yl) for _ in range(8)]
     assert losses[-1] < losses[0]
