@@ -23,11 +23,13 @@ hipError_t launch_bias_grad(const bf16raw*, float*, int, int, float*, int,
 hipError_t launch_small_wgrad(const bf16raw*, const bf16raw*, float*, float*, int, int64_t, int,
                               int, hipStream_t);
 hipError_t launch_pad_nhwc(const bf16raw*, bf16raw*, int, int, int, int, int, hipStream_t);
+hipError_t launch_s2d_stem(const bf16raw*, bf16raw*, int, int, int, int, hipStream_t);
 hipError_t launch_flip_w2d(const bf16raw*, bf16raw*, int, int, int, hipStream_t);
 hipError_t launch_conv_implicit_fwd(const bf16raw*, const bf16raw*, const float*, bf16raw*, int,
-                                    int, int, int, int, int, int, int, int, hipStream_t);
+                                    int, int, int, int, int, int, int, int, int, int,
+                                    hipStream_t);
 hipError_t launch_conv_implicit_wgrad(const bf16raw*, const bf16raw*, float*, int, int, int, int,
-                                      int, int, int, int, float*, hipStream_t);
+                                      int, int, int, int, float*, int, int, hipStream_t);
 int conv_implicit_wgrad_slices(int, int);
 hipError_t launch_cast_f64_f32(const double*, float*, int64_t, hipStream_t);
 hipError_t launch_cast_f64_bf16(const double*, bf16raw*, int64_t, hipStream_t);
@@ -175,6 +177,17 @@ at::Tensor pad_nhwc(at::Tensor x, int64_t P) {
   return xP;
 }
 
+at::Tensor s2d_stem(at::Tensor x, int64_t P) {
+  check_gpu_contig(x, at::kBFloat16, "x");
+  TORCH_CHECK(x.dim() == 4 && x.size(3) == 3 && x.size(1) % 2 == 0 && x.size(2) % 2 == 0,
+              "s2d_stem wants [B, even H, even W, 3]");
+  int B = (int)x.size(0), H = (int)x.size(1), W = (int)x.size(2);
+  auto out = at::empty({B, H / 2 + 2 * P, W / 2 + 2 * P, 16}, x.options());
+  CHECK_HIP(launch_s2d_stem((const bf16raw*)x.data_ptr(), (bf16raw*)out.data_ptr(), B, H, W,
+                            (int)P, cur_stream()));
+  return out;
+}
+
 at::Tensor flip_w2d(at::Tensor w2d, int64_t CI, int64_t KHW) {
   check_gpu_contig(w2d, at::kBFloat16, "w2d");
   TORCH_CHECK(w2d.dim() == 2 && w2d.size(1) == KHW * CI, "w2d must be [CO, KH*KW*CI]");
@@ -186,16 +199,19 @@ at::Tensor flip_w2d(at::Tensor w2d, int64_t CI, int64_t KHW) {
 }
 
 at::Tensor conv_implicit_fwd(at::Tensor xP, at::Tensor wmat, c10::optional<at::Tensor> bias,
-                             int64_t KH, int64_t KW, bool relu) {
+                             int64_t KH, int64_t KW, bool relu, int64_t OHo, int64_t OWo) {
   check_gpu_contig(xP, at::kBFloat16, "xP");
   check_gpu_contig(wmat, at::kBFloat16, "wmat");
   int B = (int)xP.size(0), Hp = (int)xP.size(1), Wp = (int)xP.size(2), CI = (int)xP.size(3);
   int OH = Hp - (int)KH + 1, OW = Wp - (int)KW + 1;
+  if (OHo <= 0) OHo = OH;
+  if (OWo <= 0) OWo = OW;
   int N = (int)wmat.size(0);
   int Kpad = (int)wmat.size(1);
   TORCH_CHECK(Kpad % 64 == 0 && Kpad >= KH * KW * CI, "wmat k-dim must be K padded to x64");
   TORCH_CHECK(CI % 16 == 0 && N % 8 == 0, "implicit conv wants CI%16==0, N%8==0");
-  auto y = at::empty({(int64_t)B * OH * OW, N}, xP.options());
+  TORCH_CHECK(OHo <= OH && OWo <= OW, "compact output dims exceed GEMM output");
+  auto y = at::empty({(int64_t)B * OHo * OWo, N}, xP.options());
   const float* bp = nullptr;
   if (bias.has_value()) {
     check_gpu_contig(*bias, at::kFloat, "bias");
@@ -204,18 +220,20 @@ at::Tensor conv_implicit_fwd(at::Tensor xP, at::Tensor wmat, c10::optional<at::T
   CHECK_HIP(launch_conv_implicit_fwd((const bf16raw*)xP.data_ptr(),
                                      (const bf16raw*)wmat.data_ptr(), bp, (bf16raw*)y.data_ptr(),
                                      B, Hp, Wp, CI, (int)KH, (int)KW, N, Kpad, relu ? 1 : 0,
-                                     cur_stream()));
+                                     (int)OHo, (int)OWo, cur_stream()));
   return y;
 }
 
 at::Tensor conv_implicit_wgrad(at::Tensor dz, at::Tensor xP, int64_t KH, int64_t KW,
-                               int64_t splitk, bool slab) {
+                               int64_t splitk, bool slab, int64_t OHo, int64_t OWo) {
   check_gpu_contig(dz, at::kBFloat16, "dz");
   check_gpu_contig(xP, at::kBFloat16, "xP");
   int B = (int)xP.size(0), Hp = (int)xP.size(1), Wp = (int)xP.size(2), CI = (int)xP.size(3);
   int OH = Hp - (int)KH + 1, OW = Wp - (int)KW + 1;
+  if (OHo <= 0) OHo = OH;
+  if (OWo <= 0) OWo = OW;
   int CO = (int)dz.size(1);
-  TORCH_CHECK(dz.size(0) == (int64_t)B * OH * OW, "dz rows must be B*OH*OW");
+  TORCH_CHECK(dz.size(0) == (int64_t)B * OHo * OWo, "dz rows must be B*OHo*OWo");
   int64_t N = KH * KW * CI;
   auto dw = at::zeros({CO, N}, xP.options().dtype(at::kFloat));
   float* wsp = nullptr;
@@ -228,7 +246,7 @@ at::Tensor conv_implicit_wgrad(at::Tensor dz, at::Tensor xP, int64_t KH, int64_t
   CHECK_HIP(launch_conv_implicit_wgrad((const bf16raw*)dz.data_ptr(),
                                        (const bf16raw*)xP.data_ptr(), dw.data_ptr<float>(), CO,
                                        B, Hp, Wp, CI, (int)KH, (int)KW, (int)splitk, wsp,
-                                       cur_stream()));
+                                       (int)OHo, (int)OWo, cur_stream()));
   return dw;
 }
 
@@ -863,9 +881,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("relu_bwd", &relu_bwd, "dz = dy * (y>0)");
   m.def("bias_grad", &bias_grad, "column-sum of dz");
   m.def("pad_nhwc", &pad_nhwc, "zero-pad ring (width P) for implicit conv");
+  m.def("s2d_stem", &s2d_stem, "fused space-to-depth + channel pad + ring pad for the 7x7 s2 stem");
   m.def("flip_w2d", &flip_w2d, "dgrad weight transform W[co,k,ci] -> W'[ci,rev(k),co]");
-  m.def("conv_implicit_fwd", &conv_implicit_fwd, "implicit-GEMM s1 conv forward / dgrad");
-  m.def("conv_implicit_wgrad", &conv_implicit_wgrad, "implicit-GEMM s1 conv weight grad");
+  m.def("conv_implicit_fwd", &conv_implicit_fwd, "implicit-GEMM s1 conv forward / dgrad",
+        py::arg("xP"), py::arg("wmat"), py::arg("bias"), py::arg("KH"), py::arg("KW"),
+        py::arg("relu"), py::arg("OHo") = -1, py::arg("OWo") = -1);
+  m.def("conv_implicit_wgrad", &conv_implicit_wgrad, "implicit-GEMM s1 conv weight grad",
+        py::arg("dz"), py::arg("xP"), py::arg("KH"), py::arg("KW"), py::arg("splitk"),
+        py::arg("slab"), py::arg("OHo") = -1, py::arg("OWo") = -1);
   m.def("cast_f64_f32", &cast_f64_f32);
   m.def("cast_f64_bf16", &cast_f64_bf16, "one-pass device Vector pack cast");
   m.def("cast_f32_bf16", &cast_f32_bf16);

@@ -36,13 +36,18 @@ __device__ __forceinline__ int ic_swz(int row, int col) {
 
 // Patch geometry: maps (output row, k) -> padded-input address.
 struct ConvGeom {
-  int OW;    // output width (stride 1: OW = Wp - KW + 1)
-  int OHW;   // OH * OW
+  int OW;    // GEMM-space output width (stride 1: OW = Wp - KW + 1)
+  int OHW;   // GEMM-space OH * OW
   int Hp, Wp;  // padded input spatial
   int CI;
   int KW;
   int KWCI;  // KW * CI
   int Kreal; // KH * KW * CI
+  // logical output dims: rows with oh >= OHo or ow >= OWo are dropped at the
+  // C-write (fwd) / read as zero dz (wgrad).  Equal to OH/OW normally; the
+  // stem's symmetric ring over-pads by one row+col and compacts here instead
+  // of a strided slice copy.
+  int OHo, OWo;
 };
 
 // ---------------------------------------------------------------------------
@@ -78,6 +83,48 @@ extern "C" hipError_t launch_pad_nhwc(const bf16raw* x, bf16raw* xP, int B, int 
   int64_t g = ceil_div_i64(total, 256);
   if (g > 8192) g = 8192;
   pad_nhwc_kernel<<<dim3((unsigned)g), dim3(256), 0, stream>>>(x, xP, B, H, W, CI, P);
+  HIP_CHECK_LAUNCH();
+  return hipSuccess;
+}
+
+// Fused stem space-to-depth: x [B,H,W,3] -> out [B, H/2+2P, W/2+2P, 16]
+// with out[b, bh+P, bw+P, (ph*2+pw)*4 + c] = x[b, 2bh+ph, 2bw+pw, c] (c < 3,
+// 4th channel zero) and a zero ring of width P — one pass replaces the
+// channel pad + 6D permute copy + pad_nhwc chain.
+__global__ void s2d_stem_kernel(const bf16raw* __restrict__ x, bf16raw* __restrict__ out, int B,
+                                int H, int W, int P) {
+  const int Hb = H >> 1, Wb = W >> 1;
+  const int Ho = Hb + 2 * P, Wo = Wb + 2 * P;
+  // one thread = one 8-channel granule = one phase row (ph) of a 2x2 block
+  int64_t total = (int64_t)B * Ho * Wo * 2;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total; i += stride) {
+    int ph = (int)(i & 1);
+    int64_t t = i >> 1;
+    int wo = (int)(t % Wo);
+    t /= Wo;
+    int ho = (int)(t % Ho);
+    int b = (int)(t / Ho);
+    shortx8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+    int bh = ho - P, bw = wo - P;
+    if (bh >= 0 && bh < Hb && bw >= 0 && bw < Wb) {
+      const bf16raw* src = x + (((int64_t)b * H + 2 * bh + ph) * W + 2 * bw) * 3;
+#pragma unroll
+      for (int c = 0; c < 3; ++c) {
+        v[c] = (short)src[c];      // pw = 0
+        v[4 + c] = (short)src[3 + c];  // pw = 1
+      }
+    }
+    *(shortx8*)(out + ((((int64_t)b * Ho + ho) * Wo + wo) * 16) + ph * 8) = v;
+  }
+}
+
+extern "C" hipError_t launch_s2d_stem(const bf16raw* x, bf16raw* out, int B, int H, int W, int P,
+                                      hipStream_t stream) {
+  int64_t total = (int64_t)B * (H / 2 + 2 * P) * (W / 2 + 2 * P) * 2;
+  int64_t g = ceil_div_i64(total, 256);
+  if (g > 8192) g = 8192;
+  s2d_stem_kernel<<<dim3((unsigned)g), dim3(256), 0, stream>>>(x, out, B, H, W, P);
   HIP_CHECK_LAUNCH();
   return hipSuccess;
 }
@@ -313,7 +360,15 @@ __global__ __launch_bounds__(WR * WC * 64, 2) void conv_implicit_fwd_kernel(
       for (int r = 0; r < 4; ++r) ep[(kg * 4 + r) * EPAD + ni * 16 + l15] = acc[mi][ni][r];
     __builtin_amdgcn_s_waitcnt(0);
     int m = m_base + mi * 16 + orow;
-    if (m < M) {
+    int64_t mout = m;
+    if (g.OWo != g.OW || g.OHo * g.OW != g.OHW) {
+      int b = m / g.OHW;
+      int rem = m - b * g.OHW;
+      int oh = rem / g.OW;
+      int ow = rem - oh * g.OW;
+      mout = (oh < g.OHo && ow < g.OWo) ? (((int64_t)b * g.OHo + oh) * g.OWo + ow) : -1;
+    }
+    if (m < M && mout >= 0) {
 #pragma unroll
       for (int h = 0; h < 2; ++h) {
         int c0 = oct * 16 + h * 8;
@@ -328,7 +383,7 @@ __global__ __launch_bounds__(WR * WC * 64, 2) void conv_implicit_fwd_kernel(
             if (RELU) val = fmaxf(val, 0.f);
             outp[j] = (short)f32_to_bf16(val);
           }
-          *(shortx8*)(y + (int64_t)m * N + n) = *(const shortx8*)outp;
+          *(shortx8*)(y + mout * N + n) = *(const shortx8*)outp;
         }
       }
     }
@@ -339,7 +394,7 @@ __global__ __launch_bounds__(WR * WC * 64, 2) void conv_implicit_fwd_kernel(
 extern "C" hipError_t launch_conv_implicit_fwd(const bf16raw* xP, const bf16raw* wmat,
                                                const float* bias, bf16raw* y, int B, int Hp,
                                                int Wp, int CI, int KH, int KW, int N, int Kpad,
-                                               int relu, hipStream_t stream) {
+                                               int relu, int OHo, int OWo, hipStream_t stream) {
   const int OH = Hp - KH + 1, OW = Wp - KW + 1;
   const int M = B * OH * OW;
   ConvGeom g;
@@ -351,6 +406,8 @@ extern "C" hipError_t launch_conv_implicit_fwd(const bf16raw* xP, const bf16raw*
   g.KW = KW;
   g.KWCI = KW * CI;
   g.Kreal = KH * KW * CI;
+  g.OHo = OHo > 0 ? OHo : OH;
+  g.OWo = OWo > 0 ? OWo : OW;
   const bool narrow = (N <= 64);
   const int bm = narrow ? 256 : 128, bn = narrow ? 64 : 128;
   dim3 grid((unsigned)ceil_div_i64(M, bm), (unsigned)ceil_div_i64(N, bn), 1);
@@ -382,12 +439,21 @@ extern "C" hipError_t launch_conv_implicit_fwd(const bf16raw* xP, const bf16raw*
 
 // A slab: [128 m-rows][32 k] of dz^T == dz rows kt..kt+31, cols m0..m0+127.
 __device__ __forceinline__ void wg_stage_a(const bf16raw* __restrict__ dz, int m0, int CO,
-                                           int kt, int kmax, bf16raw* __restrict__ regs, int t) {
+                                           int kt, int kmax, const ConvGeom g,
+                                           bf16raw* __restrict__ regs, int t) {
   int k = t >> 3;
   int r0 = (t & 7) * 16;
   int gk = kt + k;
-  if (gk < kmax) {
-    const bf16raw* base = dz + (int64_t)gk * CO;
+  int64_t gkc = gk;
+  if (g.OWo != g.OW || g.OHo * g.OW != g.OHW) {
+    int b = gk / g.OHW;
+    int rem = gk - b * g.OHW;
+    int oh = rem / g.OW;
+    int ow = rem - oh * g.OW;
+    gkc = (oh < g.OHo && ow < g.OWo) ? (((int64_t)b * g.OHo + oh) * g.OWo + ow) : -1;
+  }
+  if (gk < kmax && gkc >= 0) {
+    const bf16raw* base = dz + gkc * CO;
     int rem = CO - m0 - r0;
     if (rem >= 16) {
       const bf16raw* sp = base + m0 + r0;
@@ -475,8 +541,8 @@ __global__ __launch_bounds__(256, 2) void conv_implicit_wgrad_kernel(
 
 #define WG_LOAD(kt)                                                    \
   do {                                                                 \
-    wg_stage_a(dz, m0, CO, (kt), k_end, rA, t);                        \
-    wg_stage_a(dz, m0, CO, (kt) + 32, k_end, rA + 16, t);              \
+    wg_stage_a(dz, m0, CO, (kt), k_end, g, rA, t);                     \
+    wg_stage_a(dz, m0, CO, (kt) + 32, k_end, g, rA + 16, t);           \
     wg_stage_b(xP, n0, N, (kt), k_end, g, rB, t);                      \
     wg_stage_b(xP, n0, N, (kt) + 32, k_end, g, rB + 16, t);            \
   } while (0)
@@ -566,7 +632,7 @@ __global__ void ic_wgrad_reduce_kernel(const float* __restrict__ ws, float* __re
 
 extern "C" hipError_t launch_conv_implicit_wgrad(const bf16raw* dz, const bf16raw* xP, float* dw,
                                                  int CO, int B, int Hp, int Wp, int CI, int KH,
-                                                 int KW, int splitk, float* ws,
+                                                 int KW, int splitk, float* ws, int OHo, int OWo,
                                                  hipStream_t stream) {
   const int OH = Hp - KH + 1, OW = Wp - KW + 1;
   const int N = KH * KW * CI;
@@ -580,6 +646,8 @@ extern "C" hipError_t launch_conv_implicit_wgrad(const bf16raw* dz, const bf16ra
   g.KW = KW;
   g.KWCI = KW * CI;
   g.Kreal = N;
+  g.OHo = OHo > 0 ? OHo : OH;
+  g.OWo = OWo > 0 ? OWo : OW;
   if (splitk < 1) splitk = 1;
   int kps = K, zs = 1;
   if (splitk > 1) {

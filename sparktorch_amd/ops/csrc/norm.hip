@@ -510,10 +510,16 @@ __global__ void bn_bwd_dx_nhwc_kernel(const bf16raw* __restrict__ dy,
   }
 }
 
-// Cached-param dx: dx = a*(g') - b - (x - m)*d per channel with
-// a = gamma*invstd, b = gamma*invstd*sum_dy/N, d = gamma*invstd^2*sum_dyxhat/N
-// (algebraically dx = gamma*is*(g - sdy/N - xhat*sdyx/N)); channel octet is
-// loop-invariant as in bn_apply_nhwc_vec_kernel.
+// Cached-param dx.  Per-channel math folded to FOUR coefficients:
+//   a  = gamma*invstd
+//   d  = gamma*invstd^2*sum_dyxhat/N
+//   b2 = gamma*invstd*sum_dy/N - mean*d        (so dx = a*g - b2 - x*d)
+//   t  = beta - a*mean                          (relu mask: a*x + t > 0)
+// One 256-thread block computes them cooperatively into LDS once (C <= 512
+// on this path; every power-of-two C <= 512 divides 2048 so the grid-stride
+// octet stays loop-invariant), then every thread reads its fixed octet from
+// LDS — the old per-thread 56-scalar-global-load preamble dominated the
+// short layers' runtime (l4-class calls sat ~12x off the 3-stream roofline).
 __global__ void bn_bwd_dx_nhwc_vec_kernel(const bf16raw* __restrict__ dy, int relu,
                                           const bf16raw* __restrict__ x,
                                           const float* __restrict__ mean,
@@ -524,26 +530,33 @@ __global__ void bn_bwd_dx_nhwc_vec_kernel(const bf16raw* __restrict__ dy, int re
                                           const float* __restrict__ dgamma,
                                           bf16raw* __restrict__ dx, int C, int64_t total8,
                                           float inv_count, int train_stats) {
+  __shared__ float sa[512], sb2[512], sd[512], st[512];
+  for (int c = threadIdx.x; c < C; c += blockDim.x) {
+    float is = invstd[c];
+    float ga = gamma[c];
+    float m = mean[c];
+    float aa = ga * is;
+    float dd = 0.f, bb = 0.f;
+    if (train_stats) {
+      dd = aa * is * dgamma[c] * inv_count;
+      bb = aa * dbeta[c] * inv_count - m * dd;
+    }
+    sa[c] = aa;
+    sd[c] = dd;
+    sb2[c] = bb;
+    st[c] = (beta != nullptr ? beta[c] : 0.f) - aa * m;
+  }
+  __syncthreads();
+
   int64_t i0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  int c0 = (int)((i0 << 3) % C);
-  float a[8], b[8], d[8], m[8], is8[8], ga[8], be[8];
+  int c0 = (int)((i0 << 3) & (C - 1));  // C is a power of two on this path
+  float a[8], b2[8], d[8], t[8];
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
-    int c = c0 + j;
-    float is = invstd[c];
-    float gis = gamma[c] * is;
-    a[j] = gis;
-    is8[j] = is;
-    ga[j] = gamma[c];
-    be[j] = beta != nullptr ? beta[c] : 0.f;
-    m[j] = mean[c];
-    if (train_stats) {
-      b[j] = gis * dbeta[c] * inv_count;
-      d[j] = gis * is * dgamma[c] * inv_count;
-    } else {
-      b[j] = 0.f;
-      d[j] = 0.f;
-    }
+    a[j] = sa[c0 + j];
+    b2[j] = sb2[c0 + j];
+    d[j] = sd[c0 + j];
+    t[j] = st[c0 + j];
   }
   for (int64_t i = i0; i < total8; i += (int64_t)gridDim.x * blockDim.x) {
     int64_t off = i << 3;
@@ -553,10 +566,10 @@ __global__ void bn_bwd_dx_nhwc_vec_kernel(const bf16raw* __restrict__ dy, int re
       const shortx8 x8 = *(const shortx8*)(x + off);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        float dxm = bf16_to_f32((bf16raw)x8[j]) - m[j];
+        float xv = bf16_to_f32((bf16raw)x8[j]);
         float g = bf16_to_f32((bf16raw)g8[j]);
-        if (relu && fmaf(ga[j], dxm * is8[j], be[j]) <= 0.f) g = 0.f;
-        float o = train_stats ? (a[j] * g - b[j] - dxm * d[j]) : a[j] * g;
+        if (relu && fmaf(a[j], xv, t[j]) <= 0.f) g = 0.f;
+        float o = fmaf(a[j], g, -fmaf(d[j], xv, b2[j]));
         out[j] = (short)f32_to_bf16(o);
       }
     } else {
@@ -575,7 +588,7 @@ extern "C" hipError_t launch_bn_bwd_dx_nhwc(const bf16raw* dy, const bf16raw* yr
                                             const float* dgamma, bf16raw* dx, int C,
                                             int64_t total, float inv_count, int train_stats,
                                             hipStream_t stream) {
-  if ((C & 7) == 0 && C <= 2048 && (2048 % C) == 0) {
+  if ((C & 7) == 0 && C <= 512 && (C & (C - 1)) == 0) {
     int64_t total8 = total >> 3;
     int64_t g = ceil_div_i64(total8, 256);
     int grid = (int)(g > 8192 ? 8192 : (g < 1 ? 1 : g));

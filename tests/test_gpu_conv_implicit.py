@@ -179,3 +179,32 @@ def test_mnist_cnn_fused_trains():
     yl = torch.randint(0, 10, (4096,), device=DEV)
     losses = [tr.train_step(x, yl) for _ in range(8)]
     assert losses[-1] < losses[0]
+
+
+def test_stem_s2d_matches_torch():
+    """7x7 s2 p3 CI=3 stem via space-to-depth vs fp32 torch (fwd + wgrad)."""
+    from sparktorch_amd.ops.functional import _ConvStemS2DFn
+
+    torch.manual_seed(7)
+    B, H = 2, 32  # any even spatial works; bench uses 224
+    x32 = torch.randn(B, 3, H, H, device=DEV)
+    w32 = torch.randn(64, 3, 7, 7, device=DEV) * 0.1
+
+    xh = _nhwc(bf(x32))
+    wh = w32.clone().requires_grad_(True)
+    y = _ConvStemS2DFn.apply(xh, wh, None, False)
+    ref = F.conv2d(bf(x32).float(), bf(w32).float(), None, stride=2, padding=3)
+    ref = _nhwc(ref)
+    assert y.shape == ref.shape
+    err = (y.float() - ref).abs().max().item()
+    scale = ref.abs().max().item() + 1e-6
+    assert err / scale < 0.03, (err, scale)
+
+    gy = torch.randn_like(y.float()) * 0.1
+    y.backward(bf(gy))
+    wr = bf(w32).float().requires_grad_(True)
+    yr = F.conv2d(bf(x32).float(), wr, None, stride=2, padding=3)
+    yr.backward(bf(gy).float().permute(0, 3, 1, 2))
+    dw_err = (wh.grad - wr.grad).abs().max().item()
+    dw_scale = wr.grad.abs().max().item() + 1e-6
+    assert dw_err / dw_scale < 0.03, (dw_err, dw_scale)
