@@ -339,6 +339,9 @@ at::Tensor linear_dgrad(at::Tensor dz, at::Tensor w) {
 // tiny-output wgrad: batched outer product instead of a ~98%-dead MFMA tile
 static bool small_wgrad_path(int64_t M, int64_t N, int64_t K) {
   // slot kernel: one lane per (co, 8k) cell group; needs aligned K
+  // slot kernel's structural cap is 256 cells (one lane per cell); the wide
+  // cell-per-thread variant measured ISSUE-bound (per-CO duplicate col
+  // loads) and lost to the dead-tile MFMA path — keep the 2048 limit
   return (K & 7) == 0 && N * K <= 2048 && N <= 64 && K <= 512 && M >= 65536;
 }
 static int small_wgrad_slices(int64_t M) {

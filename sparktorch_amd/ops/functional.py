@@ -71,6 +71,10 @@ class _LinearFn(torch.autograd.Function):
         w_direct = w_notify is not None and w.grad is not None
         b_direct = b_notify is not None and b is not None and b.grad is not None
 
+        # NB: the fused dW+db ones column can push N across a tile boundary
+        # (fc2: 257 -> an extra 99.6%-dead N-tile); splitting into wgrad +
+        # bias_grad was measured NEUTRAL end-to-end (the separate column-sum
+        # pass costs what the dead tile does) — keep the single fused launch.
         if ctx.needs_input_grad[1] and w_direct and ctx.has_bias and b_direct:
             # dW and db in ONE MFMA launch (virtual ones column)
             ext.linear_wgrad_bias_into(dz, x, w.grad, b.grad, sk)
