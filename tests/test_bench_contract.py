@@ -63,3 +63,15 @@ def test_graft_entry_contract():
     g.build()  # no-op rebuild when sources unchanged; must not raise on CPU
     from sparktorch_amd import ops
     assert ops.available()
+
+
+def test_fit_contract_cpu():
+    """--mode fit: estimator-path metric with ingest inside the timed region."""
+    out = _run("--mode", "fit", "--steps", "2", "--warmup", "1", "--batch", "512")
+    for k in REQUIRED:
+        assert k in out, k
+    assert out["metric"] == "fit_samples_per_sec"
+    assert out["higher_is_better"] is True
+    assert out["config"]["ingest_included_in_timed_region"] is True
+    assert out["config"]["ingest_ms"] > 0
+    assert out["value"] > 0
