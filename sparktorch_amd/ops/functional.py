@@ -291,8 +291,9 @@ class _ConvImplicitNHWCFn(torch.autograd.Function):
         if ctx.needs_input_grad[1]:
             # measured sweep (tools/sweep_implicit_wgrad.py on MI355X):
             # best split keeps ~3136 batch rows per K-slice across the
-            # resnet shapes; slab combine beats atomics from CO >= 256
-            sk = max(16, min(256, dz.shape[0] // 3136))
+            # resnet shapes (no upper clamp — big batches want more slices);
+            # slab combine beats atomics from CO >= 256
+            sk = max(16, dz.shape[0] // 3136)
             slab = CO >= 256
             dwp = ext.conv_implicit_wgrad(dz, xP, KH, KW, sk, slab)
             dw = dwp.reshape(CO, KH, KW, CI).permute(0, 3, 1, 2)
