@@ -26,10 +26,11 @@ hipError_t launch_pad_nhwc(const bf16raw*, bf16raw*, int, int, int, int, int, hi
 hipError_t launch_s2d_stem(const bf16raw*, bf16raw*, int, int, int, int, hipStream_t);
 hipError_t launch_flip_w2d(const bf16raw*, bf16raw*, int, int, int, hipStream_t);
 hipError_t launch_conv_implicit_fwd(const bf16raw*, const bf16raw*, const float*, bf16raw*, int,
-                                    int, int, int, int, int, int, int, int, int, int,
+                                    int, int, int, int, int, int, int, int, int, int, int, int,
                                     hipStream_t);
 hipError_t launch_conv_implicit_wgrad(const bf16raw*, const bf16raw*, float*, int, int, int, int,
-                                      int, int, int, int, float*, int, int, hipStream_t);
+                                      int, int, int, int, int, int, float*, int, int,
+                                      hipStream_t);
 int conv_implicit_wgrad_slices(int, int);
 hipError_t launch_cast_f64_f32(const double*, float*, int64_t, hipStream_t);
 hipError_t launch_cast_f64_bf16(const double*, bf16raw*, int64_t, hipStream_t);
@@ -199,11 +200,12 @@ at::Tensor flip_w2d(at::Tensor w2d, int64_t CI, int64_t KHW) {
 }
 
 at::Tensor conv_implicit_fwd(at::Tensor xP, at::Tensor wmat, c10::optional<at::Tensor> bias,
-                             int64_t KH, int64_t KW, bool relu, int64_t OHo, int64_t OWo) {
+                             int64_t KH, int64_t KW, bool relu, int64_t OHo, int64_t OWo,
+                             int64_t SH, int64_t SW) {
   check_gpu_contig(xP, at::kBFloat16, "xP");
   check_gpu_contig(wmat, at::kBFloat16, "wmat");
   int B = (int)xP.size(0), Hp = (int)xP.size(1), Wp = (int)xP.size(2), CI = (int)xP.size(3);
-  int OH = Hp - (int)KH + 1, OW = Wp - (int)KW + 1;
+  int OH = (Hp - (int)KH) / (int)SH + 1, OW = (Wp - (int)KW) / (int)SW + 1;
   if (OHo <= 0) OHo = OH;
   if (OWo <= 0) OWo = OW;
   int N = (int)wmat.size(0);
@@ -219,17 +221,18 @@ at::Tensor conv_implicit_fwd(at::Tensor xP, at::Tensor wmat, c10::optional<at::T
   }
   CHECK_HIP(launch_conv_implicit_fwd((const bf16raw*)xP.data_ptr(),
                                      (const bf16raw*)wmat.data_ptr(), bp, (bf16raw*)y.data_ptr(),
-                                     B, Hp, Wp, CI, (int)KH, (int)KW, N, Kpad, relu ? 1 : 0,
-                                     (int)OHo, (int)OWo, cur_stream()));
+                                     B, Hp, Wp, CI, (int)KH, (int)KW, (int)SH, (int)SW, N, Kpad,
+                                     relu ? 1 : 0, (int)OHo, (int)OWo, cur_stream()));
   return y;
 }
 
 at::Tensor conv_implicit_wgrad(at::Tensor dz, at::Tensor xP, int64_t KH, int64_t KW,
-                               int64_t splitk, bool slab, int64_t OHo, int64_t OWo) {
+                               int64_t splitk, bool slab, int64_t OHo, int64_t OWo, int64_t SH,
+                               int64_t SW) {
   check_gpu_contig(dz, at::kBFloat16, "dz");
   check_gpu_contig(xP, at::kBFloat16, "xP");
   int B = (int)xP.size(0), Hp = (int)xP.size(1), Wp = (int)xP.size(2), CI = (int)xP.size(3);
-  int OH = Hp - (int)KH + 1, OW = Wp - (int)KW + 1;
+  int OH = (Hp - (int)KH) / (int)SH + 1, OW = (Wp - (int)KW) / (int)SW + 1;
   if (OHo <= 0) OHo = OH;
   if (OWo <= 0) OWo = OW;
   int CO = (int)dz.size(1);
@@ -245,8 +248,8 @@ at::Tensor conv_implicit_wgrad(at::Tensor dz, at::Tensor xP, int64_t KH, int64_t
   }
   CHECK_HIP(launch_conv_implicit_wgrad((const bf16raw*)dz.data_ptr(),
                                        (const bf16raw*)xP.data_ptr(), dw.data_ptr<float>(), CO,
-                                       B, Hp, Wp, CI, (int)KH, (int)KW, (int)splitk, wsp,
-                                       (int)OHo, (int)OWo, cur_stream()));
+                                       B, Hp, Wp, CI, (int)KH, (int)KW, (int)SH, (int)SW,
+                                       (int)splitk, wsp, (int)OHo, (int)OWo, cur_stream()));
   return dw;
 }
 
@@ -888,10 +891,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("flip_w2d", &flip_w2d, "dgrad weight transform W[co,k,ci] -> W'[ci,rev(k),co]");
   m.def("conv_implicit_fwd", &conv_implicit_fwd, "implicit-GEMM s1 conv forward / dgrad",
         py::arg("xP"), py::arg("wmat"), py::arg("bias"), py::arg("KH"), py::arg("KW"),
-        py::arg("relu"), py::arg("OHo") = -1, py::arg("OWo") = -1);
+        py::arg("relu"), py::arg("OHo") = -1, py::arg("OWo") = -1,
+        py::arg("SH") = 1, py::arg("SW") = 1);
   m.def("conv_implicit_wgrad", &conv_implicit_wgrad, "implicit-GEMM s1 conv weight grad",
         py::arg("dz"), py::arg("xP"), py::arg("KH"), py::arg("KW"), py::arg("splitk"),
-        py::arg("slab"), py::arg("OHo") = -1, py::arg("OWo") = -1);
+        py::arg("slab"), py::arg("OHo") = -1, py::arg("OWo") = -1,
+        py::arg("SH") = 1, py::arg("SW") = 1);
   m.def("cast_f64_f32", &cast_f64_f32);
   m.def("cast_f64_bf16", &cast_f64_bf16, "one-pass device Vector pack cast");
   m.def("cast_f32_bf16", &cast_f32_bf16);

@@ -36,8 +36,9 @@ __device__ __forceinline__ int ic_swz(int row, int col) {
 
 // Patch geometry: maps (output row, k) -> padded-input address.
 struct ConvGeom {
-  int OW;    // GEMM-space output width (stride 1: OW = Wp - KW + 1)
+  int OW;    // GEMM-space output width ((Wp - KW)/SW + 1)
   int OHW;   // GEMM-space OH * OW
+  int SH, SW;  // stride (fwd/wgrad support s>1; dgrad is s1-only)
   int Hp, Wp;  // padded input spatial
   int CI;
   int KW;
@@ -187,7 +188,7 @@ __device__ __forceinline__ void stage_patch_glds(const bf16raw* __restrict__ xP,
     int kw = r2 / g.CI;
     int ci = r2 - kw * g.CI;
     const bf16raw* gp =
-        xP + (((int64_t)b * g.Hp + oh + kh) * g.Wp + ow + kw) * g.CI + ci;
+        xP + (((int64_t)b * g.Hp + oh * g.SH + kh) * g.Wp + ow * g.SW + kw) * g.CI + ci;
     __builtin_amdgcn_global_load_lds(
         (const __attribute__((address_space(1))) unsigned int*)gp,
         (__attribute__((address_space(3))) unsigned int*)(lds + c * CH_ROWS * BKT), 16, 0, 0);
@@ -222,8 +223,8 @@ __device__ __forceinline__ void stage_patch_tail(const bf16raw* __restrict__ xP,
       int r2 = k0 - kh * g.KWCI;
       int kw = r2 / g.CI;
       int ci = r2 - kw * g.CI;
-      *(shortx8*)dst = *(const shortx8*)(xP + (((int64_t)b * g.Hp + oh + kh) * g.Wp + ow + kw) *
-                                                  g.CI + ci);
+      *(shortx8*)dst = *(const shortx8*)(xP + (((int64_t)b * g.Hp + oh * g.SH + kh) * g.Wp +
+                                                ow * g.SW + kw) * g.CI + ci);
     } else {
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
@@ -234,7 +235,7 @@ __device__ __forceinline__ void stage_patch_tail(const bf16raw* __restrict__ xP,
           int r2 = k - kh * g.KWCI;
           int kw = r2 / g.CI;
           int ci = r2 - kw * g.CI;
-          v = xP[(((int64_t)b * g.Hp + oh + kh) * g.Wp + ow + kw) * g.CI + ci];
+          v = xP[(((int64_t)b * g.Hp + oh * g.SH + kh) * g.Wp + ow * g.SW + kw) * g.CI + ci];
         }
         dst[j] = v;
       }
@@ -393,9 +394,10 @@ __global__ __launch_bounds__(WR * WC * 64, 2) void conv_implicit_fwd_kernel(
 
 extern "C" hipError_t launch_conv_implicit_fwd(const bf16raw* xP, const bf16raw* wmat,
                                                const float* bias, bf16raw* y, int B, int Hp,
-                                               int Wp, int CI, int KH, int KW, int N, int Kpad,
-                                               int relu, int OHo, int OWo, hipStream_t stream) {
-  const int OH = Hp - KH + 1, OW = Wp - KW + 1;
+                                               int Wp, int CI, int KH, int KW, int SH, int SW,
+                                               int N, int Kpad, int relu, int OHo, int OWo,
+                                               hipStream_t stream) {
+  const int OH = (Hp - KH) / SH + 1, OW = (Wp - KW) / SW + 1;
   const int M = B * OH * OW;
   ConvGeom g;
   g.OW = OW;
@@ -406,6 +408,8 @@ extern "C" hipError_t launch_conv_implicit_fwd(const bf16raw* xP, const bf16raw*
   g.KW = KW;
   g.KWCI = KW * CI;
   g.Kreal = KH * KW * CI;
+  g.SH = SH;
+  g.SW = SW;
   g.OHo = OHo > 0 ? OHo : OH;
   g.OWo = OWo > 0 ? OWo : OW;
   const bool narrow = (N <= 64);
@@ -488,7 +492,8 @@ __device__ __forceinline__ void wg_stage_b(const bf16raw* __restrict__ xP, int n
     int r2 = n - kh * g.KWCI;
     int kw = r2 / g.CI;
     int ci = r2 - kw * g.CI;
-    const bf16raw* sp = xP + (((int64_t)b * g.Hp + oh + kh) * g.Wp + ow + kw) * g.CI + ci;
+    const bf16raw* sp =
+        xP + (((int64_t)b * g.Hp + oh * g.SH + kh) * g.Wp + ow * g.SW + kw) * g.CI + ci;
     *(shortx8*)regs = *(const shortx8*)sp;
     *(shortx8*)(regs + 8) = *(const shortx8*)(sp + 8);
   } else {
@@ -632,9 +637,9 @@ __global__ void ic_wgrad_reduce_kernel(const float* __restrict__ ws, float* __re
 
 extern "C" hipError_t launch_conv_implicit_wgrad(const bf16raw* dz, const bf16raw* xP, float* dw,
                                                  int CO, int B, int Hp, int Wp, int CI, int KH,
-                                                 int KW, int splitk, float* ws, int OHo, int OWo,
-                                                 hipStream_t stream) {
-  const int OH = Hp - KH + 1, OW = Wp - KW + 1;
+                                                 int KW, int SH, int SW, int splitk, float* ws,
+                                                 int OHo, int OWo, hipStream_t stream) {
+  const int OH = (Hp - KH) / SH + 1, OW = (Wp - KW) / SW + 1;
   const int N = KH * KW * CI;
   const int K = B * OH * OW;
   ConvGeom g;
@@ -646,6 +651,8 @@ extern "C" hipError_t launch_conv_implicit_wgrad(const bf16raw* dz, const bf16ra
   g.KW = KW;
   g.KWCI = KW * CI;
   g.Kreal = N;
+  g.SH = SH;
+  g.SW = SW;
   g.OHo = OHo > 0 ? OHo : OH;
   g.OWo = OWo > 0 ? OWo : OW;
   if (splitk < 1) splitk = 1;

@@ -252,28 +252,32 @@ def _pad64(t2d):
 
 
 class _ConvImplicitNHWCFn(torch.autograd.Function):
-    """Implicit-GEMM stride-1 conv (conv_implicit.hip): no col matrix at
-    all.  Forward stages x-patches straight from a zero-ring-padded NHWC
-    tensor; wgrad reads the same padded tensor (L2-served re-reads instead
-    of a KH*KW-duplicated col stream); dgrad runs the SAME forward kernel
-    on pad(dz, KH-1-p) with flipped/transposed weights — no dcol, no
-    col2im scatter.  Also a KH*KW-fold activation-memory saving: xP is
-    saved for backward instead of the col matrix.
-    Eligible: stride 1, ph == pw <= KH-1, KH == KW, CI % 16 == 0, CO % 8 == 0."""
+    """Implicit-GEMM conv (conv_implicit.hip): no col matrix at all.
+    Forward stages x-patches straight from a zero-ring-padded NHWC tensor
+    (any stride); wgrad reads the same padded tensor (L2-served re-reads
+    instead of a KH*KW-duplicated col stream); stride-1 dgrad runs the SAME
+    forward kernel on pad(dz, KH-1-p) with flipped/transposed weights — no
+    dcol, no col2im; stride-2 dgrad is a transposed conv and keeps the
+    dcol+col2im scatter (still no im2col).  Also a KH*KW-fold
+    activation-memory saving: xP is saved for backward, not the col matrix.
+    Eligible: stride in {1,2}, ph == pw <= KH-1, KH == KW, CI % 16 == 0,
+    CO % 8 == 0."""
 
     @staticmethod
-    def forward(ctx, x, w, b, padding, relu):
+    def forward(ctx, x, w, b, stride, padding, relu):
         ext = ops.ext()
         B, H, W, CI = x.shape
         CO, _, KH, KW = w.shape
         p = padding[0]
+        s = stride[0]
         K = KH * KW * CI
         w2d = w.permute(0, 2, 3, 1).reshape(CO, K).to(torch.bfloat16).contiguous()
         xP = ext.pad_nhwc(x, p) if p > 0 else x
-        y2d = ext.conv_implicit_fwd(xP, _pad64(w2d).contiguous(), b, KH, KW, relu)
-        OH, OW = H + 2 * p - KH + 1, W + 2 * p - KW + 1
+        y2d = ext.conv_implicit_fwd(xP, _pad64(w2d).contiguous(), b, KH, KW, relu,
+                                    -1, -1, s, s)
+        OH, OW = (H + 2 * p - KH) // s + 1, (W + 2 * p - KW) // s + 1
         ctx.save_for_backward(xP, w2d, y2d)
-        ctx.meta = (B, CI, H, W, CO, KH, KW, p, OH, OW, relu, b is not None)
+        ctx.meta = (B, CI, H, W, CO, KH, KW, p, s, OH, OW, relu, b is not None)
         ctx.b_ref = b
         return y2d.view(B, OH, OW, CO)
 
@@ -281,7 +285,7 @@ class _ConvImplicitNHWCFn(torch.autograd.Function):
     def backward(ctx, dy):
         ext = ops.ext()
         xP, w2d, y2d = ctx.saved_tensors
-        B, CI, H, W, CO, KH, KW, p, OH, OW, relu, has_bias = ctx.meta
+        B, CI, H, W, CO, KH, KW, p, s, OH, OW, relu, has_bias = ctx.meta
         dy2d = dy.reshape(B * OH * OW, CO).contiguous()
         if dy2d.dtype != torch.bfloat16:
             dy2d = dy2d.to(torch.bfloat16)
@@ -295,20 +299,26 @@ class _ConvImplicitNHWCFn(torch.autograd.Function):
             # slab combine beats atomics from CO >= 256
             sk = max(16, dz.shape[0] // 3136)
             slab = CO >= 256
-            dwp = ext.conv_implicit_wgrad(dz, xP, KH, KW, sk, slab)
+            dwp = ext.conv_implicit_wgrad(dz, xP, KH, KW, sk, slab, -1, -1, s, s)
             dw = dwp.reshape(CO, KH, KW, CI).permute(0, 3, 1, 2)
         if has_bias and ctx.needs_input_grad[2]:
             db = ext.bias_grad(dz)
 
         dx = None
         if ctx.needs_input_grad[0]:
-            wf = ext.flip_w2d(w2d, CI, KH * KW)
-            ring = KH - 1 - p
-            dzv = dz.view(B, OH, OW, CO)
-            dzP = ext.pad_nhwc(dzv, ring) if ring > 0 else dzv
-            dx = ext.conv_implicit_fwd(dzP, _pad64(wf).contiguous(), None, KH, KW, False)
-            dx = dx.view(B, H, W, CI)
-        return dx, dw, db, None, None
+            if s == 1:
+                wf = ext.flip_w2d(w2d, CI, KH * KW)
+                ring = KH - 1 - p
+                dzv = dz.view(B, OH, OW, CO)
+                dzP = ext.pad_nhwc(dzv, ring) if ring > 0 else dzv
+                dx = ext.conv_implicit_fwd(dzP, _pad64(wf).contiguous(), None, KH, KW, False)
+                dx = dx.view(B, H, W, CI)
+            else:
+                # strided dgrad is a transposed conv — scatter dcol back
+                # (col2im only; no im2col anywhere on this path)
+                dcol = ext.linear_dgrad(dz, _pad64(w2d).contiguous())
+                dx = ext.col2im_nhwc(dcol, B, CI, H, W, KH, KW, s, s, p, p)
+        return dx, dw, db, None, None, None
 
 
 class _ConvStemS2DFn(torch.autograd.Function):
@@ -398,7 +408,8 @@ class _ConvStemS2DFn(torch.autograd.Function):
 
 def _implicit_eligible(CI, CO, KH, KW, stride, padding):
     return (
-        tuple(stride) == (1, 1)
+        stride[0] == stride[1]
+        and stride[0] in (1, 2)
         and KH == KW
         and padding[0] == padding[1]
         and padding[0] <= KH - 1
@@ -414,7 +425,9 @@ def hip_conv2d_nhwc(x, weight, bias=None, stride=(1, 1), padding=(0, 0), relu=Fa
             x = x.to(torch.bfloat16)
         CO, CI_w, KH, KW = weight.shape
         if _implicit_eligible(x.shape[3], CO, KH, KW, stride, padding):
-            return _ConvImplicitNHWCFn.apply(x.contiguous(), weight, bias, tuple(padding), relu)
+            return _ConvImplicitNHWCFn.apply(
+                x.contiguous(), weight, bias, tuple(stride), tuple(padding), relu
+            )
         if (
             (KH, KW) == (7, 7)
             and tuple(stride) == (2, 2)

@@ -99,7 +99,7 @@ def test_implicit_backward_matches_torch(B, CI, CO, H, W, KH, pad):
 
     xh = _nhwc(bf(x32)).requires_grad_(True)
     wh = w32.clone().requires_grad_(True)
-    y = _ConvImplicitNHWCFn.apply(xh, wh, None, (pad, pad), False)
+    y = _ConvImplicitNHWCFn.apply(xh, wh, None, (1, 1), (pad, pad), False)
     gy = torch.randn_like(y.float()) * 0.1
     y.backward(bf(gy))
 
@@ -208,3 +208,34 @@ def test_stem_s2d_matches_torch():
     dw_err = (wh.grad - wr.grad).abs().max().item()
     dw_scale = wr.grad.abs().max().item() + 1e-6
     assert dw_err / dw_scale < 0.03, (dw_err, dw_scale)
+
+
+@pytest.mark.parametrize("B,CI,CO,H,W,KH,pad", [
+    (2, 64, 128, 28, 28, 3, 1),   # resnet l2 first conv (s2 3x3)
+    (2, 64, 128, 28, 28, 1, 0),   # resnet downsample (s2 1x1)
+    (2, 16, 32, 24, 24, 3, 1),    # s2 with K tail
+])
+def test_implicit_s2_fwd_bwd_matches_torch(B, CI, CO, H, W, KH, pad):
+    torch.manual_seed(9)
+    x32 = torch.randn(B, CI, H, W, device=DEV)
+    w32 = torch.randn(CO, CI, KH, KH, device=DEV) * 0.1
+
+    xh = _nhwc(bf(x32)).requires_grad_(True)
+    wh = w32.clone().requires_grad_(True)
+    y = _ConvImplicitNHWCFn.apply(xh, wh, None, (2, 2), (pad, pad), False)
+    xr = bf(x32).float().requires_grad_(True)
+    wr = bf(w32).float().requires_grad_(True)
+    yr = F.conv2d(xr, wr, None, stride=2, padding=pad)
+    assert y.shape == _nhwc(yr).shape
+    err = (y.float() - _nhwc(yr)).abs().max().item()
+    scale = yr.abs().max().item() + 1e-6
+    assert err / scale < 0.03, (err, scale)
+
+    gy = torch.randn_like(y.float()) * 0.1
+    y.backward(bf(gy))
+    yr.backward(bf(gy).float().permute(0, 3, 1, 2))
+    dw_err = (wh.grad - wr.grad).abs().max().item()
+    assert dw_err / (wr.grad.abs().max().item() + 1e-6) < 0.03, dw_err
+    dx_ref = _nhwc(xr.grad)
+    dx_err = (xh.grad.float() - dx_ref).abs().max().item()
+    assert dx_err / (dx_ref.abs().max().item() + 1e-6) < 0.03, dx_err
