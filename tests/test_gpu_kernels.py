@@ -284,3 +284,51 @@ def test_hipgraph_train_step_matches_eager():
     torch.cuda.synchronize()
     for a, b in zip(losses[None], losses["hipgraph"]):
         assert abs(a - b) < 5e-3 + 0.01 * abs(a), (losses[None], losses["hipgraph"])
+
+
+def test_hipgraph_with_batchnorm_buffers_restored():
+    """hipGraph whole-step capture on a BN model: the 2 real warmup steps
+    must not leak BatchNorm running-stat mutations into the first replay
+    (warmup snapshot/restore covers module buffers — ADVICE round-1 fix).
+    The graphed trajectory must match an eager trainer started from the
+    same init."""
+    import copy
+
+    from sparktorch_amd.parallel.sync import SyncTrainer
+
+    def build():
+        torch.manual_seed(31)
+        m = nn.Sequential(
+            nn.Conv2d(16, 16, 3, padding=1, bias=False),
+            nn.BatchNorm2d(16),
+            nn.ReLU(),
+            nn.Flatten(),
+            nn.Linear(16 * 8 * 8, 10),
+        )
+
+        class Wrap(nn.Module):
+            def __init__(self):
+                super().__init__()
+                self.m = m
+
+            def forward(self, x):
+                return self.m(x.view(-1, 16, 8, 8))
+
+        return Wrap()
+
+    x = torch.randn(512, 16 * 8 * 8, device=DEV).to(torch.bfloat16)
+    y = torch.randint(0, 10, (512,), device=DEV)
+
+    eager = SyncTrainer(build(), nn.CrossEntropyLoss(),
+                        torch.optim.Adam(build().parameters(), lr=1e-3),
+                        device=DEV, world_size=1)
+    # NB: optimizer param groups rebuilt inside SyncTrainer; the ctor above
+    # only supplies defaults
+    graphed = SyncTrainer(build(), nn.CrossEntropyLoss(),
+                          torch.optim.Adam(build().parameters(), lr=1e-3),
+                          device=DEV, world_size=1, compile_mode="hipgraph")
+
+    le = [eager.train_step(x, y) for _ in range(6)]
+    lg = [graphed.train_step(x, y) for _ in range(6)]
+    for i, (a, b) in enumerate(zip(le, lg)):
+        assert abs(a - b) / max(abs(a), 1e-3) < 0.05, (i, le, lg)
