@@ -111,7 +111,7 @@ def test_world4_agrees_with_single_process(tmp_path):
         )
 
 
-@pytest.mark.parametrize("world", [2, 4])
+@pytest.mark.parametrize("world", [2, 4, 8])
 def test_bench_torchrun_contract_gloo(world, tmp_path):
     """The exact launch the driver uses for SCALE_rNN.json, on CPU/gloo:
     torch.distributed.run -> bench.py --gpus N; rank 0 must print one valid
