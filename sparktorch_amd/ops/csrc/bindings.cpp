@@ -238,7 +238,9 @@ at::Tensor conv_implicit_wgrad(at::Tensor dz, at::Tensor xP, int64_t KH, int64_t
   int CO = (int)dz.size(1);
   TORCH_CHECK(dz.size(0) == (int64_t)B * OHo * OWo, "dz rows must be B*OHo*OWo");
   int64_t N = KH * KW * CI;
-  auto dw = at::zeros({CO, N}, xP.options().dtype(at::kFloat));
+  // atomic combine accumulates (needs zeros); the slab reduce writes
+  auto dw = slab ? at::empty({CO, N}, xP.options().dtype(at::kFloat))
+                 : at::zeros({CO, N}, xP.options().dtype(at::kFloat));
   float* wsp = nullptr;
   at::Tensor ws;
   if (slab) {

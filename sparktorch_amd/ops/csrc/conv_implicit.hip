@@ -608,7 +608,8 @@ __global__ __launch_bounds__(256, 2) void conv_implicit_wgrad_kernel(
   }
 }
 
-// Combine slabs: dw[m,n] += sum_z ws[z,m,n].
+// Combine slabs: dw[m,n] = sum_z ws[z,m,n] (plain write — the conv wgrad
+// allocates a fresh dw, so no zero-fill pass is needed on the slab path).
 __global__ void ic_wgrad_reduce_kernel(const float* __restrict__ ws, float* __restrict__ dw,
                                        int64_t MN, int zs) {
   int64_t i0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 4;
@@ -621,15 +622,12 @@ __global__ void ic_wgrad_reduce_kernel(const float* __restrict__ ws, float* __re
 #pragma unroll
         for (int j = 0; j < 4; ++j) s[j] += v[j];
       }
-      floatx4 d = *(const floatx4*)(dw + i);
-#pragma unroll
-      for (int j = 0; j < 4; ++j) d[j] += s[j];
-      *(floatx4*)(dw + i) = d;
+      *(floatx4*)(dw + i) = s;
     } else {
       for (int64_t k = i; k < MN; ++k) {
         float s = 0.f;
         for (int z = 0; z < zs; ++z) s += ws[(int64_t)z * MN + k];
-        dw[k] += s;
+        dw[k] = s;
       }
     }
   }
