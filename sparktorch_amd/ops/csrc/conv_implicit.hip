@@ -441,12 +441,15 @@ extern "C" hipError_t launch_conv_implicit_fwd(const bf16raw* xP, const bf16raw*
 // T14 register pipeline; B rows are patch slices read straight from xP.
 // ---------------------------------------------------------------------------
 
-// A slab: [128 m-rows][32 k] of dz^T == dz rows kt..kt+31, cols m0..m0+127.
+// A slab chunk: [ROWS m-rows][4096/ROWS k] of dz^T starting at column kt
+// (dz rows kt.., cols m0..m0+ROWS-1).  256 threads x 16 elements.
+template <int ROWS>
 __device__ __forceinline__ void wg_stage_a(const bf16raw* __restrict__ dz, int m0, int CO,
                                            int kt, int kmax, const ConvGeom g,
                                            bf16raw* __restrict__ regs, int t) {
-  int k = t >> 3;
-  int r0 = (t & 7) * 16;
+  constexpr int TPK = ROWS / 16;  // threads per k-column
+  int k = t / TPK;
+  int r0 = (t % TPK) * 16;
   int gk = kt + k;
   int64_t gkc = gk;
   if (g.OWo != g.OW || g.OHo * g.OW != g.OHW) {
@@ -474,13 +477,15 @@ __device__ __forceinline__ void wg_stage_a(const bf16raw* __restrict__ dz, int m
   }
 }
 
-// B slab: [128 n-cols][32 k] — n = (kh,kw,ci), k = batch row.  A thread's
-// 16-n span stays inside one ci-run (CI % 16 == 0, n0 16-aligned).
+// B slab chunk: [ROWS n-cols][4096/ROWS k] — n = (kh,kw,ci), k = batch row.
+// A thread's 16-n span stays inside one ci-run (CI % 16 == 0, n0 16-aligned).
+template <int ROWS>
 __device__ __forceinline__ void wg_stage_b(const bf16raw* __restrict__ xP, int n0, int Nmax,
                                            int kt, int kmax, const ConvGeom g,
                                            bf16raw* __restrict__ regs, int t) {
-  int k = t >> 3;
-  int r0 = (t & 7) * 16;
+  constexpr int TPK = ROWS / 16;
+  int k = t / TPK;
+  int r0 = (t % TPK) * 16;
   int gk = kt + k;  // batch row
   int n = n0 + r0;
   if (gk < kmax && n < Nmax) {
@@ -502,23 +507,37 @@ __device__ __forceinline__ void wg_stage_b(const bf16raw* __restrict__ xP, int n
   }
 }
 
-template <int LDSTRIDE>
+// Store a staged chunk into the padded swizzled image.  col0 = the chunk's
+// absolute column base inside the BKT-wide slab; the XOR swizzle applies
+// within each 32-column half (matching the fragment-read addressing).
+template <int ROWS, int LDSTRIDE>
 __device__ __forceinline__ void wg_stage_write(bf16raw* __restrict__ lds,
-                                               const bf16raw* __restrict__ regs, int t) {
-  int k = t >> 3;
-  int r0 = (t & 7) * 16;
+                                               const bf16raw* __restrict__ regs, int t,
+                                               int col0) {
+  constexpr int TPK = ROWS / 16;
+  int c = col0 + t / TPK;
+  int r0 = (t % TPK) * 16;
+  int half = c & ~31;
+  int cl = c & 31;
 #pragma unroll
-  for (int j = 0; j < 16; ++j) lds[(r0 + j) * LDSTRIDE + ic_swz(r0 + j, k)] = regs[j];
+  for (int j = 0; j < 16; ++j)
+    lds[(r0 + j) * LDSTRIDE + half + ic_swz(r0 + j, cl)] = regs[j];
 }
 
-template <bool SLAB>
+template <bool SLAB, int WR, int WC>
 __global__ __launch_bounds__(256, 2) void conv_implicit_wgrad_kernel(
     const bf16raw* __restrict__ dz, const bf16raw* __restrict__ xP, float* __restrict__ out,
     int CO, int N, int K, ConvGeom g, int k_per_split) {
+  constexpr int BMt = WR * 64;
+  constexpr int BNt = WC * 64;
   constexpr int BKT = 64;
   constexpr int LP = BKT + IC_LDS_PAD;
-  __shared__ bf16raw As[128 * LP];
-  __shared__ bf16raw Bs[128 * LP];
+  constexpr int AKW = 4096 / BMt;    // k-width one A staging call covers
+  constexpr int ACALLS = BKT / AKW;
+  constexpr int BKW = 4096 / BNt;
+  constexpr int BCALLS = BKT / BKW;
+  __shared__ bf16raw As[BMt * LP];
+  __shared__ bf16raw Bs[BNt * LP];
 
   const int gx = gridDim.x;
   int nwg = gx * gridDim.y;
@@ -526,8 +545,8 @@ __global__ __launch_bounds__(256, 2) void conv_implicit_wgrad_kernel(
   int q = nwg >> 3, rr = nwg & 7;
   int wg = ((orig & 7) < rr ? (orig & 7) * (q + 1) : rr * (q + 1) + ((orig & 7) - rr) * q) +
            (orig >> 3);
-  const int m0 = (wg % gx) * 128;
-  const int n0 = (wg / gx) * 128;
+  const int m0 = (wg % gx) * BMt;
+  const int n0 = (wg / gx) * BNt;
 
   int k_begin = blockIdx.z * k_per_split;
   int k_end = min(K, k_begin + k_per_split);
@@ -535,31 +554,33 @@ __global__ __launch_bounds__(256, 2) void conv_implicit_wgrad_kernel(
   const int t = threadIdx.x;
   const int wid = t / WAVE;
   const int lane = t % WAVE;
-  const int wr = wid / 2;
-  const int wc = wid % 2;
+  const int wr = wid / WC;
+  const int wc = wid % WC;
   const int l15 = lane & 15, kg = lane >> 4;
 
-  alignas(16) bf16raw rA[32];
-  alignas(16) bf16raw rB[32];
+  alignas(16) bf16raw rA[ACALLS * 16];
+  alignas(16) bf16raw rB[BCALLS * 16];
 
   floatx4 acc[4][4] = {};
 
-#define WG_LOAD(kt)                                                    \
-  do {                                                                 \
-    wg_stage_a(dz, m0, CO, (kt), k_end, g, rA, t);                     \
-    wg_stage_a(dz, m0, CO, (kt) + 32, k_end, g, rA + 16, t);           \
-    wg_stage_b(xP, n0, N, (kt), k_end, g, rB, t);                      \
-    wg_stage_b(xP, n0, N, (kt) + 32, k_end, g, rB + 16, t);            \
+#define WG_LOAD(kt)                                                              \
+  do {                                                                           \
+    _Pragma("unroll") for (int ca = 0; ca < ACALLS; ++ca)                        \
+        wg_stage_a<BMt>(dz, m0, CO, (kt) + ca * AKW, k_end, g, rA + ca * 16, t); \
+    _Pragma("unroll") for (int cb = 0; cb < BCALLS; ++cb)                        \
+        wg_stage_b<BNt>(xP, n0, N, (kt) + cb * BKW, k_end, g, rB + cb * 16, t);  \
   } while (0)
 
   if (k_begin < k_end) WG_LOAD(k_begin);
 
   for (int kt = k_begin; kt < k_end; kt += BKT) {
     if (kt > k_begin) __syncthreads();
-    wg_stage_write<LP>(As, rA, t);
-    wg_stage_write<LP>(As + 32, rA + 16, t);
-    wg_stage_write<LP>(Bs, rB, t);
-    wg_stage_write<LP>(Bs + 32, rB + 16, t);
+#pragma unroll
+    for (int ca = 0; ca < ACALLS; ++ca)
+      wg_stage_write<BMt, LP>(As, rA + ca * 16, t, ca * AKW);
+#pragma unroll
+    for (int cb = 0; cb < BCALLS; ++cb)
+      wg_stage_write<BNt, LP>(Bs, rB + cb * 16, t, cb * BKW);
     __syncthreads();
 
     if (kt + BKT < k_end) WG_LOAD(kt + BKT);
@@ -659,18 +680,32 @@ extern "C" hipError_t launch_conv_implicit_wgrad(const bf16raw* dz, const bf16ra
     kps = (int)ceil_div_i64(ceil_div_i64(K, splitk), 64) * 64;
     zs = (int)ceil_div_i64(K, kps);
   }
-  dim3 grid((unsigned)ceil_div_i64(CO, 128), (unsigned)ceil_div_i64(N, 128), (unsigned)zs);
+  // CO <= 32: the 64x256 flat tile (half the dead M rows, +20% MACs per
+  // staged byte).  Measured: wins for CO=32 (MNIST conv2 class), slightly
+  // loses at CO=64 (l1/stem class, 225->207 TF) — the 2x2 tile's extra
+  // blocks apparently hide slab latency better there.
+  const bool flat = (CO <= 32);
+  const int bm = flat ? 64 : 128, bn = flat ? 256 : 128;
+  dim3 grid((unsigned)ceil_div_i64(CO, bm), (unsigned)ceil_div_i64(N, bn), (unsigned)zs);
   if (ws != nullptr) {
-    conv_implicit_wgrad_kernel<true><<<grid, dim3(256), 0, stream>>>(dz, xP, ws, CO, N, K, g,
-                                                                     kps);
+    if (flat)
+      conv_implicit_wgrad_kernel<true, 1, 4><<<grid, dim3(256), 0, stream>>>(dz, xP, ws, CO, N,
+                                                                             K, g, kps);
+    else
+      conv_implicit_wgrad_kernel<true, 2, 2><<<grid, dim3(256), 0, stream>>>(dz, xP, ws, CO, N,
+                                                                             K, g, kps);
     HIP_CHECK_LAUNCH();
     int64_t MN = (int64_t)CO * N;
     int64_t rg = ceil_div_i64(MN, 1024);
     if (rg > 2048) rg = 2048;
     ic_wgrad_reduce_kernel<<<dim3((unsigned)rg), dim3(256), 0, stream>>>(ws, dw, MN, zs);
   } else {
-    conv_implicit_wgrad_kernel<false><<<grid, dim3(256), 0, stream>>>(dz, xP, dw, CO, N, K, g,
-                                                                      kps);
+    if (flat)
+      conv_implicit_wgrad_kernel<false, 1, 4><<<grid, dim3(256), 0, stream>>>(dz, xP, dw, CO, N,
+                                                                              K, g, kps);
+    else
+      conv_implicit_wgrad_kernel<false, 2, 2><<<grid, dim3(256), 0, stream>>>(dz, xP, dw, CO, N,
+                                                                              K, g, kps);
   }
   HIP_CHECK_LAUNCH();
   return hipSuccess;
