@@ -252,7 +252,17 @@ __global__ void bias_grad_small_kernel(const bf16raw* __restrict__ dz,
   int64_t lo = (int64_t)blockIdx.x * per;
   int64_t hi = lo + per < M ? lo + per : M;
   float s[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-  for (int64_t r = lo + rg; r < hi; r += rpg) {
+  int64_t r = lo + rg;
+  for (; r + (int64_t)rpg * 8 <= hi; r += (int64_t)rpg * 8) {  // 8 loads in flight
+    shortx8 v[8];
+#pragma unroll
+    for (int u = 0; u < 8; ++u) v[u] = *(const shortx8*)(dz + (r + (int64_t)u * rpg) * N + c0);
+#pragma unroll
+    for (int u = 0; u < 8; ++u)
+#pragma unroll
+      for (int j = 0; j < 8; ++j) s[j] += bf16_to_f32((bf16raw)v[u][j]);
+  }
+  for (; r < hi; r += rpg) {
     const shortx8 v = *(const shortx8*)(dz + r * N + c0);
 #pragma unroll
     for (int j = 0; j < 8; ++j) s[j] += bf16_to_f32((bf16raw)v[j]);
@@ -342,15 +352,22 @@ __global__ void small_wgrad_kernel(const bf16raw* __restrict__ dz, const bf16raw
     const bf16raw* zp = dz + co;
     const bf16raw* cp = col + kg * 8;
     int64_t r = lo + rg;
-    for (; r + rgs * 2 <= hi; r += rgs * 2) {  // 2-deep unroll for load overlap
-      float z0 = bf16_to_f32(zp[r * CO]);
-      const shortx8 c0 = *(const shortx8*)(cp + r * K);
-      int64_t r1 = r + rgs;
-      float z1 = bf16_to_f32(zp[r1 * CO]);
-      const shortx8 c1 = *(const shortx8*)(cp + r1 * K);
+    // 8-deep row unroll: issue all 16 loads before any FMA so the wave has
+    // 8 HBM round-trips in flight (the old 2-deep version was
+    // latency-bound at ~1.6 TB/s on the conv1 shape)
+    for (; r + rgs * 8 <= hi; r += rgs * 8) {
+      float z[8];
+      shortx8 c[8];
 #pragma unroll
-      for (int jj = 0; jj < 8; ++jj)
-        acc[jj] += z0 * bf16_to_f32((bf16raw)c0[jj]) + z1 * bf16_to_f32((bf16raw)c1[jj]);
+      for (int u = 0; u < 8; ++u) {
+        int64_t ru = r + (int64_t)u * rgs;
+        z[u] = bf16_to_f32(zp[ru * CO]);
+        c[u] = *(const shortx8*)(cp + ru * K);
+      }
+#pragma unroll
+      for (int u = 0; u < 8; ++u)
+#pragma unroll
+        for (int jj = 0; jj < 8; ++jj) acc[jj] += z[u] * bf16_to_f32((bf16raw)c[u][jj]);
     }
     for (; r < hi; r += rgs) {
       float z = bf16_to_f32(zp[r * CO]);
