@@ -103,16 +103,22 @@ def handle_features_device(
 
     ext = _ops.ext()
 
-    rows = [np.asarray(d.x_train).reshape(-1) for d in data]
+    first = np.asarray(data[0].x_train)
+    if first.ndim == 1:
+        # fast path: hand np.stack the row objects directly — a per-row
+        # asarray().reshape() list comprehension costs ~0.5 us x row and
+        # dominated the 131072-row ingest
+        rows = [d.x_train for d in data]
+    else:
+        rows = [np.asarray(d.x_train).reshape(-1) for d in data]
     # stack straight INTO the pinned staging tensor: one C-level gather pass,
     # no separate pin_memory() copy of the whole matrix afterwards
-    np_dtype = np.asarray(rows[0]).dtype
     t_dtype = {np.dtype(np.float64): _torch.float64,
-               np.dtype(np.float32): _torch.float32}.get(np_dtype)
+               np.dtype(np.float32): _torch.float32}.get(first.dtype)
     if t_dtype is None:
-        rows = [r.astype(np.float32) for r in rows]
+        rows = [np.asarray(r, dtype=np.float32).reshape(-1) for r in rows]
         t_dtype = _torch.float32
-    x_pin = _torch.empty((len(rows), rows[0].size), dtype=t_dtype, pin_memory=True)
+    x_pin = _torch.empty((len(rows), first.size), dtype=t_dtype, pin_memory=True)
     np.stack(rows, out=x_pin.numpy())
     x_dev = x_pin.to(device, non_blocking=True)  # hipMemcpyAsync from pinned
     if x_dev.dtype == _torch.float64:
@@ -124,12 +130,18 @@ def handle_features_device(
 
     y = None
     if data[0].y_train is not None:
-        y_rows = [
-            [d.y_train] if np.isscalar(d.y_train)
-            else np.asarray(d.y_train, dtype=np.float32).reshape(-1)
-            for d in data
-        ]
-        y_pin = _torch.from_numpy(np.asarray(y_rows, dtype=np.float32)).pin_memory()
+        if np.isscalar(data[0].y_train):
+            # C-loop over the scalar labels (reference wraps each as [y],
+            # util.py:72-74 — same [n,1] shape, no per-row python list)
+            y_host = np.fromiter(
+                (d.y_train for d in data), dtype=np.float32, count=len(data)
+            ).reshape(-1, 1)
+        else:
+            y_host = np.asarray(
+                [np.asarray(d.y_train, dtype=np.float32).reshape(-1) for d in data],
+                dtype=np.float32,
+            )
+        y_pin = _torch.from_numpy(y_host).pin_memory()
         y = y_pin.to(device, non_blocking=True)
 
     n = x.shape[0]
