@@ -385,6 +385,89 @@ def scenario_hogwild_partition_shuffles(spark):
     assert "predictions" in res[0]
 
 
+
+
+# --------------------------------------------------------------------------
+# group: unit  (double-internal semantics the e2e flows rely on)
+# --------------------------------------------------------------------------
+
+
+def scenario_row_semantics(spark):
+    import dill
+    from pyspark.sql import Row
+
+    r = Row(a=1.0, b="x")
+    assert r["a"] == 1.0 and r.b == "x" and r[1] == "x"
+    assert r + ("z",) == (1.0, "x", "z")  # tuple concat like pyspark
+    assert r.asDict() == {"a": 1.0, "b": "x"}
+    r2 = dill.loads(dill.dumps(r))  # ships to workers with fields intact
+    assert r2["b"] == "x" and r2 == r
+
+
+def scenario_repartition_shuffles(spark):
+    df = spark.createDataFrame([(float(i),) for i in range(500)], ["v"])
+    before = [r["v"] for r in df.rdd.collect()]
+    after = [r["v"] for r in df.rdd.repartition(4).collect()]
+    assert sorted(before) == sorted(after)
+    assert before != after, "repartition must re-randomize row placement"
+
+
+def scenario_broadcast_ship_check(spark):
+    sc = spark.sparkContext
+    bc = sc.broadcast({"w": [1, 2, 3]})
+    out = sc.parallelize(list(range(8)), 4).mapPartitions(
+        lambda part: [sum(bc.value["w"]) for _ in part]
+    ).collect()
+    assert out == [6] * 8
+    try:
+        sc.broadcast(lambda x: x and (yield))  # generator fn: dill can do it...
+    except Exception:
+        pass  # eager ship-check may reject exotic closures; either is fine
+
+
+def scenario_param_machinery(spark):
+    from pyspark.ml.param import Param, Params, TypeConverters
+    from pyspark import keyword_only
+
+    class Thing(Params):
+        knob = Param(Params._dummy(), "knob", "a knob", TypeConverters.toInt)
+
+        @keyword_only
+        def __init__(self, knob=None):
+            super().__init__()
+            self._setDefault(knob=7)
+            kwargs = self._input_kwargs
+            self._set(**{k: v for k, v in kwargs.items() if v is not None})
+
+    t = Thing()
+    assert t.getOrDefault(t.knob) == 7
+    assert t.knob.parent == t.uid  # re-parented per instance
+    t2 = Thing(knob=3)
+    assert t2.getOrDefault("knob") == 3
+    c = t2.copy()
+    c._set(knob=9)
+    assert t2.getOrDefault("knob") == 3  # copy does not alias the maps
+    try:
+        Thing(knob="not an int")
+    except TypeError:
+        pass
+    else:
+        raise AssertionError("TypeConverters.toInt must reject strings")
+
+
+def scenario_barrier_allgather(spark):
+    from pyspark import BarrierTaskContext
+
+    def worker(index, part):
+        ctx = BarrierTaskContext.get()
+        got = ctx.allGather(str(index * 10))
+        yield (index, got)
+
+    rdd = spark.sparkContext.parallelize(list(range(12)), 3)
+    out = dict(rdd.barrier().mapPartitionsWithIndex(worker).collect())
+    assert out == {0: ["0", "10", "20"], 1: ["0", "10", "20"], 2: ["0", "10", "20"]}
+
+
 GROUPS = {
     "core": [
         scenario_early_stopping,
@@ -413,6 +496,13 @@ GROUPS = {
         scenario_hogwild,
         scenario_hogwild_barrier,
         scenario_hogwild_partition_shuffles,
+    ],
+    "unit": [
+        scenario_row_semantics,
+        scenario_repartition_shuffles,
+        scenario_broadcast_ship_check,
+        scenario_param_machinery,
+        scenario_barrier_allgather,
     ],
 }
 

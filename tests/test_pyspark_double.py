@@ -79,3 +79,8 @@ def test_scenarios_pipeline():
 def test_scenarios_hogwild():
     out = _run_group("hogwild")
     assert out.count("PASS") == 3
+
+
+def test_scenarios_unit():
+    out = _run_group("unit")
+    assert out.count("PASS") == 5
