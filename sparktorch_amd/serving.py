@@ -28,6 +28,7 @@ Endpoints: ``GET /health``, ``GET /info``, ``POST /predict`` with
 from __future__ import annotations
 
 import argparse
+import threading
 from typing import Any, List, Optional
 
 import numpy as np
@@ -61,6 +62,10 @@ class InferenceServer:
             self._runner = GraphedForward(self.model, device=self.device,
                                           batch_size=self.batch_size)
         self._n_served = 0
+        # FastAPI dispatches sync endpoints on a threadpool; the hipGraph
+        # replay reuses static input/output buffers, so scoring must be
+        # serialized (throughput comes from batching, not request threads)
+        self._lock = threading.Lock()
 
     # ------------------------------------------------------------------
     @torch.no_grad()
@@ -71,6 +76,10 @@ class InferenceServer:
         if feats.ndim == 1:
             feats = feats.reshape(1, -1)
         out: List[Any] = []
+        with self._lock:
+            return self._predict_locked(feats, use_vec, out)
+
+    def _predict_locked(self, feats, use_vec, out):
         for s in range(0, len(feats), self.batch_size):
             xb = torch.from_numpy(feats[s : s + self.batch_size])
             if self._runner is not None:

@@ -81,3 +81,20 @@ def test_serving_hipgraph_gpu():
     # near-ties; require 95% agreement
     agree = sum(1 for a, b in zip(preds, ref) if a == b) / len(ref)
     assert agree > 0.95, agree
+
+
+def test_predict_thread_safe():
+    """Concurrent /predict calls must serialize on the scoring engine
+    (the GPU path replays a hipGraph with static buffers)."""
+    import concurrent.futures
+
+    net = _net(out=3)
+    srv = InferenceServer(net, device="cpu", batch_size=32)
+    x = np.random.default_rng(2).normal(size=(64, 10)).astype(np.float32)
+    with torch.no_grad():
+        ref = net(torch.from_numpy(x)).argmax(dim=1).double().tolist()
+    with concurrent.futures.ThreadPoolExecutor(max_workers=8) as ex:
+        futs = [ex.submit(srv.predict, x.tolist()) for _ in range(16)]
+        for f in futs:
+            assert f.result() == ref
+    assert srv._n_served == 16 * 64
