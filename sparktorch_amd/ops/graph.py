@@ -70,4 +70,6 @@ class GraphedForward:
         if n < self.batch_size:
             self._static_in[n:].zero_()
         self._graph.replay()
+        # NB: a view of the static output buffer — consume (copy/cpu()) before
+        # the next replay overwrites it; both call sites do so immediately
         return self._static_out[:n]
