@@ -109,6 +109,11 @@ def hip_linear(
             x = x.to(torch.bfloat16)
         if not x.is_contiguous():
             x = x.contiguous()
+        if x.dim() > 2:
+            # nn.Linear semantics for N-D inputs: apply over the last dim
+            lead = x.shape[:-1]
+            y = _LinearFn.apply(x.reshape(-1, x.shape[-1]), weight, bias, relu)
+            return y.view(*lead, y.shape[-1])
         y = _LinearFn.apply(x, weight, bias, relu)
         return y
     y = F.linear(x, weight, bias)

@@ -141,3 +141,55 @@ def test_estimator_resnet_fit_transform_on_gpu():
     out = model.transform(df).collect()
     assert len(out) == 24
     assert all(r["predicted"] in (0.0, 1.0, 2.0, 3.0) for r in out)
+
+
+def test_arbitrary_module_transformer_trains():
+    """Parity guarantee beyond the model zoo: the reference accepts ANY
+    nn.Module (serialize_torch_obj is generic).  Modules the converter has
+    no native kernel for (LayerNorm, MultiheadAttention, GELU) must run
+    through torch-ROCm eager inside the same trainer, mixed with native
+    HipLinear layers, and train."""
+    import torch.nn as nn
+
+    from sparktorch_amd.parallel.sync import SyncTrainer
+
+    torch.manual_seed(21)
+
+    class TinyTransformer(nn.Module):
+        def __init__(self, d=64, nhead=4, seq=16, classes=10):
+            super().__init__()
+            self.seq, self.d = seq, d
+            self.inp = nn.Linear(49, d)          # -> HipLinear via converter
+            self.attn = nn.MultiheadAttention(d, nhead, batch_first=True)
+            self.ln1 = nn.LayerNorm(d)
+            self.ff = nn.Sequential(nn.Linear(d, 4 * d), nn.GELU(), nn.Linear(4 * d, d))
+            self.ln2 = nn.LayerNorm(d)
+            self.head = nn.Linear(d, classes)
+
+        def forward(self, x):
+            x = x.view(-1, self.seq, 49).float()
+            # native HipLinear layers emit bf16 activations; eager fp32
+            # modules (attention/LayerNorm) take an explicit .float() at the
+            # boundary — standard mixed-precision practice
+            h = self.inp(x).float()
+            a, _ = self.attn(h, h, h, need_weights=False)
+            h = self.ln1(h + a)
+            h = self.ln2(h + self.ff(h).float())
+            return self.head(h.mean(dim=1))
+
+    model = TinyTransformer()
+    opt = torch.optim.Adam(model.parameters(), lr=1e-3)
+    tr = SyncTrainer(model, nn.CrossEntropyLoss(), opt, device="cuda:0", world_size=1)
+
+    x = torch.randn(512, 16 * 49, device="cuda:0").to(torch.bfloat16)
+    y = torch.randint(0, 10, (512,), device="cuda:0")
+    losses = [tr.train_step(x, y) for _ in range(12)]
+    # steady descent (measured ~0.012/step at this lr); monotone overall
+    assert losses[-1] < losses[0] - 0.05, losses
+    assert losses[-1] < min(losses[:3]), losses
+
+    # the dense layers really did convert to the native path
+    from sparktorch_amd.ops.modules import HipLinear
+
+    assert isinstance(tr.model.inp, HipLinear)
+    assert isinstance(tr.model.head, HipLinear)
