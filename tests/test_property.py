@@ -68,3 +68,52 @@ def test_stopwords_codec_roundtrip(obj):
     sw = obj_to_stopwords(obj)
     assert isinstance(sw, list) and len(sw) == 2
     assert stopwords_to_obj(sw) == obj
+
+
+@settings(max_examples=40, deadline=None)
+@given(
+    st.integers(1, 40),           # rows
+    st.integers(1, 12),           # feature dim
+    st.sampled_from(["f64", "f32", "list"]),
+    st.booleans(),                # scalar labels or vector labels
+)
+def test_handle_features_pack_invariants(n, dim, kind, scalar_labels):
+    from sparktorch_amd.utils.data import handle_features
+    from sparktorch_amd.utils.serialize import DataObj
+
+    rng = np.random.default_rng(n * 100 + dim)
+    rows = []
+    for i in range(n):
+        x = rng.normal(size=dim)
+        if kind == "f32":
+            x = x.astype(np.float32)
+        elif kind == "list":
+            x = list(x)
+        y = float(i) if scalar_labels else rng.normal(size=3).astype(np.float32)
+        rows.append(DataObj(x, y, None, None))
+
+    d = handle_features(rows, 0.0)
+    assert d.x_train.shape == (n, dim) and d.x_train.dtype == torch.float32
+    assert d.y_train.shape == (n, 1 if scalar_labels else 3)
+    # row order and values preserved (fp32 rounding only)
+    for i in (0, n - 1):
+        assert np.allclose(d.x_train[i].numpy(),
+                           np.asarray(rows[i].x_train, dtype=np.float32), atol=0)
+
+
+@settings(max_examples=20, deadline=None)
+@given(st.integers(4, 60), st.floats(0.05, 0.6))
+def test_handle_features_validation_split_partitions(n, pct):
+    from sparktorch_amd.utils.data import handle_features
+    from sparktorch_amd.utils.serialize import DataObj
+
+    rng = np.random.default_rng(7)
+    rows = [DataObj(rng.normal(size=5), 1.0, None, None) for _ in range(n)]
+    d = handle_features(rows, pct)
+    n_val = int(n * pct)
+    if n_val == 0:
+        assert d.x_val is None
+        assert d.x_train.shape[0] == n
+    else:
+        assert d.x_val.shape[0] == n_val
+        assert d.x_train.shape[0] == n - n_val

@@ -98,3 +98,18 @@ def test_predict_thread_safe():
         for f in futs:
             assert f.result() == ref
     assert srv._n_served == 16 * 64
+
+
+def test_load_stage_roundtrip(tmp_path):
+    """Saved carrier stage -> serving loader -> same predictions."""
+    from sparktorch_amd.serving import InferenceServer, load_stage
+
+    net = _net(out=1)
+    stage = create_spark_torch_model(net, "features", "p")
+    path = str(tmp_path / "stage")
+    stage.write().overwrite().save(path)
+    srv = InferenceServer(load_stage(path), device="cpu")
+    x = [[0.3] * 10]
+    with torch.no_grad():
+        ref = float(net(torch.tensor(x)).reshape(-1)[0])
+    assert abs(srv.predict(x)[0] - ref) < 1e-6
