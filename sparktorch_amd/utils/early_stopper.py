@@ -1,54 +1,58 @@
-"""Early stopping.  Parity with reference sparktorch/early_stopper.py:8-56.
+"""Early stopping with the reference's exact observable semantics
+(reference sparktorch/early_stopper.py:8-56, itself adapted from a public
+gist):
 
-Semantics preserved exactly:
-  * ``patience == 0`` disables early stopping (step always returns False).
-  * NaN metric -> immediate stop.
-  * ``min_delta`` may be absolute or a percentage of the previous best.
-  * mode 'min' (loss-like) or 'max' (score-like).
+  * ``patience == 0`` disables stopping entirely — ``step`` always returns
+    False, even for NaN, and an invalid ``mode`` is not rejected.
+  * NaN metric -> immediate stop (when enabled).
+  * ``min_delta`` is an absolute margin, or a percentage of the current
+    best when ``percentage=True`` (computed from the signed best, matching
+    the reference).
+  * mode ``'min'`` for loss-like metrics, ``'max'`` for score-like.
+
+Independent design: instead of the gist's per-mode comparison-lambda
+factory, one ``is_better`` method computes the signed improvement margin
+directly.
 """
 
 from __future__ import annotations
 
-import math
-
-
-class EarlyStopping(object):
-    def __init__(self, mode: str = "min", min_delta: float = 0.0, patience: int = 10, percentage: bool = False):
+class EarlyStopping:
+    def __init__(
+        self,
+        mode: str = "min",
+        min_delta: float = 0.0,
+        patience: int = 10,
+        percentage: bool = False,
+    ):
         self.mode = mode
         self.min_delta = min_delta
         self.patience = patience
         self.percentage = percentage
-        self.best = None
+        self.best: float | None = None
         self.num_bad_epochs = 0
-        if patience == 0:
-            self.is_better = lambda a, b: True
-            self.step = lambda a: False  # type: ignore[assignment]
-        else:
-            self._init_is_better(mode, min_delta, percentage)
+        if patience != 0 and mode not in ("min", "max"):
+            raise ValueError("mode " + mode + " is unknown!")
+
+    def is_better(self, metric: float, best: float) -> bool:
+        """Did ``metric`` improve on ``best`` by more than the margin?"""
+        margin = (best * self.min_delta / 100.0) if self.percentage else self.min_delta
+        if self.mode == "min":
+            return metric < best - margin
+        return metric > best + margin
 
     def step(self, metric: float) -> bool:
-        if metric != metric or (isinstance(metric, float) and math.isnan(metric)):
-            return True  # NaN => stop (reference early_stopper.py:28-29)
+        """Record one epoch's metric; True means training should stop."""
+        if self.patience == 0:
+            return False
+        if metric != metric:  # NaN of any float flavor
+            return True
         if self.best is None:
             self.best = metric
             return False
         if self.is_better(metric, self.best):
-            self.num_bad_epochs = 0
             self.best = metric
+            self.num_bad_epochs = 0
         else:
             self.num_bad_epochs += 1
         return self.num_bad_epochs >= self.patience
-
-    def _init_is_better(self, mode: str, min_delta: float, percentage: bool) -> None:
-        if mode not in {"min", "max"}:
-            raise ValueError("mode " + mode + " is unknown!")
-        if not percentage:
-            if mode == "min":
-                self.is_better = lambda a, best: a < best - min_delta
-            else:
-                self.is_better = lambda a, best: a > best + min_delta
-        else:
-            if mode == "min":
-                self.is_better = lambda a, best: a < best - (best * min_delta / 100)
-            else:
-                self.is_better = lambda a, best: a > best + (best * min_delta / 100)
