@@ -1,4 +1,4 @@
-// Implicit-GEMM stride-1 NHWC conv for gfx950 — no col matrix.
+// Implicit-GEMM NHWC conv for gfx950 (stride 1 and 2) — no col matrix.
 //
 // The explicit im2col path (conv_nhwc.hip) materializes col [M, KH*KW*CI]
 // (KH*KW times the activation bytes), streams it through the GEMM, and
@@ -12,16 +12,17 @@
 //     zero-filling register fallback into the same swizzled image;
 //   * the pad ring is materialized once per tensor (pad_nhwc) — no patch
 //     address is ever out of bounds, no branches in the hot stage;
-//   * dgrad IS the forward kernel: dx = conv_s1(pad(dz, KH-1-p), flipW)
-//     with flipW[ci, (kh,kw), co] = W[co, (KH-1-kh, KW-1-kw), ci]
-//     (flip_w2d) — no dcol, no col2im scatter;
+//   * stride-1 dgrad IS the forward kernel: dx = conv_s1(pad(dz, KH-1-p),
+//     flipW) with flipW[ci, (kh,kw), co] = W[co, (KH-1-kh, KW-1-kw), ci]
+//     (flip_w2d) — no dcol, no col2im scatter; stride-2 dgrad is a
+//     transposed conv and keeps a col2im-only scatter (no im2col);
 //   * wgrad keeps gemm.hip's T14 register pipeline but stages its B tile
 //     from x patches (CI % 16 == 0) — the duplicated col stream becomes
 //     L2-served re-reads of the 1x tensor; split-K combines by fp32
 //     atomics or per-slice slabs + reduce (measured per shape).
 //
-// Eligibility (enforced by the python dispatch): stride 1, CI % 16 == 0,
-// N (output channels) >= 32; wmat column-padded to a multiple of 64.
+// Eligibility (enforced by the python dispatch): stride in {1, 2},
+// CI % 16 == 0, CO % 8 == 0; wmat column-padded to a multiple of 64.
 // Everything else keeps the explicit im2col path.
 
 #include "common.h"
