@@ -71,7 +71,7 @@ def get_barrier_context(use_barrier: bool = True):
     barrier task, otherwise the installed local one."""
     if _LOCAL_CTX is not None:
         return _LOCAL_CTX
-    try:  # pragma: no cover - Spark-only path
+    try:  # Spark barrier tasks (real pyspark or the vendored double)
         from pyspark import BarrierTaskContext
 
         return BarrierTaskContext.get()

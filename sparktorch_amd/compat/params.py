@@ -16,7 +16,7 @@ from __future__ import annotations
 
 import uuid
 
-try:  # pragma: no cover - exercised only when pyspark is installed
+try:  # exercised under vendor/pyspark (tests/test_pyspark_double.py) or real pyspark
     import pyspark  # noqa: F401
 
     HAS_PYSPARK = True
@@ -24,7 +24,7 @@ except ImportError:
     HAS_PYSPARK = False
 
 
-if HAS_PYSPARK:  # pragma: no cover
+if HAS_PYSPARK:  # exercised under vendor/pyspark or real pyspark
     from pyspark import keyword_only
     from pyspark.ml.base import Estimator, Model, Transformer
     from pyspark.ml.param import Param, Params, TypeConverters
