@@ -264,6 +264,7 @@ def handle_model(
     early_stop_patience: int = -1,
     backend: Optional[str] = None,
     compile_mode: Optional[str] = None,
+    bucket_cap_mb: float = 25.0,
 ) -> List[dict]:
     if index < 0:
         raise RuntimeError("invalid partition index")
@@ -309,6 +310,7 @@ def handle_model(
             device=dev,
             world_size=world_size,
             compile_mode=compile_mode,
+            bucket_cap_mb=bucket_cap_mb,
         )
 
         es = EarlyStopping(patience=early_stop_patience) if early_stop_patience > 0 else None
@@ -376,6 +378,7 @@ def train_distributed(
     early_stop_patience: int = -1,
     backend: Optional[str] = None,
     compile_mode: Optional[str] = None,
+    bucket_cap_mb: float = 25.0,
 ) -> dict:
     """Run sync data-parallel training over a (barrier) RDD; returns the
     trained state_dict."""
@@ -410,6 +413,7 @@ def train_distributed(
                 early_stop_patience=early_stop_patience,
                 backend=backend,
                 compile_mode=compile_mode,
+                bucket_cap_mb=bucket_cap_mb,
             )
 
         # barrier scheduling is mandatory for sync mode: every rank must be

@@ -219,6 +219,13 @@ class SparkTorch(
     miniBatch = Param(Params._dummy(), "miniBatch", "", TypeConverters.toInt)
     validationPct = Param(Params._dummy(), "validationPct", "", TypeConverters.toFloat)
     compileMode = Param(Params._dummy(), "compileMode", "", TypeConverters.toString)
+    # MI355X tunables beyond the reference's 17 Params ("adds
+    # device/RCCL tunables as new Params")
+    bucketCapMb = Param(Params._dummy(), "bucketCapMb",
+                        "gradient all-reduce bucket size (MB)", TypeConverters.toFloat)
+    backend = Param(Params._dummy(), "backend",
+                    "torch.distributed backend override (nccl=RCCL / gloo)",
+                    TypeConverters.toString)
 
     @keyword_only
     def __init__(
@@ -241,6 +248,8 @@ class SparkTorch(
         mode=None,
         device=None,
         compileMode=None,
+        bucketCapMb=None,
+        backend=None,
     ):
         super().__init__()
         self._setDefault(
@@ -262,6 +271,8 @@ class SparkTorch(
             mode="synchronous",
             device="cpu",
             compileMode=None,
+            bucketCapMb=25.0,
+            backend=None,
         )
         kwargs = self._input_kwargs
         self.setParams(**kwargs)
@@ -287,6 +298,8 @@ class SparkTorch(
         mode=None,
         device=None,
         compileMode=None,
+        bucketCapMb=None,
+        backend=None,
     ):
         kwargs = self._input_kwargs
         return self._set(**kwargs)
@@ -340,6 +353,12 @@ class SparkTorch(
     def getCompileMode(self):
         return self.getOrDefault(self.compileMode)
 
+    def getBucketCapMb(self):
+        return self.getOrDefault(self.bucketCapMb)
+
+    def getBackend(self):
+        return self.getOrDefault(self.backend)
+
     # ------------------------------------------------------------------
     def _fit(self, dataset) -> SparkTorchModel:
         inp = self.getOrDefault(self.inputCol)
@@ -380,6 +399,8 @@ class SparkTorch(
                 device=device,
                 early_stop_patience=early_stop,
                 compile_mode=compile_mode,
+                backend=self.getOrDefault(self.backend),
+                bucket_cap_mb=self.getOrDefault(self.bucketCapMb),
             )
         elif mode == "hogwild":
             from sparktorch_amd.parallel import hogwild

@@ -283,3 +283,24 @@ def test_partition_shuffles_hogwild_and_sync(data_df):
         torchObj=obj, iters=2, partitionShuffles=2, mode="synchronous", device="cpu",
     ).fit(data_df)
     assert m2.transform(data_df).count() == 400
+
+
+def test_rccl_tunable_params(data_df, general_torch_obj):
+    """The MI355X tunables beyond the reference's 17 Params reach the
+    engine: a tiny bucket cap forces multi-bucket training, and an explicit
+    backend override is honored."""
+    stm = SparkTorch(
+        inputCol="features",
+        labelCol="label",
+        predictionCol="predictions",
+        torchObj=general_torch_obj,
+        iters=3,
+        partitions=2,
+        bucketCapMb=0.001,   # every parameter tensor becomes its own bucket
+        backend="gloo",
+    )
+    assert stm.getBucketCapMb() == 0.001
+    assert stm.getBackend() == "gloo"
+    model = stm.fit(data_df)
+    res = model.transform(data_df).take(1)
+    assert "predictions" in res[0]
