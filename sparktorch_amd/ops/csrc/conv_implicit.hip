@@ -271,7 +271,10 @@ __device__ __forceinline__ void stage_w_glds(const bf16raw* __restrict__ src,
   }
 }
 
-template <int WR, int WC, bool RELU, bool BIAS>
+// NFR = active 16-column output fragments per wave (4 = the full 64-wide
+// sub-tile; 1 serves N <= 16 shapes — e.g. a CI=16 dgrad — without
+// issuing 75% dead MFMAs).
+template <int WR, int WC, bool RELU, bool BIAS, int NFR = 4>
 __global__ __launch_bounds__(WR * WC * 64, 2) void conv_implicit_fwd_kernel(
     const bf16raw* __restrict__ xP, const bf16raw* __restrict__ wmat, bf16raw* __restrict__ y,
     const float* __restrict__ bias, int M, int N, int Kpad, ConvGeom g) {
@@ -299,7 +302,7 @@ __global__ __launch_bounds__(WR * WC * 64, 2) void conv_implicit_fwd_kernel(
   const int wc = wid % WC;
   const int l15 = lane & 15, kg = lane >> 4;
 
-  floatx4 acc[4][4] = {};
+  floatx4 acc[4][NFR] = {};
 
 #define IC_STAGE(bufi, kt)                                                 \
   do {                                                                     \
@@ -319,7 +322,7 @@ __global__ __launch_bounds__(WR * WC * 64, 2) void conv_implicit_fwd_kernel(
 
 #pragma unroll
     for (int sub = 0; sub < SUBS; ++sub) {
-      cfrag_t a[4], b[4];
+      cfrag_t a[4], b[NFR];
 #pragma unroll
       for (int mi = 0; mi < 4; ++mi) {
         int row = wr * 64 + mi * 16 + l15;
@@ -327,7 +330,7 @@ __global__ __launch_bounds__(WR * WC * 64, 2) void conv_implicit_fwd_kernel(
         a[mi] = *(const cfrag_t*)&As[buf][row * BKT + ((kq ^ (row & (SLOTS - 1))) << 3)];
       }
 #pragma unroll
-      for (int ni = 0; ni < 4; ++ni) {
+      for (int ni = 0; ni < NFR; ++ni) {
         int rowb = wc * 64 + ni * 16 + l15;
         int kq = kg + sub * 4;
         b[ni] = *(const cfrag_t*)&Bs[buf][rowb * BKT + ((kq ^ (rowb & (SLOTS - 1))) << 3)];
@@ -335,7 +338,7 @@ __global__ __launch_bounds__(WR * WC * 64, 2) void conv_implicit_fwd_kernel(
 #pragma unroll
       for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
-        for (int ni = 0; ni < 4; ++ni)
+        for (int ni = 0; ni < NFR; ++ni)
           acc[mi][ni] =
               __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[mi], b[ni], acc[mi][ni], 0, 0, 0);
     }
@@ -357,7 +360,7 @@ __global__ __launch_bounds__(WR * WC * 64, 2) void conv_implicit_fwd_kernel(
 #pragma unroll
   for (int mi = 0; mi < 4; ++mi) {
 #pragma unroll
-    for (int ni = 0; ni < 4; ++ni)
+    for (int ni = 0; ni < NFR; ++ni)
 #pragma unroll
       for (int r = 0; r < 4; ++r) ep[(kg * 4 + r) * EPAD + ni * 16 + l15] = acc[mi][ni][r];
     __builtin_amdgcn_s_waitcnt(0);
@@ -375,7 +378,7 @@ __global__ __launch_bounds__(WR * WC * 64, 2) void conv_implicit_fwd_kernel(
       for (int h = 0; h < 2; ++h) {
         int c0 = oct * 16 + h * 8;
         int n = n_base + c0;
-        if (n < N) {
+        if (c0 < NFR * 16 && n < N) {
           alignas(16) short outp[8];
           const float* src = ep + orow * EPAD + c0;
 #pragma unroll
@@ -431,6 +434,11 @@ extern "C" hipError_t launch_conv_implicit_fwd(const bf16raw* xP, const bf16raw*
   if (relu && bias) IC_DISPATCH(true, true);
   else if (relu) IC_DISPATCH(true, false);
   else if (bias) IC_DISPATCH(false, true);
+  else if (narrow && N <= 16)
+    // single-fragment column (dgrad of CI<=16 convs): 1/4 the MFMAs of the
+    // generic narrow tile, same staging
+    conv_implicit_fwd_kernel<4, 1, false, false, 1><<<grid, block, 0, stream>>>(
+        xP, wmat, y, bias, M, N, Kpad, g);
   else IC_DISPATCH(false, false);
 #undef IC_DISPATCH
   HIP_CHECK_LAUNCH();
