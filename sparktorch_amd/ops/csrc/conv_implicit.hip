@@ -421,9 +421,17 @@ extern "C" hipError_t launch_conv_implicit_fwd(const bf16raw* xP, const bf16raw*
   dim3 grid((unsigned)ceil_div_i64(M, bm), (unsigned)ceil_div_i64(N, bn), 1);
   dim3 block(256);
 
+  // active fragment columns: N <= 16 -> 1, N <= 32 -> 2, else the full 64
+  // (small-CO shapes otherwise issue 50-75% dead MFMAs in the narrow tile)
 #define IC_DISPATCH(RELUV, BIASV)                                                           \
   do {                                                                                      \
-    if (narrow)                                                                             \
+    if (narrow && N <= 16)                                                                  \
+      conv_implicit_fwd_kernel<4, 1, RELUV, BIASV, 1><<<grid, block, 0, stream>>>(          \
+          xP, wmat, y, bias, M, N, Kpad, g);                                                \
+    else if (narrow && N <= 32)                                                             \
+      conv_implicit_fwd_kernel<4, 1, RELUV, BIASV, 2><<<grid, block, 0, stream>>>(          \
+          xP, wmat, y, bias, M, N, Kpad, g);                                                \
+    else if (narrow)                                                                        \
       conv_implicit_fwd_kernel<4, 1, RELUV, BIASV><<<grid, block, 0, stream>>>(             \
           xP, wmat, y, bias, M, N, Kpad, g);                                                \
     else                                                                                    \
@@ -434,11 +442,6 @@ extern "C" hipError_t launch_conv_implicit_fwd(const bf16raw* xP, const bf16raw*
   if (relu && bias) IC_DISPATCH(true, true);
   else if (relu) IC_DISPATCH(true, false);
   else if (bias) IC_DISPATCH(false, true);
-  else if (narrow && N <= 16)
-    // single-fragment column (dgrad of CI<=16 convs): 1/4 the MFMAs of the
-    // generic narrow tile, same staging
-    conv_implicit_fwd_kernel<4, 1, false, false, 1><<<grid, block, 0, stream>>>(
-        xP, wmat, y, bias, M, N, Kpad, g);
   else IC_DISPATCH(false, false);
 #undef IC_DISPATCH
   HIP_CHECK_LAUNCH();
