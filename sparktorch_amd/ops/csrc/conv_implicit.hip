@@ -536,7 +536,9 @@ __device__ __forceinline__ void wg_stage_write(bf16raw* __restrict__ lds,
     lds[(r0 + j) * LDSTRIDE + half + ic_swz(r0 + j, cl)] = regs[j];
 }
 
-template <bool SLAB, int WR, int WC>
+// MFR = active 16-row output fragments per wave (4 = full 64 rows; 2
+// serves CO <= 32 without half-dead MFMAs).
+template <bool SLAB, int WR, int WC, int MFR = 4>
 __global__ __launch_bounds__(256, 2) void conv_implicit_wgrad_kernel(
     const bf16raw* __restrict__ dz, const bf16raw* __restrict__ xP, float* __restrict__ out,
     int CO, int N, int K, ConvGeom g, int k_per_split) {
@@ -573,7 +575,7 @@ __global__ __launch_bounds__(256, 2) void conv_implicit_wgrad_kernel(
   alignas(16) bf16raw rA[ACALLS * 16];
   alignas(16) bf16raw rB[BCALLS * 16];
 
-  floatx4 acc[4][4] = {};
+  floatx4 acc[MFR][4] = {};
 
 #define WG_LOAD(kt)                                                              \
   do {                                                                           \
@@ -599,9 +601,9 @@ __global__ __launch_bounds__(256, 2) void conv_implicit_wgrad_kernel(
 
 #pragma unroll
     for (int sub = 0; sub < 2; ++sub) {
-      cfrag_t a[4], b[4];
+      cfrag_t a[MFR], b[4];
 #pragma unroll
-      for (int mi = 0; mi < 4; ++mi) {
+      for (int mi = 0; mi < MFR; ++mi) {
         int row = wr * 64 + mi * 16 + l15;
         a[mi] = *(const cfrag_t*)&As[row * LP + sub * 32 + ic_swz(row, kg * 8)];
       }
@@ -611,7 +613,7 @@ __global__ __launch_bounds__(256, 2) void conv_implicit_wgrad_kernel(
         b[ni] = *(const cfrag_t*)&Bs[rowb * LP + sub * 32 + ic_swz(rowb, kg * 8)];
       }
 #pragma unroll
-      for (int mi = 0; mi < 4; ++mi)
+      for (int mi = 0; mi < MFR; ++mi)
 #pragma unroll
         for (int ni = 0; ni < 4; ++ni)
           acc[mi][ni] =
@@ -623,7 +625,7 @@ __global__ __launch_bounds__(256, 2) void conv_implicit_wgrad_kernel(
   const int m_base = m0 + wr * 64;
   const int n_base = n0 + wc * 64;
 #pragma unroll
-  for (int mi = 0; mi < 4; ++mi) {
+  for (int mi = 0; mi < MFR; ++mi) {
 #pragma unroll
     for (int ni = 0; ni < 4; ++ni) {
       int n = n_base + ni * 16 + l15;
@@ -701,8 +703,9 @@ extern "C" hipError_t launch_conv_implicit_wgrad(const bf16raw* dz, const bf16ra
   dim3 grid((unsigned)ceil_div_i64(CO, bm), (unsigned)ceil_div_i64(N, bn), (unsigned)zs);
   if (ws != nullptr) {
     if (flat)
-      conv_implicit_wgrad_kernel<true, 1, 4><<<grid, dim3(256), 0, stream>>>(dz, xP, ws, CO, N,
-                                                                             K, g, kps);
+      // CO <= 32 -> only 2 of the 4 row fragments are live
+      conv_implicit_wgrad_kernel<true, 1, 4, 2><<<grid, dim3(256), 0, stream>>>(dz, xP, ws, CO,
+                                                                                N, K, g, kps);
     else
       conv_implicit_wgrad_kernel<true, 2, 2><<<grid, dim3(256), 0, stream>>>(dz, xP, ws, CO, N,
                                                                              K, g, kps);
@@ -713,8 +716,8 @@ extern "C" hipError_t launch_conv_implicit_wgrad(const bf16raw* dz, const bf16ra
     ic_wgrad_reduce_kernel<<<dim3((unsigned)rg), dim3(256), 0, stream>>>(ws, dw, MN, zs);
   } else {
     if (flat)
-      conv_implicit_wgrad_kernel<false, 1, 4><<<grid, dim3(256), 0, stream>>>(dz, xP, dw, CO, N,
-                                                                              K, g, kps);
+      conv_implicit_wgrad_kernel<false, 1, 4, 2><<<grid, dim3(256), 0, stream>>>(dz, xP, dw, CO,
+                                                                                 N, K, g, kps);
     else
       conv_implicit_wgrad_kernel<false, 2, 2><<<grid, dim3(256), 0, stream>>>(dz, xP, dw, CO, N,
                                                                               K, g, kps);
