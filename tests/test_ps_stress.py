@@ -19,6 +19,7 @@ import torch.nn as nn
 from sparktorch_amd.compat.local import free_port
 from sparktorch_amd.models.simple_net import Net
 from sparktorch_amd.parallel.hogwild import (
+    get_main,
     get_state_dict,
     put_deltas_to_server,
     put_early_stop,
@@ -116,3 +117,63 @@ def test_ps_windowed_early_stop_under_concurrency():
         assert any(stop_seen), "early stop never tripped on a flat loss series"
     finally:
         srv.stop_server()
+
+
+def test_http_helpers_retry_exactly_once():
+    """The hogwild HTTP helpers retry exactly once (reference
+    hogwild.py:31-57): one transient failure is absorbed, two are fatal —
+    except put_deltas_to_server, which silently tolerates a lost update
+    (reference hogwild.py:44-49)."""
+    import json
+    from http.server import BaseHTTPRequestHandler, HTTPServer
+
+    from sparktorch_amd.compat.local import free_port
+
+    calls = {"n": 0, "fail_first": 0}
+
+    class Flaky(BaseHTTPRequestHandler):
+        def log_message(self, *a):
+            pass
+
+        def _go(self):
+            calls["n"] += 1
+            if calls["n"] <= calls["fail_first"]:
+                self.send_response(503)
+                self.end_headers()
+                self.wfile.write(b"down")
+                return
+            self.send_response(200)
+            self.end_headers()
+            self.wfile.write(json.dumps({"stop": False}).encode())
+
+        def do_GET(self):
+            self._go()
+
+        def do_POST(self):
+            self.rfile.read(int(self.headers.get("Content-Length", 0) or 0))
+            self._go()
+
+    port = free_port()
+    httpd = HTTPServer(("127.0.0.1", port), Flaky)
+    t = threading.Thread(target=httpd.serve_forever, daemon=True)
+    t.start()
+    ps = "127.0.0.1:%d" % port
+    try:
+        # one transient failure -> absorbed by the single retry
+        calls.update(n=0, fail_first=1)
+        assert get_main(ps) is not None
+        assert calls["n"] == 2
+
+        # two failures -> raises (no third attempt)
+        calls.update(n=0, fail_first=2)
+        with pytest.raises(RuntimeError):
+            get_main(ps)
+        assert calls["n"] == 2
+
+        # put_deltas tolerates total failure silently
+        calls.update(n=0, fail_first=99)
+        put_deltas_to_server(ps, [torch.zeros(2)])
+        assert calls["n"] == 2
+    finally:
+        httpd.shutdown()
+        httpd.server_close()
