@@ -177,3 +177,41 @@ def test_http_helpers_retry_exactly_once():
     finally:
         httpd.shutdown()
         httpd.server_close()
+
+
+def test_barrier_fail_fast_on_dead_worker():
+    """A barrier task that dies without reporting (OOM-kill shape) must fail
+    the stage within seconds — not hang until the stage timeout.  Guards the
+    dead-worker detection in vendor/pyspark/rdd.py:_collect_barrier."""
+    import os
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ)
+    env["PYTHONPATH"] = os.pathsep.join(
+        [os.path.join(repo, "vendor"), os.path.join(repo, "tests"), repo]
+        + [p for p in env.get("PYTHONPATH", "").split(os.pathsep) if p]
+    )
+    code = (
+        "import time\n"
+        "from pyspark import SparkContext\n"
+        "from barrier_crash_worker import crash_partition_one\n"
+        "sc = SparkContext()\n"
+        "rdd = sc.parallelize(list(range(30)), 3)\n"
+        "t0 = time.time()\n"
+        "try:\n"
+        "    rdd.barrier().mapPartitionsWithIndex(crash_partition_one).collect(timeout_s=120)\n"
+        "except Exception as e:\n"
+        "    took = time.time() - t0\n"
+        "    msg = str(e)\n"
+        "    assert 'died' in msg or 'exit' in msg.lower(), msg\n"
+        "    assert took < 60, took\n"
+        "    print('FAILFAST_OK %.1fs' % took)\n"
+        "else:\n"
+        "    raise SystemExit('barrier stage did not fail')\n"
+    )
+    r = subprocess.run([sys.executable, "-c", code], env=env, cwd=repo,
+                       capture_output=True, text=True, timeout=150)
+    assert r.returncode == 0, r.stdout + r.stderr
+    assert "FAILFAST_OK" in r.stdout, r.stdout + r.stderr
