@@ -203,3 +203,36 @@ def test_public_api_surface_matches_reference():
         "device", "useBarrier", "useVectorOut",
     ):
         assert hasattr(est, "get" + p[0].upper() + p[1:]) or hasattr(est, p), p
+
+
+def test_select_backend_ranks_per_host(monkeypatch):
+    """The nccl-vs-gloo choice compares ranks-on-THIS-host to visible GPUs,
+    not global world size (the round-1 advisor's multi-node finding)."""
+    import warnings
+
+    import torch
+
+    from sparktorch_amd.parallel.rendezvous import select_backend
+
+    # explicit backend always wins
+    assert select_backend("cuda:0", backend="gloo", world_size=8) == "gloo"
+    # cpu device -> gloo regardless
+    assert select_backend("cpu", world_size=8) == "gloo"
+
+    monkeypatch.setattr(torch.cuda, "is_available", lambda: True)
+    monkeypatch.setattr(torch.cuda, "device_count", lambda: 8)
+
+    # 2 nodes x 8 GPUs: world 16 but only 8 ranks local -> still nccl
+    assert select_backend("cuda", world_size=16, local_ranks=8) == "nccl"
+    # oversubscribed host -> gloo with a warning
+    with warnings.catch_warnings(record=True) as w:
+        warnings.simplefilter("always")
+        assert select_backend("cuda", world_size=16, local_ranks=16) == "gloo"
+    assert any("gloo" in str(x.message) for x in w)
+    # no local_ranks: LOCAL_WORLD_SIZE env consulted before assuming world-local
+    monkeypatch.setenv("LOCAL_WORLD_SIZE", "4")
+    assert select_backend("cuda", world_size=32) == "nccl"
+    monkeypatch.delenv("LOCAL_WORLD_SIZE")
+    with warnings.catch_warnings(record=True):
+        warnings.simplefilter("ignore")
+        assert select_backend("cuda", world_size=32) == "gloo"
