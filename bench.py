@@ -308,6 +308,13 @@ def run_time_to_loss(args, model, device, on_gpu, world, rank) -> int:
     while steps < args.max_steps:
         loss = trainer.train_step(x, y)
         steps += 1
+        if world > 1:
+            # collective stop: every rank must take the same branch or the
+            # next step's all-reduce deadlocks (rank-seeded data -> local
+            # losses differ); stop only when the WORST rank hits target
+            lt = torch.tensor([loss], device=device if on_gpu else "cpu")
+            dist.all_reduce(lt, op=dist.ReduceOp.MAX)
+            loss = float(lt)
         if loss < args.target_loss:
             break
     if on_gpu:
