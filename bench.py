@@ -336,7 +336,8 @@ def run_time_to_loss(args, model, device, on_gpu, world, rank) -> int:
             "data": "synthetic",
             "config": {"model": args.model, "global_batch": args.batch,
                        "target_loss": args.target_loss, "reached": loss < args.target_loss,
-                       "final_loss": loss, "seq_len": None, "parallelism": "dp1"},
+                       "final_loss": loss, "seq_len": None,
+                       "parallelism": "dp%d" % (world if world > 1 else 1)},
         }), flush=True)
     if world > 1:
         dist.destroy_process_group()
