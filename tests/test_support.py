@@ -236,3 +236,30 @@ def test_select_backend_ranks_per_host(monkeypatch):
     with warnings.catch_warnings(record=True):
         warnings.simplefilter("ignore")
         assert select_backend("cuda", world_size=32) == "gloo"
+
+
+def test_pick_device_refuses_silent_cpu_fallback():
+    """device='cuda' without a GPU must raise, not silently train on CPU
+    (the round-end 'native code not loaded' guard)."""
+    import torch
+
+    from sparktorch_amd.parallel.rendezvous import pick_device
+
+    if torch.cuda.is_available():
+        pytest.skip("host has a GPU")
+    with pytest.raises(RuntimeError, match="refusing"):
+        pick_device("cuda", rank=0)
+    assert pick_device("cpu", rank=0) == "cpu"
+    assert pick_device("", rank=3) == "cpu"
+
+
+def test_ops_ext_raises_when_extension_missing(monkeypatch):
+    """ops.ext() must fail loudly (no eager fallback) when _sparkhip is
+    absent — GPU kernels silently replaced by torch would invalidate every
+    benchmark."""
+    import sparktorch_amd.ops as ops
+
+    monkeypatch.setattr(ops, "_EXT", False)
+    assert ops.available() is False
+    with pytest.raises(RuntimeError, match="_sparkhip"):
+        ops.ext()
