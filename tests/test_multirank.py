@@ -136,3 +136,29 @@ def test_bench_torchrun_contract_gloo(world, tmp_path):
     assert out["config"]["parallelism"] == "dp%d" % world
     assert out["config"]["global_batch"] == 512 * world
     assert out["value"] > 0
+
+
+def test_bench_time_to_loss_world2_no_deadlock(tmp_path):
+    """time_to_loss at world>1: ranks see different data, so the stop
+    decision must be collective — a rank breaking on its local loss would
+    leave the others deadlocked in the next all-reduce.  Runs the real
+    torchrun launch on gloo."""
+    port = _free_port()
+    proc = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run",
+         "--nnodes=1", "--nproc-per-node", "2",
+         "--master-addr", "127.0.0.1", "--master-port", str(port),
+         os.path.join(REPO, "bench.py"),
+         "--mode", "time_to_loss", "--batch", "1024",
+         "--max-steps", "30", "--target-loss", "2.1"],
+        capture_output=True,
+        text=True,
+        timeout=420,
+        cwd=REPO,
+    )
+    assert proc.returncode == 0, proc.stdout + "\n" + proc.stderr
+    line = next(l for l in proc.stdout.splitlines() if l.startswith("{"))
+    out = json.loads(line)
+    assert out["metric"] == "time_to_loss_s"
+    assert out["n_gpus"] == 2
+    assert out["config"]["parallelism"] == "dp2"
