@@ -545,6 +545,33 @@ def hip_max_pool2d_nhwc(x, kernel_size: int, stride: Optional[int] = None, paddi
     return y.permute(0, 2, 3, 1).contiguous()
 
 
+class _ReluFn(torch.autograd.Function):
+    """Standalone ReLU on the native path.  Forward reuses the relu_bwd
+    kernel as x*(x>0)==relu(x) — no extra kernel needed.  Exists so
+    ``relu(maxpool(z))`` can replace ``maxpool(relu(z))`` (equal for the
+    monotone max), moving the backward mask onto the pooled (4x smaller for
+    a 2x2 pool) tensor instead of the full conv output."""
+
+    @staticmethod
+    def forward(ctx, x):
+        y = ops.ext().relu_bwd(x, x)
+        ctx.save_for_backward(y)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (y,) = ctx.saved_tensors
+        return ops.ext().relu_bwd(dy.contiguous().to(torch.bfloat16), y)
+
+
+def hip_relu(x):
+    if x.is_cuda:
+        if x.dtype != torch.bfloat16:
+            x = x.to(torch.bfloat16)
+        return _ReluFn.apply(x.contiguous())
+    return F.relu(x)
+
+
 class _GlobalAvgPoolNHWCFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x):

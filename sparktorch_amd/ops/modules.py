@@ -28,6 +28,7 @@ from sparktorch_amd.ops.functional import (
     hip_linear,
     hip_max_pool2d,
     hip_max_pool2d_nhwc,
+    hip_relu,
     hip_mse,
 )
 
@@ -208,7 +209,10 @@ class MnistCNNFused(nn.Module):
     def __init__(self):
         super().__init__()
         self.conv1 = HipConv2d(1, 16, kernel_size=5, activation="relu", layout="nhwc")
-        self.conv2 = HipConv2d(16, 32, kernel_size=3, activation="relu", layout="nhwc")
+        # conv2's ReLU commutes with the monotone maxpool: applying it AFTER
+        # the pool (on the 4x smaller tensor) gives bit-equal activations
+        # while the backward mask pass shrinks 4x
+        self.conv2 = HipConv2d(16, 32, kernel_size=3, layout="nhwc")
         self.dropout = HipDropout(p=0.25, channel_wise=True, layout="nhwc")
         self.fc = HipLinear(3872, 10)
 
@@ -218,7 +222,7 @@ class MnistCNNFused(nn.Module):
         x = x.view(-1, 1, 28, 28).permute(0, 2, 3, 1).contiguous()  # NHWC entry
         x = self.conv1(x)
         x = self.conv2(x)
-        x = hip_max_pool2d_nhwc(x, 2)
+        x = hip_relu(hip_max_pool2d_nhwc(x, 2))
         x = self.dropout(x)
         B, H, W, C = x.shape
         x = x.reshape(B, H * W * C)
