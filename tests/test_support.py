@@ -263,3 +263,32 @@ def test_ops_ext_raises_when_extension_missing(monkeypatch):
     assert ops.available() is False
     with pytest.raises(RuntimeError, match="_sparkhip"):
         ops.ext()
+
+
+def test_relu_pool_commute_identity():
+    """MnistCNNFused applies conv2's ReLU after the 2x2 maxpool; this is
+    only valid because relu(maxpool(z)) == maxpool(relu(z)) for the
+    monotone max — pinned here over random tensors including all-negative
+    windows and exact ties."""
+    import torch.nn.functional as F
+
+    torch.manual_seed(0)
+    for _ in range(5):
+        z = torch.randn(4, 16, 8, 8)
+        z[0, 0] = -1.0                      # all-negative window
+        z[1, 1, :2, :2] = 0.5               # exact ties inside a window
+        a = F.relu(F.max_pool2d(z, 2))
+        b = F.max_pool2d(F.relu(z), 2)
+        assert torch.equal(a, b)
+
+    # the fused module still matches the eager reference on CPU
+    from sparktorch_amd.models.mnist import MnistCNN
+    from sparktorch_amd.ops.modules import MnistCNNFused
+
+    m = MnistCNNFused()
+    ref = MnistCNN()
+    ref.load_state_dict(m.state_dict())
+    m.eval(), ref.eval()
+    x = torch.randn(8, 784)
+    with torch.no_grad():
+        assert torch.allclose(m(x), ref(x), atol=1e-5)
