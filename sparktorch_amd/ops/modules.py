@@ -200,8 +200,9 @@ class HipDropout(nn.Module):
 
 class MnistCNNFused(nn.Module):
     """The reference example CNN (examples/cnn_network.py:6-24) on the fully
-    native path, channels-last end to end: NHWC implicit-GEMM convs with
-    fused ReLU, NHWC maxpool, counter-based NHWC Dropout2d.  The fc layer
+    native path, channels-last end to end: NHWC implicit-GEMM convs (conv1
+    with fused ReLU; conv2's ReLU runs after the pool — see forward), NHWC
+    maxpool, counter-based NHWC Dropout2d.  The fc layer
     expects the reference's NCHW flatten order, so its (tiny) weight is
     viewed in NHWC order per forward instead of permuting the (huge)
     activation.  state_dict-compatible with models.mnist.MnistCNN."""
