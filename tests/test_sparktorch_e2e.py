@@ -304,3 +304,26 @@ def test_rccl_tunable_params(data_df, general_torch_obj):
     model = stm.fit(data_df)
     res = model.transform(data_df).take(1)
     assert "predictions" in res[0]
+
+
+def test_sync_fit_eight_partitions():
+    """Estimator-level world_size=8 synchronous fit (the 8-GPU node shape on
+    CPU/gloo): 8 barrier tasks, 8-rank rendezvous, bucketed all-reduce, and
+    every rank's final state identical by construction."""
+    rng = np.random.RandomState(7)
+    feats = rng.normal(0.0, 1.0, (320, 10))
+    labels = [float(i % 2) for i in range(320)]
+    df = LocalDataFrame.from_arrays(feats, labels, num_partitions=8)
+    torch_obj = serialize_torch_obj(
+        nn.Sequential(nn.Linear(10, 8), nn.ReLU(), nn.Linear(8, 1)),
+        nn.MSELoss(), torch.optim.Adam, lr=0.01,
+    )
+    est = SparkTorch(
+        inputCol="features", labelCol="label", predictionCol="predicted",
+        torchObj=torch_obj, iters=3, verbose=0, mode="synchronous", partitions=8,
+    )
+    model = est.fit(df)
+    out = model.transform(df)
+    preds = [r["predicted"] for r in out.collect()]
+    assert len(preds) == 320
+    assert all(np.isfinite(p) for p in preds)
